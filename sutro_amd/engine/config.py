@@ -1,0 +1,51 @@
+"""Engine configuration, sized for MI355X (288 GB HBM3E per GPU)."""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from ..models.registry import ModelSpec
+
+KV_BLOCK_SIZE = 32  # tokens per KV page (multiple of 16 for dwordx4-aligned rows)
+
+
+@dataclass
+class EngineConfig:
+    spec: ModelSpec
+    device: str = "cuda"            # "cuda" (=HIP on ROCm) or "cpu"
+    dtype: torch.dtype = torch.bfloat16
+    kv_block_size: int = KV_BLOCK_SIZE
+    max_model_len: int = 8192        # scheduler cap on prompt+output length
+    max_num_seqs: int = 1024         # max concurrently running sequences
+    max_tokens_per_step: int = 16384  # token budget per scheduler step (prefill chunking)
+    gpu_memory_utilization: float = 0.90
+    num_kv_blocks: Optional[int] = None  # None = derive from free memory
+    default_max_new_tokens: int = 256
+    seed: int = 0
+    enforce_eager: bool = False      # disable hipGraph capture of the decode step
+    # tensor parallelism (process group set up by the caller)
+    tp_size: int = 1
+    tp_rank: int = 0
+
+    def __post_init__(self) -> None:
+        if self.device == "cpu":
+            self.dtype = torch.float32
+        if self.max_model_len > self.spec.max_context:
+            self.max_model_len = self.spec.max_context
+
+    def derive_num_kv_blocks(self, free_bytes: int) -> int:
+        """KV blocks that fit in `free_bytes` (both K and V, all layers)."""
+        spec = self.spec
+        kvh = spec.num_kv_heads // self.tp_size if spec.num_kv_heads >= self.tp_size else 1
+        bytes_per_block = (
+            2  # K and V
+            * spec.num_layers
+            * kvh
+            * self.kv_block_size
+            * spec.head_dim
+            * (2 if self.dtype in (torch.bfloat16, torch.float16) else 4)
+        )
+        return max(16, int(free_bytes // bytes_per_block))

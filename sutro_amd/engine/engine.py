@@ -1,0 +1,301 @@
+"""LLMEngine: continuous-batching execution loop (generative + embedding).
+
+One engine per GPU process. `step()` runs one scheduler iteration: build the
+flat token batch, forward the model, sample (or pool, for embedding models),
+update request state, and return per-step stats. The service layer drives it
+from a worker thread/process; `bench.py` drives it directly.
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..models.qwen3 import Qwen3Model
+from ..models.registry import ModelSpec
+from .batch import ForwardBatch, ScheduledBatch
+from .config import EngineConfig
+from .guided import GuidedFSM
+from .kv_cache import PagedKVCache
+from .request import FinishReason, Request, SamplingParams
+from .sampler import Sampler
+from .scheduler import Scheduler
+from .tokenizer import EOS_ID, TOKENIZER_VOCAB, get_tokenizer
+
+
+@dataclass
+class StepStats:
+    scheduled_tokens: int = 0
+    prefill_tokens: int = 0
+    output_tokens: int = 0
+    finished: List[Request] = field(default_factory=list)
+
+
+class LLMEngine:
+    def __init__(self, cfg: EngineConfig, model: Optional[torch.nn.Module] = None):
+        self.cfg = cfg
+        self.spec: ModelSpec = cfg.spec
+        self.device = cfg.device
+        self.tokenizer = get_tokenizer()
+        if model is None:
+            model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len)
+            model.init_random_weights(cfg.seed)
+        self.model = model.to(cfg.device).eval()
+
+        num_blocks = cfg.num_kv_blocks
+        if num_blocks is None:
+            if cfg.device.startswith("cuda"):
+                free, total = torch.cuda.mem_get_info()
+                headroom = int((1.0 - cfg.gpu_memory_utilization) * total)
+                num_blocks = cfg.derive_num_kv_blocks(max(0, free - headroom))
+                # cap the block-table width the scheduler can ever need
+                max_blocks_useful = (
+                    (cfg.max_model_len + cfg.kv_block_size - 1) // cfg.kv_block_size
+                ) * cfg.max_num_seqs
+                num_blocks = min(num_blocks, max_blocks_useful)
+            else:
+                num_blocks = 1024
+        kvh = self.spec.num_kv_heads
+        if cfg.tp_size > 1:
+            kvh = max(1, kvh // cfg.tp_size)
+        self.kv = PagedKVCache(
+            num_layers=self.spec.num_layers,
+            num_blocks=num_blocks,
+            num_kv_heads=kvh,
+            block_size=cfg.kv_block_size,
+            head_dim=self.spec.head_dim,
+            dtype=cfg.dtype,
+            device=cfg.device,
+        )
+        self.scheduler = Scheduler(cfg, self.kv)
+        self.sampler = Sampler(cfg.device, seed=cfg.seed, vocab_limit=TOKENIZER_VOCAB)
+
+        self._next_req_id = 0
+        self._fsms: Dict[int, GuidedFSM] = {}
+        self._next_fsm_id = 0
+        self._emb_sums: Dict[int, torch.Tensor] = {}
+        self.embeddings: Dict[int, np.ndarray] = {}
+
+        # cumulative stats
+        self.total_prompt_tokens = 0
+        self.total_output_tokens = 0
+
+    # ---- admission ----
+
+    def register_fsm(self, schema: dict) -> int:
+        fsm = GuidedFSM.from_schema(schema, device=self.device)
+        fsm_id = self._next_fsm_id
+        self._next_fsm_id += 1
+        self._fsms[fsm_id] = fsm
+        return fsm_id
+
+    def add_request(
+        self,
+        prompt_token_ids: List[int],
+        sampling: Optional[SamplingParams] = None,
+        fsm_id: Optional[int] = None,
+        priority: int = 0,
+        arrival_idx: int = 0,
+        truncate: bool = True,
+    ) -> Request:
+        sampling = sampling or SamplingParams(max_tokens=self.cfg.default_max_new_tokens)
+        limit = self.cfg.max_model_len - max(1, min(sampling.max_tokens, 64))
+        if len(prompt_token_ids) > limit:
+            if truncate:
+                prompt_token_ids = prompt_token_ids[:limit]
+            else:
+                raise ValueError(
+                    f"prompt of {len(prompt_token_ids)} tokens exceeds max_model_len "
+                    f"{self.cfg.max_model_len} and truncate_rows is False"
+                )
+        req = Request(
+            req_id=self._next_req_id,
+            prompt_token_ids=list(prompt_token_ids),
+            sampling=sampling,
+            fsm_id=fsm_id,
+            arrival_idx=arrival_idx,
+        )
+        if fsm_id is not None:
+            req.fsm_state = self._fsms[fsm_id].start_state()
+        self._next_req_id += 1
+        self.scheduler.add_request(req, priority)
+        return req
+
+    def abort_request(self, req: Request) -> None:
+        self.scheduler.abort_request(req)
+
+    def has_work(self) -> bool:
+        return self.scheduler.has_work()
+
+    # ---- the step ----
+
+    def _build_forward_batch(self, sb: ScheduledBatch) -> ForwardBatch:
+        S = len(sb.reqs)
+        T = sb.total_tokens
+        input_ids = np.empty(T, dtype=np.int64)
+        positions = np.empty(T, dtype=np.int64)
+        slots = np.empty(T, dtype=np.int64)
+        qlocs = np.zeros(S + 1, dtype=np.int32)
+        seq_lens = np.empty(S, dtype=np.int32)
+        bs = self.kv.block_size
+        max_blocks = 1
+        cursor = 0
+        for s, (req, c) in enumerate(zip(sb.reqs, sb.num_new_tokens)):
+            start = req.num_computed_tokens
+            table = self.kv.block_tables[req.req_id]
+            max_blocks = max(max_blocks, len(table))
+            for j in range(c):
+                pos = start + j
+                input_ids[cursor] = req.token_at(pos)
+                positions[cursor] = pos
+                slots[cursor] = table[pos // bs] * bs + pos % bs
+                cursor += 1
+            qlocs[s + 1] = cursor
+            seq_lens[s] = start + c
+        block_tables = np.zeros((S, max_blocks), dtype=np.int32)
+        for s, req in enumerate(sb.reqs):
+            t = self.kv.block_tables[req.req_id]
+            block_tables[s, : len(t)] = t
+
+        # rows that need logits: completing prefills + all decodes
+        logits_rows: List[int] = []
+        for s in range(sb.num_prefills):
+            req, c = sb.reqs[s], sb.num_new_tokens[s]
+            if req.num_computed_tokens + c == req.num_prompt_tokens:
+                logits_rows.append(int(qlocs[s + 1]) - 1)
+        for s in range(sb.num_prefills, S):
+            logits_rows.append(int(qlocs[s + 1]) - 1)
+
+        dev = self.device
+        return ForwardBatch(
+            input_ids=torch.from_numpy(input_ids).to(dev),
+            positions=torch.from_numpy(positions).to(dev),
+            slot_mapping=torch.from_numpy(slots).to(dev),
+            block_tables=torch.from_numpy(block_tables).to(dev),
+            seq_lens=torch.from_numpy(seq_lens).to(dev),
+            query_start_locs=torch.from_numpy(qlocs).to(dev),
+            num_decodes_tail=sb.num_decodes,
+            logits_idx=torch.tensor(logits_rows, dtype=torch.long, device=dev),
+            max_seq_len=int(seq_lens.max()) if S else 0,
+            max_query_len=int(max(sb.num_new_tokens)) if S else 0,
+        )
+
+    def _sampling_reqs(self, sb: ScheduledBatch) -> List[Request]:
+        out = []
+        for s in range(sb.num_prefills):
+            req, c = sb.reqs[s], sb.num_new_tokens[s]
+            if req.num_computed_tokens + c == req.num_prompt_tokens:
+                out.append(req)
+        out.extend(sb.reqs[sb.num_prefills:])
+        return out
+
+    @torch.no_grad()
+    def step(self) -> StepStats:
+        sb = self.scheduler.schedule()
+        stats = StepStats()
+        if not sb.reqs:
+            return stats
+        fb = self._build_forward_batch(sb)
+        hidden = self.model(fb, self.kv)
+
+        stats.scheduled_tokens = sb.total_tokens
+
+        if self.spec.embedding:
+            self._embedding_update(sb, fb, hidden, stats)
+            self._advance_computed(sb)
+            stats.prefill_tokens = sb.total_tokens
+            self.total_prompt_tokens += stats.prefill_tokens
+            return stats
+
+        sample_reqs = self._sampling_reqs(sb)
+        if sample_reqs:
+            logits = self.model.compute_logits(hidden[fb.logits_idx])
+            fsm_mask = self._fsm_masks(sample_reqs, logits.device)
+            tokens, lps = self.sampler.sample(logits, sample_reqs, fsm_mask)
+        else:
+            tokens, lps = [], []
+
+        self._advance_computed(sb)
+
+        for req, tok, lp in zip(sample_reqs, tokens, lps):
+            self._apply_sampled(req, int(tok), float(lp), stats)
+        stats.prefill_tokens = sum(sb.num_new_tokens[: sb.num_prefills])
+        self.total_prompt_tokens += stats.prefill_tokens
+        self.total_output_tokens += stats.output_tokens
+        return stats
+
+    def _advance_computed(self, sb: ScheduledBatch) -> None:
+        for req, c in zip(sb.reqs, sb.num_new_tokens):
+            req.num_computed_tokens += c
+
+    def _fsm_masks(self, reqs: List[Request], device) -> Optional[torch.Tensor]:
+        if all(r.fsm_id is None for r in reqs):
+            return None
+        mask = torch.ones((len(reqs), TOKENIZER_VOCAB), dtype=torch.bool, device=device)
+        for i, r in enumerate(reqs):
+            if r.fsm_id is not None:
+                mask[i] = self._fsms[r.fsm_id].mask_for(r.fsm_state).to(device)
+        return mask
+
+    def _apply_sampled(self, req: Request, tok: int, lp: float, stats: StepStats) -> None:
+        sp = req.sampling
+        stop_ids = set(sp.stop_token_ids or ()) | {EOS_ID}
+        if tok in stop_ids:
+            self.scheduler.finish(req, FinishReason.STOP)
+            stats.finished.append(req)
+            return
+        req.output_token_ids.append(tok)
+        req.cumulative_logprob += lp
+        stats.output_tokens += 1
+        if req.fsm_id is not None:
+            fsm = self._fsms[req.fsm_id]
+            req.fsm_state = fsm.advance(req.fsm_state, tok)
+            if fsm.must_stop(req.fsm_state):
+                self.scheduler.finish(req, FinishReason.STOP)
+                stats.finished.append(req)
+                return
+        if len(req.output_token_ids) >= sp.max_tokens or req.total_len >= self.cfg.max_model_len:
+            self.scheduler.finish(req, FinishReason.LENGTH)
+            stats.finished.append(req)
+
+    # ---- embedding mode ----
+
+    def _embedding_update(self, sb: ScheduledBatch, fb: ForwardBatch,
+                          hidden: torch.Tensor, stats: StepStats) -> None:
+        qlocs = fb.query_start_locs
+        for s, (req, c) in enumerate(zip(sb.reqs, sb.num_new_tokens)):
+            a, b = int(qlocs[s]), int(qlocs[s + 1])
+            chunk_sum = hidden[a:b].float().sum(dim=0)
+            if req.req_id in self._emb_sums:
+                self._emb_sums[req.req_id] += chunk_sum
+            else:
+                self._emb_sums[req.req_id] = chunk_sum
+            if req.num_computed_tokens + c == req.num_prompt_tokens:
+                vec = self._emb_sums.pop(req.req_id) / max(1, req.num_prompt_tokens)
+                vec = vec / (vec.norm() + 1e-12)
+                self.embeddings[req.req_id] = vec.cpu().numpy()
+                self.scheduler.finish(req, FinishReason.STOP)
+                stats.finished.append(req)
+
+    # ---- convenience: run a list of prompts to completion (tests/smoke) ----
+
+    def generate(
+        self,
+        prompts: List[str],
+        sampling: Optional[SamplingParams] = None,
+        system_prompt: Optional[str] = None,
+        schema: Optional[dict] = None,
+    ) -> List[str]:
+        fsm_id = self.register_fsm(schema) if schema else None
+        reqs = []
+        for i, p in enumerate(prompts):
+            ids = self.tokenizer.render_prompt(p, system_prompt)
+            sp = sampling or SamplingParams(max_tokens=32, temperature=0.8)
+            reqs.append(self.add_request(ids, sp, fsm_id=fsm_id, arrival_idx=i))
+        while self.has_work():
+            self.step()
+        return [self.tokenizer.decode(r.output_token_ids) for r in reqs]

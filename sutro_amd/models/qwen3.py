@@ -1,0 +1,205 @@
+"""Qwen3-family decoder (dense + MoE) built on the MI355X op set.
+
+One architecture covers the whole registry: RMSNorm + RoPE-NeoX + GQA paged
+attention + SwiGLU MLP (dense) or top-k routed SwiGLU experts (MoE).
+GEMMs go through F.linear (hipBLASLt on ROCm); everything between GEMMs is a
+hand-written HIP kernel on GPU (see sutro_amd/ops).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..engine.batch import ForwardBatch
+from ..engine.kv_cache import PagedKVCache
+from ..ops import torch_ref
+from .registry import ModelSpec
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, size: int, eps: float, dtype: torch.dtype):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(size, dtype=dtype))
+        self.eps = eps
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.rmsnorm(x, self.weight, self.eps)
+
+
+class Qwen3Attention(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int,
+                 tp_size: int = 1):
+        super().__init__()
+        assert spec.num_heads % tp_size == 0
+        self.layer_idx = layer_idx
+        self.num_heads = spec.num_heads // tp_size
+        self.num_kv_heads = max(1, spec.num_kv_heads // tp_size)
+        self.head_dim = spec.head_dim
+        self.q_size = self.num_heads * self.head_dim
+        self.kv_size = self.num_kv_heads * self.head_dim
+        self.scale = 1.0 / math.sqrt(self.head_dim)
+        h = spec.hidden_size
+        self.qkv_proj = nn.Linear(h, self.q_size + 2 * self.kv_size, bias=False,
+                                  dtype=dtype)
+        self.o_proj = nn.Linear(self.q_size, h, bias=False, dtype=dtype)
+        self.qk_norm = spec.qk_norm
+        if spec.qk_norm:
+            self.q_norm = RMSNorm(self.head_dim, spec.rms_eps, dtype)
+            self.k_norm = RMSNorm(self.head_dim, spec.rms_eps, dtype)
+
+    def forward(self, x: torch.Tensor, fb: ForwardBatch, kv: PagedKVCache,
+                cos_sin: torch.Tensor) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = self.qkv_proj(x)
+        q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
+        q = q.view(T, self.num_heads, self.head_dim)
+        k = k.view(T, self.num_kv_heads, self.head_dim)
+        v = v.view(T, self.num_kv_heads, self.head_dim)
+        if self.qk_norm:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
+        q = q.contiguous()
+        k = k.contiguous()
+        v = v.contiguous()
+        q = ops.rope_and_cache(q, k, v, fb.positions, fb.slot_mapping,
+                               kv.k_cache[self.layer_idx], kv.v_cache[self.layer_idx],
+                               cos_sin)
+        o = ops.paged_attention(q, kv.k_cache[self.layer_idx],
+                                kv.v_cache[self.layer_idx], fb.block_tables,
+                                fb.seq_lens, fb.query_start_locs, self.scale,
+                                fb.num_decodes_tail)
+        return self.o_proj(o.view(T, self.q_size))
+
+
+class Qwen3MLP(nn.Module):
+    def __init__(self, hidden: int, intermediate: int, dtype: torch.dtype):
+        super().__init__()
+        self.gate_up_proj = nn.Linear(hidden, 2 * intermediate, bias=False, dtype=dtype)
+        self.down_proj = nn.Linear(intermediate, hidden, bias=False, dtype=dtype)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.silu_mul(self.gate_up_proj(x)))
+
+
+class Qwen3MoE(nn.Module):
+    """Top-k routed SwiGLU experts. Round-1 execution: sort tokens by expert and
+    run per-expert GEMMs on contiguous segments (grouped-GEMM kernel later)."""
+
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype):
+        super().__init__()
+        self.num_experts = spec.num_experts
+        self.top_k = spec.experts_per_token
+        h, m = spec.hidden_size, spec.moe_intermediate_size
+        self.router = nn.Linear(h, spec.num_experts, bias=False, dtype=dtype)
+        self.gate_up = nn.Parameter(torch.empty(spec.num_experts, h, 2 * m, dtype=dtype))
+        self.down = nn.Parameter(torch.empty(spec.num_experts, m, h, dtype=dtype))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, h = x.shape
+        weights, idx = torch_ref.topk_softmax_router(self.router(x), self.top_k)
+        flat_expert = idx.reshape(-1)                      # [T*k]
+        flat_tok = torch.arange(T, device=x.device).repeat_interleave(self.top_k)
+        order = torch.argsort(flat_expert, stable=True)
+        sorted_expert = flat_expert[order]
+        sorted_tok = flat_tok[order]
+        gathered = x[sorted_tok]                           # [T*k, h]
+        out_sorted = torch.empty_like(gathered)
+        counts = torch.bincount(sorted_expert, minlength=self.num_experts)
+        start = 0
+        counts_l = counts.tolist()
+        for e in range(self.num_experts):
+            n = counts_l[e]
+            if n == 0:
+                continue
+            seg = gathered[start:start + n]
+            act = ops.silu_mul(seg @ self.gate_up[e])
+            out_sorted[start:start + n] = act @ self.down[e]
+            start += n
+        # scatter-add back with routing weights
+        out = torch.zeros_like(x, dtype=torch.float32)
+        w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
+        out.index_add_(0, sorted_tok, out_sorted.float() * w_sorted)
+        return out.to(x.dtype)
+
+
+class Qwen3Block(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int):
+        super().__init__()
+        self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
+        self.self_attn = Qwen3Attention(spec, dtype, layer_idx)
+        self.post_attention_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
+        if spec.num_experts > 0:
+            self.mlp = Qwen3MoE(spec, dtype)
+        else:
+            self.mlp = Qwen3MLP(spec.hidden_size, spec.intermediate_size, dtype)
+
+    def forward(self, x, residual, fb: ForwardBatch, kv, cos_sin):
+        if residual is None:
+            residual = x
+            x = self.input_layernorm(x)
+        else:
+            x, residual = ops.fused_add_rmsnorm(x, residual,
+                                                self.input_layernorm.weight,
+                                                self.input_layernorm.eps)
+        x = self.self_attn(x, fb, kv, cos_sin)
+        x, residual = ops.fused_add_rmsnorm(x, residual,
+                                            self.post_attention_layernorm.weight,
+                                            self.post_attention_layernorm.eps)
+        x = self.mlp(x)
+        return x, residual
+
+
+class Qwen3Model(nn.Module):
+    """Full decoder. For embedding specs there is no lm_head; for generative
+    specs `compute_logits` projects selected rows through it."""
+
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, max_len: int):
+        super().__init__()
+        self.spec = spec
+        self.embed_tokens = nn.Embedding(spec.vocab_size, spec.hidden_size, dtype=dtype)
+        self.layers = nn.ModuleList(
+            [Qwen3Block(spec, dtype, i) for i in range(spec.num_layers)]
+        )
+        self.norm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
+        if not spec.embedding:
+            if spec.tie_embeddings:
+                self.lm_head = None
+            else:
+                self.lm_head = nn.Linear(spec.hidden_size, spec.vocab_size,
+                                         bias=False, dtype=dtype)
+        cos_sin = torch_ref.rope_cos_sin(max_len, spec.head_dim, spec.rope_theta)
+        self.register_buffer("cos_sin", cos_sin, persistent=False)
+
+    @torch.no_grad()
+    def forward(self, fb: ForwardBatch, kv: PagedKVCache) -> torch.Tensor:
+        x = self.embed_tokens(fb.input_ids)
+        residual = None
+        for layer in self.layers:
+            x, residual = layer(x, residual, fb, kv, self.cos_sin)
+        x, _ = ops.fused_add_rmsnorm(x, residual, self.norm.weight, self.norm.eps)
+        return x  # [T, hidden]
+
+    @torch.no_grad()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        if self.lm_head is not None:
+            return self.lm_head(hidden)
+        return F.linear(hidden, self.embed_tokens.weight)
+
+    @torch.no_grad()
+    def init_random_weights(self, seed: int = 0) -> None:
+        """Deterministic random init (there is no checkpoint source offline)."""
+        gen = torch.Generator().manual_seed(seed)
+        for name, p in sorted(self.named_parameters()):
+            if p.dim() >= 2:
+                fan_in = p.shape[-1] if "embed" not in name else p.shape[0]
+                std = 0.02 if "embed" in name else (1.0 / math.sqrt(fan_in))
+                with torch.no_grad():
+                    cpu = torch.randn(p.shape, generator=gen, dtype=torch.float32) * std
+                    p.copy_(cpu.to(p.dtype))
+            # norm weights stay at ones
