@@ -1,0 +1,135 @@
+"""Guided decoding: schema->regex->DFA correctness and token masks."""
+
+import json
+
+import pytest
+from pydantic import BaseModel
+
+from sutro_amd.engine.guided import (
+    DFA,
+    GuidedFSM,
+    compile_dfa,
+    schema_to_regex,
+)
+from sutro_amd.engine.tokenizer import BYTE_OFFSET, EOS_ID
+
+
+def dfa_for(schema: dict) -> DFA:
+    return compile_dfa(schema_to_regex(schema))
+
+
+def test_integer_schema():
+    d = dfa_for({"type": "integer"})
+    assert d.matches(b"0")
+    assert d.matches(b"-123")
+    assert d.matches(b"42")
+    assert not d.matches(b"01")
+    assert not d.matches(b"abc")
+    assert not d.matches(b"")
+
+
+def test_bounded_integer():
+    d = dfa_for({"type": "integer", "minimum": 0, "maximum": 10})
+    for v in range(0, 11):
+        assert d.matches(str(v).encode())
+    assert not d.matches(b"11")
+    assert not d.matches(b"-1")
+
+
+def test_number_schema():
+    d = dfa_for({"type": "number"})
+    assert d.matches(b"3.25")
+    assert d.matches(b"-0.5")
+    assert not d.matches(b"3.")
+
+
+def test_string_schema():
+    d = dfa_for({"type": "string"})
+    assert d.matches(b'"hello"')
+    assert d.matches(b'"with \\"escape\\""')
+    assert not d.matches(b"hello")
+    assert not d.matches(b'"unterminated')
+
+
+def test_enum_schema():
+    d = dfa_for({"enum": ["yes", "no"]})
+    assert d.matches(b'"yes"')
+    assert d.matches(b'"no"')
+    assert not d.matches(b'"maybe"')
+
+
+def test_boolean_null():
+    assert dfa_for({"type": "boolean"}).matches(b"true")
+    assert dfa_for({"type": "boolean"}).matches(b"false")
+    assert dfa_for({"type": "null"}).matches(b"null")
+
+
+def test_array_schema():
+    d = dfa_for({"type": "array", "items": {"type": "integer"},
+                 "minItems": 1, "maxItems": 3})
+    assert d.matches(b"[1]")
+    assert d.matches(b"[1,2,3]")
+    assert not d.matches(b"[]")
+    assert not d.matches(b"[1,2,3,4]")
+
+
+def test_object_schema():
+    class M(BaseModel):
+        name: str
+        age: int
+        active: bool
+
+    d = dfa_for(M.model_json_schema())
+    assert d.matches(b'{"name":"bob","age":4,"active":true}')
+    assert not d.matches(b'{"age":4,"name":"bob","active":true}')  # fixed order
+    assert not d.matches(b'{"name":"bob","age":4}')
+
+
+def test_nested_object_with_refs():
+    class Inner(BaseModel):
+        x: int
+
+    class Outer(BaseModel):
+        inner: Inner
+        tag: str
+
+    d = dfa_for(Outer.model_json_schema())
+    assert d.matches(b'{"inner":{"x":1},"tag":"t"}')
+
+
+def test_anyof_optional():
+    d = dfa_for({"anyOf": [{"type": "integer"}, {"type": "null"}]})
+    assert d.matches(b"7")
+    assert d.matches(b"null")
+
+
+def test_fsm_masks_drive_valid_generation():
+    """Walking the FSM greedily by lowest allowed token must produce a string
+    the DFA accepts -- for any schema."""
+    schema = {"type": "object", "properties": {
+        "a": {"type": "integer", "minimum": 0, "maximum": 3},
+        "b": {"enum": ["u", "v"]}}}
+    fsm = GuidedFSM.from_schema(schema)
+    state = fsm.start_state()
+    out = []
+    for _ in range(200):
+        mask = fsm.mask_for(state)
+        allowed = mask.nonzero().flatten().tolist()
+        assert allowed, "FSM must always allow at least one token"
+        tok = allowed[0] if allowed[0] != EOS_ID or len(allowed) == 1 else allowed[1]
+        if tok == EOS_ID:
+            break
+        out.append(tok)
+        state = fsm.advance(state, tok)
+    data = bytes(t - BYTE_OFFSET for t in out)
+    assert fsm.dfa.matches(data)
+    json.loads(data)  # also valid JSON
+
+
+def test_classify_like_schema_has_enum_tail():
+    schema = {"type": "object", "properties": {
+        "scratchpad": {"type": "string", "maxLength": 32},
+        "classification": {"enum": ["Pos", "Neg"]}}}
+    d = dfa_for(schema)
+    assert d.matches(b'{"scratchpad":"ok","classification":"Pos"}')
+    assert not d.matches(b'{"scratchpad":"ok","classification":"Other"}')

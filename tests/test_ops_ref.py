@@ -1,0 +1,128 @@
+"""Torch reference-op correctness vs independent dense formulations.
+
+These same references are the oracle the HIP kernels are tested against on GPU
+(tests/test_gpu_kernels.py)."""
+
+import math
+
+import pytest
+import torch
+
+from sutro_amd.ops import torch_ref as R
+
+
+def test_rmsnorm_matches_formula():
+    x = torch.randn(5, 64)
+    w = torch.randn(64)
+    y = R.rmsnorm(x, w, 1e-6)
+    ref = x / torch.sqrt((x ** 2).mean(-1, keepdim=True) + 1e-6) * w
+    assert torch.allclose(y, ref, atol=1e-5)
+
+
+def test_fused_add_rmsnorm():
+    x, res = torch.randn(4, 32), torch.randn(4, 32)
+    y, new_res = R.fused_add_rmsnorm(x, res, torch.ones(32), 1e-6)
+    assert torch.allclose(new_res, x + res, atol=1e-6)
+    assert torch.allclose(y, R.rmsnorm(x + res, torch.ones(32), 1e-6), atol=1e-6)
+
+
+def test_silu_mul():
+    x = torch.randn(3, 16)
+    y = R.silu_mul(x)
+    g, u = x.chunk(2, -1)
+    assert torch.allclose(y, torch.nn.functional.silu(g) * u, atol=1e-5)
+
+
+def test_rope_preserves_norm_and_position0_identity():
+    T, H, D = 6, 2, 8
+    q = torch.randn(T, H, D)
+    k = torch.randn(T, 1, D)
+    cs = R.rope_cos_sin(32, D, 10000.0)
+    pos = torch.zeros(T, dtype=torch.long)
+    q2, k2 = R.apply_rope(q, k, pos, cs)
+    assert torch.allclose(q2, q, atol=1e-6)  # position 0 = identity
+    pos = torch.arange(T)
+    q3, _ = R.apply_rope(q, k, pos, cs)
+    assert torch.allclose(q3.norm(dim=-1), q.norm(dim=-1), atol=1e-4)
+
+
+def test_rope_relative_property():
+    """<rope(q,p) , rope(k,p+d)> depends only on d (per head-dim pair)."""
+    D = 8
+    cs = R.rope_cos_sin(64, D, 10000.0)
+    q = torch.randn(1, 1, D)
+    k = torch.randn(1, 1, D)
+    dots = []
+    for p in (0, 5, 11):
+        qp, _ = R.apply_rope(q, q.clone(), torch.tensor([p]), cs)
+        kp, _ = R.apply_rope(k, k.clone(), torch.tensor([p + 3]), cs)
+        dots.append((qp * kp).sum().item())
+    assert max(dots) - min(dots) < 1e-4
+
+
+def test_paged_attention_matches_sdpa():
+    """Paged attention over scattered blocks == dense causal SDPA."""
+    torch.manual_seed(0)
+    bs, hq, hk, d = 4, 4, 2, 16
+    num_blocks = 32
+    kc = torch.zeros(num_blocks, hk, bs, d)
+    vc = torch.zeros(num_blocks, hk, bs, d)
+    # two sequences with different lengths; new tokens = suffix
+    seq_lens = [10, 7]
+    new_counts = [3, 1]
+    tables = [[5, 9, 2], [7, 30]]
+    qs = []
+    full_k = []
+    full_v = []
+    slot_rows = []
+    for s, L in enumerate(seq_lens):
+        K = torch.randn(L, hk, d)
+        V = torch.randn(L, hk, d)
+        full_k.append(K)
+        full_v.append(V)
+        for pos in range(L):
+            b, off = tables[s][pos // bs], pos % bs
+            kc[b, :, off] = K[pos]
+            vc[b, :, off] = V[pos]
+        qs.append(torch.randn(new_counts[s], hq, d))
+    q = torch.cat(qs, 0)
+    max_blocks = max(len(t) for t in tables)
+    bt = torch.zeros(2, max_blocks, dtype=torch.int32)
+    for s, t in enumerate(tables):
+        bt[s, :len(t)] = torch.tensor(t, dtype=torch.int32)
+    qlocs = torch.tensor([0, 3, 4], dtype=torch.int32)
+    out = R.paged_attention(q, kc, vc, bt,
+                            torch.tensor(seq_lens, dtype=torch.int32),
+                            qlocs, 1.0 / math.sqrt(d))
+    # dense reference per sequence/head
+    group = hq // hk
+    for s, L in enumerate(seq_lens):
+        nq = new_counts[s]
+        for h in range(hq):
+            K = full_k[s][:, h // group]      # [L, d]
+            V = full_v[s][:, h // group]
+            for i in range(nq):
+                pos = L - nq + i
+                qv = qs[s][i, h]
+                att = (K[:pos + 1] @ qv) / math.sqrt(d)
+                p = torch.softmax(att, 0)
+                ref = p @ V[:pos + 1]
+                got = out[int(qlocs[s]) + i, h]
+                assert torch.allclose(got, ref, atol=1e-4), (s, h, i)
+
+
+def test_mean_pool_normalize():
+    h = torch.randn(7, 16)
+    locs = torch.tensor([0, 3, 7], dtype=torch.int32)
+    out = R.mean_pool_normalize(h, locs)
+    ref0 = h[:3].mean(0)
+    ref0 = ref0 / ref0.norm()
+    assert torch.allclose(out[0], ref0, atol=1e-5)
+    assert abs(out[1].norm().item() - 1.0) < 1e-5
+
+
+def test_topk_router_normalized():
+    logits = torch.randn(5, 8)
+    w, idx = R.topk_softmax_router(logits, 2)
+    assert torch.allclose(w.sum(-1), torch.ones(5), atol=1e-6)
+    assert idx.shape == (5, 2)

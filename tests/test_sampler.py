@@ -1,0 +1,79 @@
+"""Sampler correctness: greedy, masks, seeds, logprobs, top-k/top-p."""
+
+import math
+
+import numpy as np
+import torch
+
+from sutro_amd.engine.request import Request, SamplingParams
+from sutro_amd.engine.sampler import Sampler, seeded_uniform
+
+
+def _reqs(n, **kw):
+    return [Request(req_id=i, prompt_token_ids=[3, 4],
+                    sampling=SamplingParams(**kw)) for i in range(n)]
+
+
+def test_greedy_is_argmax():
+    s = Sampler("cpu", vocab_limit=16)
+    logits = torch.randn(4, 32)
+    toks, lps = s.sample(logits, _reqs(4, temperature=0.0))
+    assert toks == logits[:, :16].argmax(dim=-1).tolist()
+
+
+def test_logprob_matches_log_softmax():
+    s = Sampler("cpu", vocab_limit=8)
+    logits = torch.randn(1, 8)
+    toks, lps = s.sample(logits, _reqs(1, temperature=0.0))
+    ref = torch.log_softmax(logits[0, :8], dim=-1)[toks[0]].item()
+    assert abs(lps[0] - ref) < 1e-5
+
+
+def test_fsm_mask_restricts_support():
+    s = Sampler("cpu", vocab_limit=8)
+    logits = torch.zeros(3, 8)
+    mask = torch.zeros(3, 8, dtype=torch.bool)
+    mask[:, 5] = True
+    toks, _ = s.sample(logits, _reqs(3, temperature=1.0), fsm_mask=mask)
+    assert toks == [5, 5, 5]
+
+
+def test_top_k_restricts_support():
+    s = Sampler("cpu", vocab_limit=16)
+    logits = torch.arange(16, dtype=torch.float32).repeat(64, 1)
+    toks, _ = s.sample(logits, _reqs(64, temperature=1.0, top_k=2))
+    assert set(toks).issubset({14, 15})
+
+
+def test_top_p_restricts_support():
+    s = Sampler("cpu", vocab_limit=4)
+    # probs ~ [0.97, 0.01, 0.01, 0.01] with top_p=0.5 -> always token 0
+    logits = torch.tensor([[8.0, 0.0, 0.0, 0.0]]).repeat(32, 1)
+    toks, _ = s.sample(logits, _reqs(32, temperature=1.0, top_p=0.5))
+    assert set(toks) == {0}
+
+
+def test_seeded_stream_deterministic():
+    u1 = seeded_uniform(np.array([7, 7], dtype=np.uint64),
+                        np.array([1, 2], dtype=np.uint64))
+    u2 = seeded_uniform(np.array([7, 7], dtype=np.uint64),
+                        np.array([1, 2], dtype=np.uint64))
+    assert (u1 == u2).all()
+    assert u1[0] != u1[1]
+    assert ((0 <= u1) & (u1 < 1)).all()
+
+
+def test_temperature_sampling_distribution():
+    """Sampling should roughly follow softmax probabilities."""
+    s = Sampler("cpu", seed=123, vocab_limit=4)
+    logits = torch.tensor([[2.0, 1.0, 0.0, -1.0]]).repeat(2000, 1)
+    toks, _ = s.sample(logits, _reqs(2000, temperature=1.0))
+    counts = np.bincount(toks, minlength=4) / 2000
+    expect = torch.softmax(torch.tensor([2.0, 1.0, 0.0, -1.0]), 0).numpy()
+    assert np.abs(counts - expect).max() < 0.05
+
+
+def test_cumulative_logprob_confidence_bounds():
+    lp = -0.5
+    conf = math.exp(lp)
+    assert 0 < conf < 1

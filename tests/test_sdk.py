@@ -1,0 +1,252 @@
+"""Client-surface tests against the in-process local service (tiny CPU engines),
+plus transport retry-policy tests with mocked HTTP (mirroring the reference's
+`tests/test_sdk.py` mechanisms)."""
+
+import json
+import os
+import time
+from unittest.mock import MagicMock, patch
+
+import pandas as pd
+import pytest
+
+from sutro_amd.interfaces import JobStatus
+
+
+def _await(client, job_id, timeout=120):
+    return client.await_job_completion(job_id, timeout=timeout)
+
+
+SP = {"max_tokens": 8, "temperature": 0.8}
+
+
+def test_infer_results_input_ordered(local_client):
+    rows = [f"row number {i}" for i in range(6)]
+    job_id = local_client.infer(rows, model="qwen-3.5-2b", stay_attached=False,
+                                sampling_params=SP)
+    df = _await(local_client, job_id)
+    assert len(df) == 6
+    assert list(df.columns)[0] == "inference_result"
+    # per-row confidence present
+    assert "confidence_score" in df.columns
+
+
+def test_infer_include_inputs_and_logprobs(local_client):
+    job_id = local_client.infer(["a", "b"], model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    _await(local_client, job_id)
+    df = local_client.get_job_results(job_id, include_inputs=True,
+                                      include_cumulative_logprobs=True,
+                                      disable_cache=True)
+    assert list(df.columns)[:1] == ["inputs"]
+    assert df["inputs"].tolist() == ["a", "b"]
+    assert "cumulative_logprobs" in df.columns
+    assert all(lp <= 0 for lp in df["cumulative_logprobs"])
+
+
+def test_name_description_limits(local_client):
+    with pytest.raises(ValueError):
+        local_client.infer(["x"], name="n" * 46, stay_attached=False)
+    with pytest.raises(ValueError):
+        local_client.infer(["x"], description="d" * 513, stay_attached=False)
+
+
+def test_id_column_only_for_url(local_client):
+    with pytest.raises(ValueError):
+        local_client.infer(["x"], id_column="id", stay_attached=False)
+
+
+def test_dry_run_cost_estimate(local_client):
+    est = local_client.infer(["some input"] * 10, model="qwen-3.5-2b",
+                             dry_run=True)
+    assert isinstance(est, float) and est > 0
+
+
+def test_job_status_and_list(local_client):
+    job_id = local_client.infer(["hello"], model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    _await(local_client, job_id)
+    assert local_client.get_job_status(job_id) == JobStatus.SUCCEEDED.value
+    jobs = local_client.list_jobs()
+    assert job_id in jobs["job_id"].tolist()
+    row = jobs[jobs.job_id == job_id].iloc[0]
+    assert row["num_rows"] == 1
+    assert row["input_tokens"] > 0
+
+
+def test_results_cache_hit_skips_service(local_client):
+    job_id = local_client.infer(["cache me"], model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    df1 = _await(local_client, job_id)
+    with patch.object(local_client, "do_request") as mocked:
+        df2 = local_client.get_job_results(job_id)
+        mocked.assert_not_called()
+    assert df1["inference_result"].tolist() == df2["inference_result"].tolist()
+
+
+def test_structured_output_unpack(local_client):
+    schema = {"type": "object", "properties": {
+        "label": {"enum": ["A", "B"]},
+        "score": {"type": "integer", "minimum": 0, "maximum": 5}}}
+    job_id = local_client.infer(["classify this"], model="qwen-3.5-2b",
+                                output_schema=schema, stay_attached=False)
+    df = _await(local_client, job_id)
+    assert "label" in df.columns and "score" in df.columns
+    assert df["label"].iloc[0] in ("A", "B")
+    assert 0 <= int(df["score"].iloc[0]) <= 5
+
+
+def test_unpack_collision_raises(local_client):
+    schema = {"type": "object", "properties": {
+        "inference_result": {"enum": ["A"]}}}
+    job_id = local_client.infer(["x"], model="qwen-3.5-2b",
+                                output_schema=schema, stay_attached=False)
+    local_client.await_job_completion(job_id, obtain_results=False)
+    with pytest.raises(ValueError):
+        local_client.get_job_results(job_id)
+
+
+def test_with_original_df_join(local_client):
+    src = pd.DataFrame({"text": ["u", "v"], "meta": [1, 2]})
+    job_id = local_client.infer(src, column="text", model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    local_client.await_job_completion(job_id, obtain_results=False)
+    df = local_client.get_job_results(job_id, with_original_df=src)
+    assert list(df["meta"]) == [1, 2]
+    assert "inference_result" in df.columns
+
+
+def test_cancel_job(local_client):
+    job_id = local_client.infer([f"r{i}" for i in range(50)],
+                                model="qwen-3.5-2b", stay_attached=False,
+                                sampling_params={"max_tokens": 200})
+    out = local_client.cancel_job(job_id)
+    status = out["job_status"][job_id]
+    assert status in (JobStatus.CANCELLED.value, JobStatus.CANCELLING.value)
+
+
+def test_quotas_shape(local_client):
+    quotas = local_client.get_quotas()
+    assert len(quotas) >= 2
+    assert {"row_quota", "token_quota"} <= set(quotas[0])
+
+
+def test_row_quota_enforced(local_client):
+    svc = local_client._local_transport().service
+    svc.quotas[0]["row_quota"] = 2
+    with pytest.raises(RuntimeError):
+        local_client.infer(["a", "b", "c"], model="qwen-3.5-2b",
+                           stay_attached=False)
+    svc.quotas[0]["row_quota"] = 100_000
+
+
+def test_try_authentication(local_client):
+    assert local_client.try_authentication("any-key")["authenticated"] is True
+
+
+def test_datasets_roundtrip(local_client, tmp_path):
+    ds = local_client.create_dataset()
+    assert ds.startswith("dataset-")
+    df = pd.DataFrame({"text": ["hello", "world"]})
+    local_client.upload_to_dataset(ds, data=df)
+    files = local_client.list_dataset_files(ds)
+    assert len(files) == 1
+    job_id = local_client.infer(ds, column="text", model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    res = _await(local_client, job_id)
+    assert len(res) == 2
+    # download round trip
+    local_client.download_from_dataset(ds, files[0], str(tmp_path))
+    assert (tmp_path / files[0]).exists()
+
+
+def test_functions_create_and_run(local_client):
+    local_client.create_function("sentiment-fn", model="qwen-3.5-2b",
+                                 system_prompt="classify sentiment",
+                                 output_schema={"type": "object", "properties": {
+                                     "label": {"enum": ["pos", "neg"]}}})
+    out = local_client.run_function("sentiment-fn", "I love it")
+    assert set(out) >= {"response", "confidence", "predictions", "run_id", "usage"}
+    assert json.loads(out["response"])["label"] in ("pos", "neg")
+
+
+def test_batch_run_function(local_client):
+    local_client.create_function("batch-fn", model="qwen-3.5-2b")
+    df = pd.DataFrame({"a": [1, 2], "b": ["x", "y"]})
+    job_id = local_client.batch_run_function(df, "batch-fn")
+    res = _await(local_client, job_id)
+    assert len(res) == 2
+
+
+def test_reasoning_model_output_unpacks(local_client):
+    job_id = local_client.infer(["think about this"],
+                                model="qwen-3.5-2b-thinking",
+                                stay_attached=False,
+                                sampling_params={"max_tokens": 4096})
+    df = _await(local_client, job_id)
+    assert "content" in df.columns and "reasoning_content" in df.columns
+
+
+def test_progress_stream_protocol(local_client):
+    svc = local_client._local_transport().service
+    job_id = local_client.infer(["p1", "p2"], model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    updates = list(svc.stream_progress(job_id, poll=0.02))
+    kinds = {u["update_type"] for u in updates}
+    assert "progress" in kinds and "tokens" in kinds
+    progress_vals = [u["result"] for u in updates if u["update_type"] == "progress"]
+    assert progress_vals == sorted(progress_vals)  # monotone
+    assert progress_vals[-1] == 2
+
+
+def test_embedding_job_results(local_client):
+    df = local_client.embed(["one", "two", "three"],
+                            model="qwen-3-embedding-0.6b")
+    vecs = df["inference_result"].tolist()
+    assert len(vecs) == 3
+    assert all(len(v) > 0 for v in vecs)
+
+
+# ---- transport retry policy (mocked HTTP, reference TestRequestRetries) ----
+
+
+def _resp(code, body=None):
+    m = MagicMock()
+    m.status_code = code
+    m.json.return_value = body or {}
+    m.text = json.dumps(body or {})
+    return m
+
+
+def _http_client():
+    from sutro_amd.sdk import Sutro
+
+    c = Sutro(api_key="k", base_url="https://example.invalid")
+    return c
+
+
+def test_submission_never_retries_524():
+    c = _http_client()
+    with patch("requests.post", return_value=_resp(524)) as post:
+        resp = c.do_request("POST", "/batch-inference", {"x": 1}, max_retries=0)
+        assert resp.status_code == 524
+        assert post.call_count == 1
+
+
+def test_get_retries_on_524_then_succeeds():
+    c = _http_client()
+    seq = [_resp(524), _resp(524), _resp(200, {"ok": True})]
+    with patch("requests.get", side_effect=seq) as get, \
+            patch("time.sleep") as slept:
+        resp = c.do_request("GET", "/list-jobs")
+        assert resp.status_code == 200
+        assert get.call_count == 3
+        assert slept.call_count == 2
+
+
+def test_non_524_errors_not_retried():
+    c = _http_client()
+    with patch("requests.get", return_value=_resp(500)) as get:
+        resp = c.do_request("GET", "/list-jobs")
+        assert resp.status_code == 500
+        assert get.call_count == 1
