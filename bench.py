@@ -1,0 +1,180 @@
+"""Flagship benchmark: Qwen3-32B `so.infer`-shaped batch inference throughput.
+
+Measures whole-node output tokens/sec of the continuous-batching engine on
+synthetic data with random-init weights (BASELINE.json metric). Data-parallel:
+one rank per GPU over RCCL, each rank runs a full engine replica on its own
+row shard (weak scaling: per-GPU work fixed).
+
+    python bench.py --gpus 1 --steps 32 --warmup 8
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 \
+        --master-addr 127.0.0.1 bench.py --gpus 8 --steps 32 --warmup 8
+
+One engine `step()` = one scheduler iteration (prefill chunks + one decode
+token for every running sequence). The engine is kept saturated by refilling
+finished rows from a synthetic job queue, so the timed region is the
+steady state of a large batch job (prefill+decode mix included).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def build_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--warmup", type=int, default=8)
+    p.add_argument("--model", type=str, default="qwen-3-32b")
+    p.add_argument("--batch", type=int, default=512,
+                   help="rows in flight per GPU (max_num_seqs)")
+    p.add_argument("--prompt-len", type=int, default=128)
+    p.add_argument("--max-new", type=int, default=128)
+    p.add_argument("--tokens-per-step", type=int, default=16384)
+    p.add_argument("--device", type=str, default=None)
+    p.add_argument("--kv-blocks", type=int, default=None)
+    return p.parse_args()
+
+
+def main():
+    args = build_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    have_gpu = torch.cuda.is_available()
+    device = args.device or ("cuda" if have_gpu else "cpu")
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        backend = "nccl" if have_gpu else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import get_model_spec, tiny_spec_for_tests
+
+    if device == "cpu":
+        # no GPU in this container: run the contract end-to-end on a tiny spec
+        spec = tiny_spec_for_tests()
+        args.batch = min(args.batch, 32)
+        args.prompt_len = min(args.prompt_len, 32)
+    else:
+        spec = get_model_spec(args.model)
+
+    cfg = EngineConfig(
+        spec=spec,
+        device=device,
+        max_model_len=max(256, args.prompt_len + args.max_new + 32),
+        max_num_seqs=args.batch,
+        max_tokens_per_step=args.tokens_per_step,
+        num_kv_blocks=args.kv_blocks if device != "cpu" else 512,
+        seed=rank,
+    )
+    t_init0 = time.time()
+    eng = LLMEngine(cfg)
+    init_s = time.time() - t_init0
+
+    # synthetic row stream: prompts of prompt_len byte-tokens, seeded per rank
+    rng = np.random.default_rng(1234 + rank)
+
+    def make_prompt():
+        body = rng.integers(3, 259, size=args.prompt_len - 1).tolist()
+        return [1] + body  # BOS + random bytes
+
+    sp_kwargs = dict(max_tokens=args.max_new, temperature=0.8, top_p=0.95)
+    arrival = [0]
+
+    def refill():
+        while len(eng.scheduler.running) + len(eng.scheduler.waiting_p0) < args.batch:
+            eng.add_request(make_prompt(), SamplingParams(**sp_kwargs),
+                            arrival_idx=arrival[0])
+            arrival[0] += 1
+
+    refill()
+
+    def sync():
+        if have_gpu:
+            torch.cuda.synchronize()
+        if dist is not None:
+            dist.barrier()
+            if have_gpu:
+                torch.cuda.synchronize()
+
+    # warmup: W untimed steps (covers initial prefill wave)
+    for _ in range(args.warmup):
+        refill()
+        eng.step()
+
+    sync()
+    t0 = time.time()
+    out_tokens = 0
+    sched_tokens = 0
+    for _ in range(args.steps):
+        refill()
+        stats = eng.step()
+        out_tokens += stats.output_tokens
+        sched_tokens += stats.scheduled_tokens
+    if have_gpu:
+        torch.cuda.synchronize()
+    t1 = time.time()
+    elapsed = t1 - t0
+    sync()
+
+    if dist is not None:
+        te = torch.tensor([elapsed], dtype=torch.float64,
+                          device=device if have_gpu else "cpu")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed = float(te.item())
+        tt = torch.tensor([out_tokens, sched_tokens], dtype=torch.float64,
+                          device=device if have_gpu else "cpu")
+        dist.all_reduce(tt, op=dist.ReduceOp.SUM)
+        out_tokens, sched_tokens = float(tt[0].item()), float(tt[1].item())
+
+    n_gpus = world if world > 1 else args.gpus
+    value = out_tokens / elapsed if elapsed > 0 else 0.0
+    if rank == 0:
+        result = {
+            "metric": "output_tokens_per_sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000.0 / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if device != "cpu" else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": spec.name,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": args.prompt_len + args.max_new,
+                "parallelism": f"dp{n_gpus}",
+                "prompt_len": args.prompt_len,
+                "max_new_tokens": args.max_new,
+                "rows_per_hour": round(value / max(1, args.max_new) * 3600, 1),
+                "scheduled_tokens_per_sec": round(sched_tokens / elapsed, 1),
+                "engine_init_s": round(init_s, 1),
+            },
+        }
+        print(json.dumps(result))
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

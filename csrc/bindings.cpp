@@ -1,0 +1,144 @@
+// Python bindings for the sutro-amd CDNA4 kernels.
+// Tensors are validated here; raw pointers + the current HIP stream go to the
+// extern "C" launchers in the .hip translation units.
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+#define CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+#define CHECK_BF16(x) \
+  TORCH_CHECK((x).scalar_type() == at::kBFloat16, #x " must be bf16")
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+extern "C" {
+void sutro_rmsnorm(void*, const void*, const void*, float, long, int,
+                   hipStream_t);
+void sutro_fused_add_rmsnorm(void*, void*, const void*, float, long, int,
+                             hipStream_t);
+void sutro_silu_mul(void*, const void*, long, int, hipStream_t);
+void sutro_rope_and_cache(void*, void*, const void*, const long*, const long*,
+                          void*, void*, const float*, int, int, int, int, int,
+                          hipStream_t);
+void sutro_mean_pool_normalize(float*, const void*, const int*, int, int,
+                               hipStream_t);
+void sutro_attn_decode(void*, const void*, const void*, const void*,
+                       const int*, const int*, int, int, int, int, int, float,
+                       hipStream_t);
+void sutro_attn_prefill(void*, const void*, const void*, const void*,
+                        const int*, const int*, const int*, const int*,
+                        const int*, int, int, int, int, float, hipStream_t);
+void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_CUDA(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  const long C = x.size(-1);
+  const long rows = x.numel() / C;
+  TORCH_CHECK(C % 8 == 0 && C <= 16384, "unsupported row size ", C);
+  sutro_rmsnorm(out.data_ptr(), x.data_ptr(), w.data_ptr(), (float)eps, rows,
+                (int)C, cur_stream());
+}
+
+void fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual, torch::Tensor w,
+                       double eps) {
+  CHECK_CUDA(x); CHECK_CONTIG(x); CHECK_BF16(x); CHECK_CONTIG(residual);
+  const long C = x.size(-1);
+  const long rows = x.numel() / C;
+  TORCH_CHECK(C % 8 == 0 && C <= 16384, "unsupported row size ", C);
+  sutro_fused_add_rmsnorm(x.data_ptr(), residual.data_ptr(), w.data_ptr(),
+                          (float)eps, rows, (int)C, cur_stream());
+}
+
+void silu_mul(torch::Tensor out, torch::Tensor x) {
+  CHECK_CUDA(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  const long I = x.size(-1) / 2;
+  const long T = x.numel() / (2 * I);
+  TORCH_CHECK(I % 8 == 0, "intermediate size must be a multiple of 8");
+  sutro_silu_mul(out.data_ptr(), x.data_ptr(), T, (int)I, cur_stream());
+}
+
+void rope_and_cache(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor positions, torch::Tensor slot_mapping,
+                    torch::Tensor k_cache, torch::Tensor v_cache,
+                    torch::Tensor cos_sin) {
+  CHECK_CUDA(q); CHECK_CONTIG(q); CHECK_BF16(q);
+  CHECK_CONTIG(k); CHECK_CONTIG(v); CHECK_CONTIG(k_cache);
+  TORCH_CHECK(cos_sin.scalar_type() == at::kFloat, "cos_sin must be f32");
+  const int T = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hk = k.size(1);
+  const int bs = k_cache.size(2);
+  TORCH_CHECK(D % 2 == 0 && D / 2 <= 128, "unsupported head_dim ", D);
+  sutro_rope_and_cache(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                       positions.data_ptr<long>(), slot_mapping.data_ptr<long>(),
+                       k_cache.data_ptr(), v_cache.data_ptr(),
+                       cos_sin.data_ptr<float>(), T, Hq, Hk, D, bs,
+                       cur_stream());
+}
+
+void mean_pool_normalize(torch::Tensor out, torch::Tensor hidden,
+                         torch::Tensor qlocs) {
+  CHECK_CUDA(hidden); CHECK_CONTIG(hidden); CHECK_BF16(hidden);
+  const int S = out.size(0), H = hidden.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden size must be a multiple of 8");
+  sutro_mean_pool_normalize(out.data_ptr<float>(), hidden.data_ptr(),
+                            qlocs.data_ptr<int>(), S, H, cur_stream());
+}
+
+void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
+                     torch::Tensor v_cache, torch::Tensor block_tables,
+                     torch::Tensor seq_lens, torch::Tensor qlocs, double scale,
+                     int64_t num_decodes, torch::Tensor tile_seq,
+                     torch::Tensor tile_q0, int64_t prefill_token_count) {
+  CHECK_CUDA(q); CHECK_CONTIG(q); CHECK_BF16(q); CHECK_CONTIG(k_cache);
+  const int Hq = q.size(1), D = q.size(2);
+  const int Hk = k_cache.size(1);
+  const int bs = k_cache.size(2);
+  const int S = seq_lens.size(0);
+  TORCH_CHECK(D == 128, "attention kernels support head_dim 128 (got ", D, ")");
+  TORCH_CHECK(bs == 32, "attention kernels require kv_block_size 32");
+  TORCH_CHECK(Hq % Hk == 0 && Hq / Hk <= 8,
+              "GQA group size must divide and be <= 8");
+  const int bt_stride = block_tables.size(1);
+  const int n_tiles = tile_seq.numel();
+  if (n_tiles > 0) {
+    sutro_attn_prefill(out.data_ptr(), q.data_ptr(), k_cache.data_ptr(),
+                       v_cache.data_ptr(), block_tables.data_ptr<int>(),
+                       seq_lens.data_ptr<int>(), qlocs.data_ptr<int>(),
+                       tile_seq.data_ptr<int>(), tile_q0.data_ptr<int>(),
+                       n_tiles, bt_stride, Hq, Hk, (float)scale, cur_stream());
+  }
+  if (num_decodes > 0) {
+    const long dec_off = prefill_token_count;  // decode rows are the tail
+    const int seq_offset = S - (int)num_decodes;
+    const u_int16_t* qp = (const u_int16_t*)q.data_ptr();
+    u_int16_t* op = (u_int16_t*)out.data_ptr();
+    sutro_attn_decode(op + dec_off * Hq * D, qp + dec_off * Hq * D,
+                      k_cache.data_ptr(), v_cache.data_ptr(),
+                      block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                      bt_stride, (int)num_decodes, Hq, Hk, seq_offset,
+                      (float)scale, cur_stream());
+  }
+}
+
+torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
+  CHECK_CUDA(a); CHECK_BF16(a);
+  auto c = torch::zeros({32, 32}, a.options().dtype(at::kFloat));
+  sutro_mfma32_probe(c.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                     cur_stream());
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual+=x; x=rmsnorm");
+  m.def("silu_mul", &silu_mul, "fused SwiGLU");
+  m.def("rope_and_cache", &rope_and_cache, "RoPE + paged KV write");
+  m.def("mean_pool_normalize", &mean_pool_normalize, "varlen mean pool + L2");
+  m.def("paged_attention", &paged_attention, "paged prefill+decode attention");
+  m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
+}

@@ -41,3 +41,7 @@ class ForwardBatch:
     logits_idx: torch.Tensor       # [n] long - flat rows needing logits
     max_seq_len: int               # host-side max of seq_lens
     max_query_len: int             # host-side max new tokens per seq
+    # prefill q-tile map for the GPU flash kernel (32 rows per tile)
+    tile_seq: Optional[torch.Tensor] = None   # [n_tiles] int32 seq index
+    tile_q0: Optional[torch.Tensor] = None    # [n_tiles] int32 local row start
+    prefill_token_count: int = 0              # flat rows belonging to prefills

@@ -42,7 +42,8 @@ class LLMEngine:
         self.device = cfg.device
         self.tokenizer = get_tokenizer()
         if model is None:
-            model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len)
+            with torch.device(cfg.device):
+                model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len)
             model.init_random_weights(cfg.seed)
         self.model = model.to(cfg.device).eval()
 
@@ -170,6 +171,16 @@ class LLMEngine:
         for s in range(sb.num_prefills, S):
             logits_rows.append(int(qlocs[s + 1]) - 1)
 
+        # prefill q-tile map (32 rows per tile) for the GPU flash kernel
+        tiles_s: List[int] = []
+        tiles_q0: List[int] = []
+        for s in range(sb.num_prefills):
+            c = sb.num_new_tokens[s]
+            for j in range(0, c, 32):
+                tiles_s.append(s)
+                tiles_q0.append(j)
+        prefill_token_count = int(qlocs[sb.num_prefills])
+
         dev = self.device
         return ForwardBatch(
             input_ids=torch.from_numpy(input_ids).to(dev),
@@ -182,6 +193,9 @@ class LLMEngine:
             logits_idx=torch.tensor(logits_rows, dtype=torch.long, device=dev),
             max_seq_len=int(seq_lens.max()) if S else 0,
             max_query_len=int(max(sb.num_new_tokens)) if S else 0,
+            tile_seq=torch.tensor(tiles_s, dtype=torch.int32, device=dev),
+            tile_q0=torch.tensor(tiles_q0, dtype=torch.int32, device=dev),
+            prefill_token_count=prefill_token_count,
         )
 
     def _sampling_reqs(self, sb: ScheduledBatch) -> List[Request]:

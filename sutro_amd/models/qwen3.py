@@ -73,7 +73,8 @@ class Qwen3Attention(nn.Module):
         o = ops.paged_attention(q, kv.k_cache[self.layer_idx],
                                 kv.v_cache[self.layer_idx], fb.block_tables,
                                 fb.seq_lens, fb.query_start_locs, self.scale,
-                                fb.num_decodes_tail)
+                                fb.num_decodes_tail, fb.tile_seq, fb.tile_q0,
+                                fb.prefill_token_count)
         return self.o_proj(o.view(T, self.q_size))
 
 
@@ -193,13 +194,20 @@ class Qwen3Model(nn.Module):
 
     @torch.no_grad()
     def init_random_weights(self, seed: int = 0) -> None:
-        """Deterministic random init (there is no checkpoint source offline)."""
-        gen = torch.Generator().manual_seed(seed)
+        """Deterministic random init, generated on the parameters' own device
+        (there is no checkpoint source offline; a 32B model must not round-trip
+        through host RAM)."""
+        import zlib
+
         for name, p in sorted(self.named_parameters()):
             if p.dim() >= 2:
                 fan_in = p.shape[-1] if "embed" not in name else p.shape[0]
                 std = 0.02 if "embed" in name else (1.0 / math.sqrt(fan_in))
-                with torch.no_grad():
-                    cpu = torch.randn(p.shape, generator=gen, dtype=torch.float32) * std
-                    p.copy_(cpu.to(p.dtype))
-            # norm weights stay at ones
+                dev = p.device
+                gen = torch.Generator(device=dev)
+                gen.manual_seed(seed * 1000003 + zlib.crc32(name.encode()))
+                r = torch.randn(p.shape, generator=gen, dtype=torch.float32,
+                                device=dev)
+                p.copy_((r * std).to(p.dtype))
+            else:
+                p.fill_(1.0)  # norm weights

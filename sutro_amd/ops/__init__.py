@@ -87,21 +87,32 @@ def rope_and_cache(
     return q
 
 
+_EMPTY_I32: Optional[torch.Tensor] = None
+
+
 def paged_attention(
     q: torch.Tensor, k_cache: torch.Tensor, v_cache: torch.Tensor,
     block_tables: torch.Tensor, seq_lens: torch.Tensor,
     query_start_locs: torch.Tensor, scale: float,
     num_decodes_tail: int = 0,
+    tile_seq: Optional[torch.Tensor] = None,
+    tile_q0: Optional[torch.Tensor] = None,
+    prefill_token_count: int = 0,
 ) -> torch.Tensor:
     """Attention of new tokens vs full cached KV (causal within the new chunk).
 
-    `num_decodes_tail`: how many of the trailing sequences are single-token
-    decodes (lets the GPU path route them to the decode kernel)."""
+    `num_decodes_tail`: trailing sequences that are single-token decodes (the
+    GPU path routes them to the decode kernel). `tile_seq`/`tile_q0`: int32
+    device arrays mapping prefill q-tiles of 32 rows to (seq, local row)."""
     if _use_hip(q):
         out = torch.empty_like(q)
+        if tile_seq is None:
+            tile_seq = torch.empty(0, dtype=torch.int32, device=q.device)
+            tile_q0 = tile_seq
         _require_hip().paged_attention(out, q, k_cache, v_cache, block_tables,
                                        seq_lens, query_start_locs, scale,
-                                       num_decodes_tail)
+                                       num_decodes_tail, tile_seq, tile_q0,
+                                       prefill_token_count)
         return out
     return torch_ref.paged_attention(q, k_cache, v_cache, block_tables,
                                      seq_lens, query_start_locs, scale)

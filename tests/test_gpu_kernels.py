@@ -1,0 +1,270 @@
+"""GPU numerics tests: every CDNA4 HIP kernel vs the plain-PyTorch fp32
+reference (sutro_amd.ops.torch_ref) on the same bf16 inputs."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from sutro_amd import _C, ops
+    from sutro_amd.ops import torch_ref as R
+else:  # collected on CPU; every test is skipped by the marker filter anyway
+    _C = ops = R = None
+
+
+def require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+
+
+DEV = "cuda"
+
+
+def bf(x):
+    return x.to(torch.bfloat16).to(DEV)
+
+
+def test_hip_extension_loaded():
+    require_gpu()
+    assert ops.hip_available(), "sutro_amd._C must be importable on the GPU box"
+
+
+def test_mfma32_layout_probe():
+    require_gpu()
+    torch.manual_seed(0)
+    # asymmetric B catches transposed layouts (guide: A=I-check rule)
+    a = torch.randn(32, 16)
+    b = torch.randn(16, 32) * torch.arange(1, 33).float() / 8.0
+    c = _C.mfma32_probe(bf(a), bf(b))
+    ref = (bf(a).float() @ bf(b).float())
+    assert torch.allclose(c.cpu(), ref.cpu(), atol=2e-2, rtol=2e-2), (
+        (c.cpu() - ref.cpu()).abs().max()
+    )
+
+
+@pytest.mark.parametrize("rows,cols", [(64, 5120), (3, 1024), (256, 4096)])
+def test_rmsnorm(rows, cols):
+    require_gpu()
+    torch.manual_seed(1)
+    x = bf(torch.randn(rows, cols))
+    w = bf(torch.randn(cols) * 0.5 + 1.0)
+    got = ops.rmsnorm(x, w, 1e-6).float().cpu()
+    ref = R.rmsnorm(x.cpu(), w.cpu(), 1e-6).float()
+    assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_rmsnorm_small_rows():
+    require_gpu()
+    x = bf(torch.randn(100, 4, 128))  # qk-norm shape [T, H, D]
+    w = bf(torch.ones(128) * 1.3)
+    got = ops.rmsnorm(x, w, 1e-6).float().cpu()
+    ref = R.rmsnorm(x.cpu(), w.cpu(), 1e-6).float()
+    assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_fused_add_rmsnorm():
+    require_gpu()
+    torch.manual_seed(2)
+    x = bf(torch.randn(64, 2048))
+    res = bf(torch.randn(64, 2048))
+    x_cpu, res_cpu = x.cpu(), res.cpu()
+    w = bf(torch.randn(2048) * 0.1 + 1.0)
+    y, new_res = ops.fused_add_rmsnorm(x, res, w, 1e-6)
+    ref_y, ref_res = R.fused_add_rmsnorm(x_cpu, res_cpu, w.cpu(), 1e-6)
+    assert torch.allclose(new_res.float().cpu(), ref_res.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(y.float().cpu(), ref_y.float(), atol=3e-2, rtol=3e-2)
+
+
+def test_silu_mul():
+    require_gpu()
+    x = bf(torch.randn(128, 2 * 1024))
+    got = ops.silu_mul(x).float().cpu()
+    ref = R.silu_mul(x.cpu()).float()
+    assert torch.allclose(got, ref, atol=2e-2, rtol=2e-2)
+
+
+def _mk_cache(nb, hk, bs, d):
+    return (torch.zeros(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV),
+            torch.zeros(nb, hk, bs, d, dtype=torch.bfloat16, device=DEV))
+
+
+def test_rope_and_cache():
+    require_gpu()
+    torch.manual_seed(3)
+    T, Hq, Hk, D, bs = 33, 8, 2, 128, 32
+    q = bf(torch.randn(T, Hq, D))
+    k = bf(torch.randn(T, Hk, D))
+    v = bf(torch.randn(T, Hk, D))
+    q0, k0, v0 = q.cpu(), k.cpu(), v.cpu()
+    pos = torch.randint(0, 500, (T,), dtype=torch.long)
+    slots = torch.randperm(4 * bs)[:T].to(torch.long)
+    cs = R.rope_cos_sin(512, D, 1e6).to(DEV)
+    kc, vc = _mk_cache(4, Hk, bs, D)
+    q_out = ops.rope_and_cache(q, k, v, pos.to(DEV), slots.to(DEV), kc, vc, cs)
+    rq, rk = R.apply_rope(q0.float(), k0.float(), pos, cs.cpu())
+    kc_ref = torch.zeros(4, Hk, bs, D)
+    vc_ref = torch.zeros(4, Hk, bs, D)
+    R.write_kv_cache(rk, v0.float(), kc_ref, vc_ref, slots)
+    assert torch.allclose(q_out.float().cpu(), rq.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(kc.float().cpu(), kc_ref.to(torch.bfloat16).float(),
+                          atol=3e-2, rtol=3e-2)
+    assert torch.allclose(vc.float().cpu(), vc_ref.to(torch.bfloat16).float(),
+                          atol=3e-2, rtol=3e-2)
+
+
+def _attn_setup(seq_lens, new_counts, Hq=8, Hk=2, D=128, bs=32, seed=5):
+    """Build a scattered paged cache + q batch; returns gpu tensors + cpu refs."""
+    torch.manual_seed(seed)
+    S = len(seq_lens)
+    total_blocks = sum((L + bs - 1) // bs for L in seq_lens) + 3
+    perm = torch.randperm(total_blocks).tolist()
+    kc, vc = _mk_cache(total_blocks, Hk, bs, D)
+    tables = []
+    i = 0
+    for L in seq_lens:
+        n = (L + bs - 1) // bs
+        tables.append(perm[i:i + n])
+        i += n
+    max_b = max(len(t) for t in tables)
+    bt = torch.zeros(S, max_b, dtype=torch.int32)
+    for s, t in enumerate(tables):
+        bt[s, :len(t)] = torch.tensor(t, dtype=torch.int32)
+    qs = []
+    for s, L in enumerate(seq_lens):
+        K = bf(torch.randn(L, Hk, D))
+        V = bf(torch.randn(L, Hk, D))
+        slots = torch.tensor(
+            [tables[s][p // bs] * bs + p % bs for p in range(L)], dtype=torch.long)
+        R.write_kv_cache(K, V, kc, vc, slots.to(DEV))
+        qs.append(bf(torch.randn(new_counts[s], Hq, D)))
+    q = torch.cat(qs, 0)
+    qlocs = torch.tensor([0] + list(torch.tensor(new_counts).cumsum(0)),
+                         dtype=torch.int32)
+    sl = torch.tensor(seq_lens, dtype=torch.int32)
+    return q, kc, vc, bt.to(DEV), sl.to(DEV), qlocs.to(DEV), sl, qlocs, bt
+
+
+def test_attn_decode_matches_ref():
+    require_gpu()
+    seq_lens = [1, 31, 32, 33, 100, 257]
+    new_counts = [1] * len(seq_lens)
+    q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(seq_lens, new_counts)
+    scale = 1.0 / math.sqrt(128)
+    got = ops.paged_attention(q, kc, vc, bt, sl, ql, scale,
+                              num_decodes_tail=len(seq_lens),
+                              prefill_token_count=0)
+    ref = R.paged_attention(q.float().cpu(), kc.float().cpu(), vc.float().cpu(),
+                            bt_c, sl_c, ql_c, scale)
+    assert torch.allclose(got.float().cpu(), ref, atol=3e-2, rtol=3e-2), (
+        (got.float().cpu() - ref).abs().max()
+    )
+
+
+def test_attn_prefill_matches_ref():
+    require_gpu()
+    # mixed: fresh prefill (ctx=0), chunked continuation (ctx>0), ragged sizes
+    seq_lens = [40, 64, 100, 7]
+    new_counts = [40, 32, 33, 7]
+    q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(seq_lens, new_counts)
+    tiles_s, tiles_q0 = [], []
+    for s, c in enumerate(new_counts):
+        for j in range(0, c, 32):
+            tiles_s.append(s)
+            tiles_q0.append(j)
+    scale = 1.0 / math.sqrt(128)
+    got = ops.paged_attention(
+        q, kc, vc, bt, sl, ql, scale, num_decodes_tail=0,
+        tile_seq=torch.tensor(tiles_s, dtype=torch.int32, device=DEV),
+        tile_q0=torch.tensor(tiles_q0, dtype=torch.int32, device=DEV),
+        prefill_token_count=int(sum(new_counts)))
+    ref = R.paged_attention(q.float().cpu(), kc.float().cpu(), vc.float().cpu(),
+                            bt_c, sl_c, ql_c, scale)
+    err = (got.float().cpu() - ref).abs().max()
+    assert torch.allclose(got.float().cpu(), ref, atol=5e-2, rtol=5e-2), err
+
+
+def test_attn_mixed_batch():
+    require_gpu()
+    seq_lens = [50, 90, 33, 65, 129]
+    new_counts = [50, 40, 1, 1, 1]  # 2 prefills + 3 decodes
+    q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(seq_lens, new_counts)
+    tiles_s, tiles_q0 = [], []
+    for s in range(2):
+        for j in range(0, new_counts[s], 32):
+            tiles_s.append(s)
+            tiles_q0.append(j)
+    scale = 1.0 / math.sqrt(128)
+    got = ops.paged_attention(
+        q, kc, vc, bt, sl, ql, scale, num_decodes_tail=3,
+        tile_seq=torch.tensor(tiles_s, dtype=torch.int32, device=DEV),
+        tile_q0=torch.tensor(tiles_q0, dtype=torch.int32, device=DEV),
+        prefill_token_count=90)
+    ref = R.paged_attention(q.float().cpu(), kc.float().cpu(), vc.float().cpu(),
+                            bt_c, sl_c, ql_c, scale)
+    assert torch.allclose(got.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gqa_group_sizes():
+    require_gpu()
+    for hq, hk in ((8, 8), (8, 4), (16, 2)):
+        seq_lens = [70, 33]
+        q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(
+            seq_lens, [1, 1], Hq=hq, Hk=hk)
+        scale = 1.0 / math.sqrt(128)
+        got = ops.paged_attention(q, kc, vc, bt, sl, ql, scale,
+                                  num_decodes_tail=2, prefill_token_count=0)
+        ref = R.paged_attention(q.float().cpu(), kc.float().cpu(),
+                                vc.float().cpu(), bt_c, sl_c, ql_c, scale)
+        assert torch.allclose(got.float().cpu(), ref, atol=3e-2, rtol=3e-2), (hq, hk)
+
+
+def test_mean_pool_normalize():
+    require_gpu()
+    h = bf(torch.randn(50, 1024))
+    locs = torch.tensor([0, 10, 11, 50], dtype=torch.int32, device=DEV)
+    got = ops.mean_pool_normalize(h, locs).cpu()
+    ref = R.mean_pool_normalize(h.float().cpu(), locs.cpu())
+    assert torch.allclose(got, ref, atol=2e-2, rtol=2e-2)
+
+
+def test_engine_e2e_gpu():
+    require_gpu()
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import get_model_spec
+
+    cfg = EngineConfig(spec=get_model_spec("qwen-3-0.6b"), device="cuda",
+                       max_model_len=512, num_kv_blocks=512,
+                       max_tokens_per_step=2048, max_num_seqs=64)
+    eng = LLMEngine(cfg)
+    outs = eng.generate([f"row {i}" for i in range(8)],
+                        sampling=SamplingParams(max_tokens=16, temperature=0.8))
+    assert len(outs) == 8
+    assert eng.total_output_tokens >= 8
+
+
+def test_engine_guided_json_gpu():
+    require_gpu()
+    import json
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import get_model_spec
+
+    cfg = EngineConfig(spec=get_model_spec("qwen-3-0.6b"), device="cuda",
+                       max_model_len=2048, num_kv_blocks=512,
+                       max_tokens_per_step=2048)
+    eng = LLMEngine(cfg)
+    schema = {"type": "object", "properties": {
+        "label": {"enum": ["A", "B"]},
+        "score": {"type": "integer", "minimum": 0, "maximum": 9}}}
+    outs = eng.generate(["classify"],
+                        sampling=SamplingParams(max_tokens=512, temperature=1.0),
+                        schema=schema)
+    d = json.loads(outs[0])
+    assert d["label"] in ("A", "B") and 0 <= d["score"] <= 9
