@@ -72,8 +72,17 @@ class LLMEngine:
             dtype=cfg.dtype,
             device=cfg.device,
         )
+        # block 0 is reserved scratch: hipGraph padding rows read/write it
+        self.scratch_block = self.kv.allocator.allocate(1)[0]
         self.scheduler = Scheduler(cfg, self.kv)
         self.sampler = Sampler(cfg.device, seed=cfg.seed, vocab_limit=TOKENIZER_VOCAB)
+
+        self.graph_runner = None
+        if (cfg.device.startswith("cuda") and not cfg.enforce_eager
+                and not self.spec.embedding):
+            from .graph_runner import DecodeGraphRunner
+
+            self.graph_runner = DecodeGraphRunner(self)
 
         self._next_req_id = 0
         self._fsms: Dict[int, GuidedFSM] = {}
@@ -213,31 +222,48 @@ class LLMEngine:
         stats = StepStats()
         if not sb.reqs:
             return stats
-        fb = self._build_forward_batch(sb)
-        hidden = self.model(fb, self.kv)
-
         stats.scheduled_tokens = sb.total_tokens
 
-        if self.spec.embedding:
-            self._embedding_update(sb, fb, hidden, stats)
-            self._advance_computed(sb)
-            stats.prefill_tokens = sb.total_tokens
-            self.total_prompt_tokens += stats.prefill_tokens
-            return stats
-
-        sample_reqs = self._sampling_reqs(sb)
-        if sample_reqs:
-            logits = self.model.compute_logits(hidden[fb.logits_idx])
-            fsm_mask = self._fsm_masks(sample_reqs, logits.device)
-            tokens, lps = self.sampler.sample(logits, sample_reqs, fsm_mask)
+        # Split mixed steps: decodes replay a hipGraph, prefills run eager.
+        sub_batches: List[ScheduledBatch] = []
+        if (self.graph_runner is not None and sb.num_decodes > 0):
+            if sb.num_prefills > 0:
+                sub_batches.append(ScheduledBatch(
+                    reqs=sb.reqs[: sb.num_prefills],
+                    num_new_tokens=sb.num_new_tokens[: sb.num_prefills],
+                    num_prefills=sb.num_prefills))
+            sub_batches.append(ScheduledBatch(
+                reqs=sb.reqs[sb.num_prefills:],
+                num_new_tokens=[1] * sb.num_decodes,
+                num_prefills=0))
         else:
-            tokens, lps = [], []
+            sub_batches.append(sb)
 
-        self._advance_computed(sb)
+        for sub in sub_batches:
+            if (self.graph_runner is not None and self.graph_runner.can_run(sub)):
+                logits = self.graph_runner.run(sub)
+                sample_reqs = list(sub.reqs)
+            else:
+                fb = self._build_forward_batch(sub)
+                hidden = self.model(fb, self.kv)
+                if self.spec.embedding:
+                    self._embedding_update(sub, fb, hidden, stats)
+                    self._advance_computed(sub)
+                    stats.prefill_tokens += sub.total_tokens
+                    continue
+                sample_reqs = self._sampling_reqs(sub)
+                logits = (self.model.compute_logits(hidden[fb.logits_idx])
+                          if sample_reqs else None)
+            if sample_reqs:
+                fsm_mask = self._fsm_masks(sample_reqs, logits.device)
+                tokens, lps = self.sampler.sample(logits, sample_reqs, fsm_mask)
+            else:
+                tokens, lps = [], []
+            self._advance_computed(sub)
+            for req, tok, lp in zip(sample_reqs, tokens, lps):
+                self._apply_sampled(req, int(tok), float(lp), stats)
+            stats.prefill_tokens += sum(sub.num_new_tokens[: sub.num_prefills])
 
-        for req, tok, lp in zip(sample_reqs, tokens, lps):
-            self._apply_sampled(req, int(tok), float(lp), stats)
-        stats.prefill_tokens = sum(sb.num_new_tokens[: sb.num_prefills])
         self.total_prompt_tokens += stats.prefill_tokens
         self.total_output_tokens += stats.output_tokens
         return stats

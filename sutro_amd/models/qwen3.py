@@ -58,15 +58,12 @@ class Qwen3Attention(nn.Module):
         T = x.shape[0]
         qkv = self.qkv_proj(x)
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
-        q = q.view(T, self.num_heads, self.head_dim)
-        k = k.view(T, self.num_kv_heads, self.head_dim)
-        v = v.view(T, self.num_kv_heads, self.head_dim)
+        q = q.reshape(T, self.num_heads, self.head_dim).contiguous()
+        k = k.reshape(T, self.num_kv_heads, self.head_dim).contiguous()
+        v = v.reshape(T, self.num_kv_heads, self.head_dim).contiguous()
         if self.qk_norm:
             q = self.q_norm(q)
             k = self.k_norm(k)
-        q = q.contiguous()
-        k = k.contiguous()
-        v = v.contiguous()
         q = ops.rope_and_cache(q, k, v, fb.positions, fb.slot_mapping,
                                kv.k_cache[self.layer_idx], kv.v_cache[self.layer_idx],
                                cos_sin)
