@@ -39,6 +39,8 @@ def build_args():
     p.add_argument("--tokens-per-step", type=int, default=16384)
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
+    p.add_argument("--no-refill", action="store_true",
+                   help="measure pure decode of the initial batch (no new rows)")
     return p.parse_args()
 
 
@@ -98,6 +100,8 @@ def main():
     arrival = [0]
 
     def refill():
+        if args.no_refill and arrival[0] >= args.batch:
+            return
         while len(eng.scheduler.running) + len(eng.scheduler.waiting_p0) < args.batch:
             eng.add_request(make_prompt(), SamplingParams(**sp_kwargs),
                             arrival_idx=arrival[0])

@@ -33,6 +33,9 @@ void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         const int*, const int*, const int*, const int*,
                         const int*, int, int, int, int, float, hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
+void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
+                    const float*, const void*, const void*, float, int, int,
+                    int, int, int, int, hipStream_t);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -125,6 +128,27 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   }
 }
 
+void qkv_prep(torch::Tensor qkv, torch::Tensor q_out, torch::Tensor k_cache,
+              torch::Tensor v_cache, torch::Tensor positions,
+              torch::Tensor slot_mapping, torch::Tensor cos_sin,
+              c10::optional<torch::Tensor> q_norm_w,
+              c10::optional<torch::Tensor> k_norm_w, double eps) {
+  CHECK_CUDA(qkv); CHECK_CONTIG(qkv); CHECK_BF16(qkv);
+  CHECK_CONTIG(q_out); CHECK_CONTIG(k_cache); CHECK_CONTIG(v_cache);
+  const int T = qkv.size(0);
+  const int Hq = q_out.size(1), D = q_out.size(2);
+  const int Hk = k_cache.size(1);
+  const int bs = k_cache.size(2);
+  const int row_stride = qkv.size(1);
+  TORCH_CHECK(row_stride == Hq * D + 2 * Hk * D, "qkv width mismatch");
+  const void* qw = q_norm_w ? q_norm_w->data_ptr() : nullptr;
+  const void* kw = k_norm_w ? k_norm_w->data_ptr() : nullptr;
+  sutro_qkv_prep(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
+                 v_cache.data_ptr(), positions.data_ptr<long>(),
+                 slot_mapping.data_ptr<long>(), cos_sin.data_ptr<float>(), qw,
+                 kw, (float)eps, T, Hq, Hk, D, bs, row_stride, cur_stream());
+}
+
 torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
   CHECK_CUDA(a); CHECK_BF16(a);
   auto c = torch::zeros({32, 32}, a.options().dtype(at::kFloat));
@@ -138,6 +162,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual+=x; x=rmsnorm");
   m.def("silu_mul", &silu_mul, "fused SwiGLU");
   m.def("rope_and_cache", &rope_and_cache, "RoPE + paged KV write");
+  m.def("qkv_prep", &qkv_prep, "fused qk-norm + RoPE + KV write + q gather");
   m.def("mean_pool_normalize", &mean_pool_normalize, "varlen mean pool + L2");
   m.def("paged_attention", &paged_attention, "paged prefill+decode attention");
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");

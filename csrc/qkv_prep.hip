@@ -1,0 +1,102 @@
+// Fused post-QKV-GEMM prep: per-head RMSNorm (Qwen3 qk-norm) + RoPE (NeoX)
+// + paged KV-cache scatter + contiguous-q gather, reading the qkv projection
+// output in place (no .contiguous() copies, no separate norm/rope launches).
+//
+// qkv: [T, Hq*D + 2*Hk*D] (the GEMM output, row stride = that width)
+//   q heads  -> normed+rotated -> q_out [T, Hq, D]
+//   k heads  -> normed+rotated -> k_cache slot
+//   v heads  -> copied         -> v_cache slot
+// One wave per (token, head) item over Hq+2*Hk heads.
+#include "common.h"
+
+__global__ void qkv_prep_kernel(
+    const u16* __restrict__ qkv,     // [T, row_stride]
+    u16* __restrict__ q_out,         // [T, Hq, D]
+    u16* __restrict__ k_cache,       // [nb, Hk, bs, D]
+    u16* __restrict__ v_cache,
+    const long* __restrict__ pos,    // [T]
+    const long* __restrict__ slots,  // [T]
+    const float* __restrict__ cos_sin,  // [max_pos, D]
+    const u16* __restrict__ qw,      // [D] or null
+    const u16* __restrict__ kw,      // [D] or null
+    float eps, int T, int Hq, int Hk, int D, int bs, int row_stride) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int H = Hq + 2 * Hk;
+  const long item = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  if (item >= (long)T * H) return;
+  const int t = (int)(item / H);
+  const int h = (int)(item - (long)t * H);
+  const int half = D / 2;
+  const int pairs = half / (int)WAVE ? half / (int)WAVE : 1;  // D>=128 -> >=1
+
+  const long slot = slots[t];
+  const long blk = slot / bs, off = slot - blk * bs;
+
+  if (h >= Hq + Hk) {
+    // V head: plain copy into the paged cache
+    const int vh = h - Hq - Hk;
+    const u16* src = qkv + (long)t * row_stride + Hq * D + Hk * D + vh * D;
+    u16* dst = v_cache + (((long)blk * Hk + vh) * bs + off) * D;
+    for (int j = lane; j < D / 8; j += (int)WAVE)
+      *(u16x8*)(dst + j * 8) = *(const u16x8*)(src + j * 8);
+    return;
+  }
+
+  const bool is_q = h < Hq;
+  const u16* src = is_q ? qkv + (long)t * row_stride + h * D
+                        : qkv + (long)t * row_stride + Hq * D + (h - Hq) * D;
+  u16* dst = is_q ? q_out + ((long)t * Hq + h) * D
+                  : k_cache + (((long)blk * Hk + (h - Hq)) * bs + off) * D;
+  const u16* w = is_q ? qw : kw;
+
+  // load the row: lane holds pairs (d, d+half) for d = lane + j*WAVE
+  float x1[2], x2[2];
+  float ssq = 0.f;
+#pragma unroll 2
+  for (int j = 0; j < 2; ++j) {
+    if (j >= pairs) break;
+    const int d = lane + j * (int)WAVE;
+    if (d >= half) continue;  // D < 128: upper lanes idle
+    x1[j] = bf2f(src[d]);
+    x2[j] = bf2f(src[d + half]);
+    ssq += x1[j] * x1[j] + x2[j] * x2[j];
+  }
+  if (w != nullptr) {
+    ssq = wave_sum_f32(ssq);
+    const float inv = rsqrtf(ssq / (float)D + eps);
+#pragma unroll 2
+    for (int j = 0; j < 2; ++j) {
+      if (j >= pairs) break;
+      const int d = lane + j * (int)WAVE;
+      if (d >= half) continue;
+      x1[j] *= inv * bf2f(w[d]);
+      x2[j] *= inv * bf2f(w[d + half]);
+    }
+  }
+  const float* cs = cos_sin + pos[t] * D;
+#pragma unroll 2
+  for (int j = 0; j < 2; ++j) {
+    if (j >= pairs) break;
+    const int d = lane + j * (int)WAVE;
+    if (d >= half) continue;
+    const float c = cs[d], sn = cs[d + half];
+    dst[d] = f2bf(x1[j] * c - x2[j] * sn);
+    dst[d + half] = f2bf(x2[j] * c + x1[j] * sn);
+  }
+}
+
+extern "C" void sutro_qkv_prep(const void* qkv, void* q_out, void* k_cache,
+                               void* v_cache, const long* pos,
+                               const long* slots, const float* cos_sin,
+                               const void* qw, const void* kw, float eps, int T,
+                               int Hq, int Hk, int D, int bs, int row_stride,
+                               hipStream_t s) {
+  const long items = (long)T * (Hq + 2 * Hk);
+  if (items == 0) return;
+  const int wpb = 4;
+  const long blocks = (items + wpb - 1) / wpb;
+  hipLaunchKernelGGL(qkv_prep_kernel, dim3((unsigned)blocks), dim3(wpb * WAVE),
+                     0, s, (const u16*)qkv, (u16*)q_out, (u16*)k_cache,
+                     (u16*)v_cache, pos, slots, cos_sin, (const u16*)qw,
+                     (const u16*)kw, eps, T, Hq, Hk, D, bs, row_stride);
+}

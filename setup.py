@@ -15,6 +15,7 @@ SOURCES = [
     "csrc/rmsnorm.hip",
     "csrc/elementwise.hip",
     "csrc/rope_cache.hip",
+    "csrc/qkv_prep.hip",
     "csrc/pool.hip",
     "csrc/attn_decode.hip",
     "csrc/attn_prefill.hip",

@@ -21,6 +21,10 @@ class EngineConfig:
     max_model_len: int = 8192        # scheduler cap on prompt+output length
     max_num_seqs: int = 1024         # max concurrently running sequences
     max_tokens_per_step: int = 16384  # token budget per scheduler step (prefill chunking)
+    # throughput policy: while decodes are running, hold back new prefills
+    # until this many prompt tokens have accumulated (amortizes the eager
+    # prefill pass; hipGraph decode steps stay pure). 0 = admit eagerly.
+    min_prefill_batch_tokens: int = 4096
     gpu_memory_utilization: float = 0.90
     num_kv_blocks: Optional[int] = None  # None = derive from free memory
     default_max_new_tokens: int = 256

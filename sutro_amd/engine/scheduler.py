@@ -120,7 +120,23 @@ class Scheduler:
             scheduled.add(id(req))
             budget -= c
 
-        # 3) admit new requests (p0 ahead of p1)
+        # 3) admit new requests (p0 ahead of p1). While decodes run, hold
+        # back until enough prefill work accumulates to amortize the pass.
+        thr = self.cfg.min_prefill_batch_tokens
+        if decode_reqs and not prefill_reqs and thr > 0:
+            avail = 0
+            for q in (self.waiting_p0, self.waiting_p1):
+                for r in q:
+                    avail += r.num_prompt_tokens - r.num_computed_tokens
+                    if avail >= thr:
+                        break
+                if avail >= thr:
+                    break
+            if avail < min(thr, self.cfg.max_tokens_per_step):
+                return ScheduledBatch(reqs=prefill_reqs + decode_reqs,
+                                      num_new_tokens=prefill_counts
+                                      + [1] * len(decode_reqs),
+                                      num_prefills=len(prefill_reqs))
         for q in (self.waiting_p0, self.waiting_p1):
             while q and budget > 0 and len(self.running) < self.cfg.max_num_seqs:
                 req = q[0]

@@ -57,16 +57,14 @@ class Qwen3Attention(nn.Module):
                 cos_sin: torch.Tensor) -> torch.Tensor:
         T = x.shape[0]
         qkv = self.qkv_proj(x)
-        q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size], dim=-1)
-        q = q.reshape(T, self.num_heads, self.head_dim).contiguous()
-        k = k.reshape(T, self.num_kv_heads, self.head_dim).contiguous()
-        v = v.reshape(T, self.num_kv_heads, self.head_dim).contiguous()
-        if self.qk_norm:
-            q = self.q_norm(q)
-            k = self.k_norm(k)
-        q = ops.rope_and_cache(q, k, v, fb.positions, fb.slot_mapping,
-                               kv.k_cache[self.layer_idx], kv.v_cache[self.layer_idx],
-                               cos_sin)
+        q = ops.fused_qkv_prep(
+            qkv, self.num_heads, self.num_kv_heads, self.head_dim,
+            fb.positions, fb.slot_mapping,
+            kv.k_cache[self.layer_idx], kv.v_cache[self.layer_idx], cos_sin,
+            self.q_norm.weight if self.qk_norm else None,
+            self.k_norm.weight if self.qk_norm else None,
+            self.q_norm.eps if self.qk_norm else 1e-6,
+        )
         o = ops.paged_attention(q, kv.k_cache[self.layer_idx],
                                 kv.v_cache[self.layer_idx], fb.block_tables,
                                 fb.seq_lens, fb.query_start_locs, self.scale,

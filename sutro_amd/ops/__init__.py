@@ -70,6 +70,40 @@ def silu_mul(x: torch.Tensor) -> torch.Tensor:
     return torch_ref.silu_mul(x)
 
 
+def fused_qkv_prep(
+    qkv: torch.Tensor,            # [T, Hq*D + 2*Hk*D] (qkv GEMM output)
+    num_heads: int, num_kv_heads: int, head_dim: int,
+    positions: torch.Tensor, slot_mapping: torch.Tensor,
+    k_cache: torch.Tensor, v_cache: torch.Tensor,
+    cos_sin: torch.Tensor,
+    q_norm_w: Optional[torch.Tensor] = None,
+    k_norm_w: Optional[torch.Tensor] = None,
+    eps: float = 1e-6,
+) -> torch.Tensor:
+    """Per-head qk-RMSNorm + RoPE + paged KV write; returns contiguous q
+    [T, Hq, D]. One fused kernel on GPU (reads the GEMM output in place)."""
+    T = qkv.shape[0]
+    if _use_hip(qkv):
+        q_out = torch.empty((T, num_heads, head_dim), dtype=qkv.dtype,
+                            device=qkv.device)
+        _require_hip().qkv_prep(qkv, q_out, k_cache, v_cache, positions,
+                                slot_mapping, cos_sin, q_norm_w, k_norm_w, eps)
+        return q_out
+    q_size = num_heads * head_dim
+    kv_size = num_kv_heads * head_dim
+    q, k, v = qkv.split([q_size, kv_size, kv_size], dim=-1)
+    q = q.reshape(T, num_heads, head_dim).contiguous()
+    k = k.reshape(T, num_kv_heads, head_dim).contiguous()
+    v = v.reshape(T, num_kv_heads, head_dim).contiguous()
+    if q_norm_w is not None:
+        q = torch_ref.rmsnorm(q, q_norm_w, eps)
+    if k_norm_w is not None:
+        k = torch_ref.rmsnorm(k, k_norm_w, eps)
+    q, k = torch_ref.apply_rope(q, k, positions, cos_sin)
+    torch_ref.write_kv_cache(k, v, k_cache, v_cache, slot_mapping)
+    return q
+
+
 def rope_and_cache(
     q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     positions: torch.Tensor, slot_mapping: torch.Tensor,

@@ -115,6 +115,29 @@ def test_rope_and_cache():
                           atol=3e-2, rtol=3e-2)
 
 
+def test_qkv_prep_fused():
+    require_gpu()
+    torch.manual_seed(7)
+    T, Hq, Hk, D, bs = 29, 8, 2, 128, 32
+    qkv = bf(torch.randn(T, (Hq + 2 * Hk) * D))
+    pos = torch.randint(0, 400, (T,), dtype=torch.long)
+    slots = torch.randperm(4 * bs)[:T].to(torch.long)
+    cs = R.rope_cos_sin(512, D, 1e6).to(DEV)
+    qw = bf(torch.randn(D) * 0.2 + 1.0)
+    kw = bf(torch.randn(D) * 0.2 + 1.0)
+    kc, vc = _mk_cache(4, Hk, bs, D)
+    q_out = ops.fused_qkv_prep(qkv, Hq, Hk, D, pos.to(DEV), slots.to(DEV),
+                               kc, vc, cs, qw, kw, 1e-6)
+    # CPU reference composition on the same bf16 inputs
+    kc_ref = torch.zeros(4, Hk, bs, D)
+    vc_ref = torch.zeros(4, Hk, bs, D)
+    q_ref = ops.fused_qkv_prep(qkv.cpu(), Hq, Hk, D, pos, slots, kc_ref,
+                               vc_ref, cs.cpu(), qw.cpu(), kw.cpu(), 1e-6)
+    assert torch.allclose(q_out.float().cpu(), q_ref.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(kc.float().cpu(), kc_ref.float(), atol=3e-2, rtol=3e-2)
+    assert torch.allclose(vc.float().cpu(), vc_ref.float(), atol=3e-2, rtol=3e-2)
+
+
 def _attn_setup(seq_lens, new_counts, Hq=8, Hk=2, D=128, bs=32, seed=5):
     """Build a scattered paged cache + q batch; returns gpu tensors + cpu refs."""
     torch.manual_seed(seed)
