@@ -41,6 +41,8 @@ def build_args():
     p.add_argument("--kv-blocks", type=int, default=None)
     p.add_argument("--no-refill", action="store_true",
                    help="measure pure decode of the initial batch (no new rows)")
+    p.add_argument("--tune", action="store_true",
+                   help="run TunableOp GEMM tuning and write the results file")
     return p.parse_args()
 
 
@@ -54,6 +56,12 @@ def main():
     device = args.device or ("cuda" if have_gpu else "cpu")
     if have_gpu:
         torch.cuda.set_device(local_rank)
+
+    if args.tune:
+        os.environ["SUTRO_AMD_TUNABLEOP_TUNE"] = "1"
+        os.environ.setdefault("SUTRO_AMD_TUNABLEOP_FILE",
+                              "gpurun_out/tunableop_gfx950.csv")
+        os.makedirs("gpurun_out", exist_ok=True)
 
     dist = None
     if world > 1:
@@ -176,6 +184,10 @@ def main():
             },
         }
         print(json.dumps(result))
+    if args.tune and have_gpu:
+        import torch.cuda.tunable as tunable
+
+        tunable.write_file()
     if dist is not None:
         dist.destroy_process_group()
 

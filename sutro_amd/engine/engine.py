@@ -35,12 +35,38 @@ class StepStats:
     finished: List[Request] = field(default_factory=list)
 
 
+def _setup_tunableop(cfg: EngineConfig) -> None:
+    """Enable PyTorch TunableOp (hipBLASLt algorithm selection) with the
+    shipped gfx950 tuning table. Set SUTRO_AMD_TUNABLEOP_TUNE=1 to re-tune
+    (slow warmup; call torch.cuda.tunable.write_file() afterwards)."""
+    if not cfg.device.startswith("cuda"):
+        return
+    import os as _os
+
+    if _os.environ.get("SUTRO_AMD_TUNABLEOP", "1") == "0":
+        return
+    try:
+        import torch.cuda.tunable as tunable
+    except ImportError:
+        return
+    path = _os.environ.get(
+        "SUTRO_AMD_TUNABLEOP_FILE",
+        _os.path.join(_os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))),
+                      "data", "tunableop_gfx950.csv"))
+    tunable.enable(True)
+    tunable.set_filename(path)
+    if _os.path.exists(path):
+        tunable.read_file(path)
+    tunable.tuning_enable(_os.environ.get("SUTRO_AMD_TUNABLEOP_TUNE", "0") == "1")
+
+
 class LLMEngine:
     def __init__(self, cfg: EngineConfig, model: Optional[torch.nn.Module] = None):
         self.cfg = cfg
         self.spec: ModelSpec = cfg.spec
         self.device = cfg.device
         self.tokenizer = get_tokenizer()
+        _setup_tunableop(cfg)
         if model is None:
             with torch.device(cfg.device):
                 model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len)
