@@ -187,7 +187,9 @@ def main():
     if args.tune and have_gpu:
         import torch.cuda.tunable as tunable
 
-        tunable.write_file()
+        if hasattr(tunable, "write_file"):
+            tunable.write_file()
+        # newer torch flushes tuned results to the filename automatically
     if dist is not None:
         dist.destroy_process_group()
 
