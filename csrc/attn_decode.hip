@@ -1,15 +1,18 @@
 // Paged decode attention (one new token per sequence), GQA, head_dim 128.
 //
 // Mapping: ONE WAVE per (decode_seq, kv_head). The wave walks the sequence's
-// KV pages (page == KV block, block_size 32). Per page:
-//   phase A: K page -> registers (2 lanes per position, 64 dims each);
-//            scores for the G grouped q-heads via f32 FMA; online softmax.
-//   phase B: V page staged to LDS coalesced; lanes switch to dim-parallel
-//            (2 dims/lane) and accumulate P*V with broadcast P reads.
-// Decode is KV-bandwidth-bound (~8 flop/byte); the VALU path here has ~2x
-// issue headroom over the 6.3 TB/s HBM ceiling, so no MFMA is needed.
-// No cross-wave barriers: waves in a block process independent sequences of
-// different lengths (a block barrier would deadlock).
+// KV pages (page == KV block, block_size 32), software-pipelined:
+//   - K page p+1 is prefetched into a second register buffer while page p
+//     computes (2x-unrolled loop; static buffers per §5.4 rule 20 — a
+//     runtime-indexed buffer array would spill to scratch).
+//   - V page p is loaded into registers at iteration start and only staged to
+//     LDS after phase A (T14 issue-early/write-late: HBM latency hides under
+//     the score computation).
+// Per page: phase A scores 2 lanes/position (64 dims each) against the G
+// grouped q-heads with f32 FMA + online softmax; phase B accumulates P*V
+// dim-parallel (2 dims/lane) from LDS with broadcast P reads.
+// Decode is KV-bandwidth-bound (~8 flop/byte); no MFMA needed.
+// No cross-wave barriers: waves in a block serve independent sequences.
 #include "common.h"
 
 #define BS 32       // KV page size (tokens) == EngineConfig.kv_block_size
@@ -21,6 +24,97 @@ struct DecodeSmem {
   u16 vstage[BS * DHEAD];   // V page, row-major bf16
   float p[MAXG * BS];       // softmax weights for current page
 };
+
+struct DecodeCtx {
+  const u16* k_cache;
+  const u16* v_cache;
+  long kv_base;     // element offset of (block, kv_head) slab, set per page
+  int p_pos;        // position this lane scores (phase A): lane>>1
+  int half;         // which 64-dim half: lane&1
+  int lane;
+  int valid;        // valid positions in current page
+  int G;
+};
+
+__device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
+                                        long kv_base, int p_pos, int half) {
+  const u16* krow = k_cache + kv_base + (long)p_pos * DHEAD + half * 64;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) kreg[j] = *(const u16x8*)(krow + j * 8);
+}
+
+__device__ __forceinline__ void load_v8(u16x8* vreg, const u16* v_cache,
+                                        long kv_base, int lane) {
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int off = (j * (int)WAVE + lane) * 8;
+    vreg[j] = *(const u16x8*)(v_cache + kv_base + off);
+  }
+}
+
+__device__ __forceinline__ void stage_v(DecodeSmem* sm, const u16x8* vreg,
+                                        int lane) {
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = vreg[j];
+}
+
+struct SoftmaxState {
+  float m[MAXG], lsum[MAXG], acc0[MAXG], acc1[MAXG];
+};
+
+// phase A: scores for page from K regs + online-softmax update + P -> LDS
+__device__ __forceinline__ void phase_a(DecodeSmem* sm, SoftmaxState& st,
+                                        const u16x8* kreg, const DecodeCtx& c) {
+  float kf[64];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+#pragma unroll
+    for (int t = 0; t < 8; ++t) kf[j * 8 + t] = bf2f(kreg[j][t]);
+  }
+  for (int g = 0; g < c.G; ++g) {
+    const float* qv = sm->qs + g * DHEAD + c.half * 64;
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      f32x4 q4 = *(const f32x4*)(qv + j * 4);
+      s = fmaf(kf[j * 4 + 0], q4[0], s);
+      s = fmaf(kf[j * 4 + 1], q4[1], s);
+      s = fmaf(kf[j * 4 + 2], q4[2], s);
+      s = fmaf(kf[j * 4 + 3], q4[3], s);
+    }
+    s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
+    if (c.p_pos >= c.valid) s = -1e30f;
+    const float tile_max = wave_max_f32(s);
+    const float m_new = fmaxf(st.m[g], tile_max);
+    const float alpha = __expf(st.m[g] - m_new);
+    const float p_val = __expf(s - m_new);
+    const float tile_sum = wave_sum_f32(p_val) * 0.5f;  // each pos in 2 lanes
+    st.lsum[g] = st.lsum[g] * alpha + tile_sum;
+    st.acc0[g] *= alpha;
+    st.acc1[g] *= alpha;
+    st.m[g] = m_new;
+    if (c.half == 0) sm->p[g * BS + c.p_pos] = p_val;
+  }
+}
+
+// phase B: PV accumulate, lane owns dims (2*lane, 2*lane+1)
+__device__ __forceinline__ void phase_b(DecodeSmem* sm, SoftmaxState& st,
+                                        const DecodeCtx& c) {
+  const int d0 = c.lane * 2;
+  for (int pos = 0; pos < c.valid; ++pos) {
+    u16x2 v2 = *(const u16x2*)(sm->vstage + pos * DHEAD + d0);
+    const float v0 = bf2f(v2[0]), v1 = bf2f(v2[1]);
+    const float* prow = sm->p + pos;  // strided by BS per head
+#pragma unroll
+    for (int g = 0; g < MAXG; ++g) {
+      if (g >= c.G) break;
+      const float pv = prow[g * BS];
+      st.acc0[g] = fmaf(pv, v0, st.acc0[g]);
+      st.acc1[g] = fmaf(pv, v1, st.acc1[g]);
+    }
+  }
+}
 
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
@@ -37,95 +131,61 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   if (item >= (long)n_dec * Hk) return;
   const int sd = (int)(item / Hk);
   const int kh = (int)(item - (long)sd * Hk);
-  const int sg = seq_offset + sd;  // row in block_tables/seq_lens
+  const int sg = seq_offset + sd;
   const int L = seq_lens[sg];
   const int npages = (L + BS - 1) / BS;
+  const int* bt = block_tables + (long)sg * bt_stride;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   DecodeSmem* sm = ((DecodeSmem*)smem_raw) + wid;
 
-  // stage Q (G heads x 128 dims) into LDS as f32, pre-scaled
   for (int g = 0; g < G; ++g) {
     const u16* qrow = q + ((long)sd * Hq + kh * G + g) * DHEAD;
     for (int d = lane; d < DHEAD; d += WAVE)
       sm->qs[g * DHEAD + d] = bf2f(qrow[d]) * scale;
   }
 
-  const int p_pos = lane >> 1;        // position this lane scores (phase A)
-  const int half = lane & 1;          // which 64-dim half
-  float m[MAXG], lsum[MAXG], acc0[MAXG], acc1[MAXG];
+  DecodeCtx c;
+  c.k_cache = k_cache; c.v_cache = v_cache;
+  c.p_pos = lane >> 1; c.half = lane & 1; c.lane = lane; c.G = G;
+
+  SoftmaxState st;
 #pragma unroll
   for (int g = 0; g < MAXG; ++g) {
-    m[g] = -1e30f; lsum[g] = 0.f; acc0[g] = 0.f; acc1[g] = 0.f;
+    st.m[g] = -1e30f; st.lsum[g] = 0.f; st.acc0[g] = 0.f; st.acc1[g] = 0.f;
   }
+
+  auto slab = [&](int pg) {
+    return (((long)bt[pg] * Hk + kh) * BS) * DHEAD;
+  };
+
+  u16x8 kA[8], kB[8], vbuf[8];
+  long base0 = slab(0);
+  load_k8(kA, k_cache, base0, c.p_pos, c.half);
 
   for (int pg = 0; pg < npages; ++pg) {
-    const int blk = block_tables[(long)sg * bt_stride + pg];
-    const long kv_base = (((long)blk * Hk + kh) * BS) * DHEAD;
-    const int valid = min(BS, L - pg * BS);
-
-    // ---- K page -> f32 registers (this lane: pos p_pos, dims half*64..+64)
-    float kf[64];
-    {
-      const u16* krow = k_cache + kv_base + (long)p_pos * DHEAD + half * 64;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        u16x8 kv8 = *(const u16x8*)(krow + j * 8);
-#pragma unroll
-        for (int t = 0; t < 8; ++t) kf[j * 8 + t] = bf2f(kv8[t]);
-      }
+    const long base_cur = slab(pg);
+    const long base_nxt = (pg + 1 < npages) ? slab(pg + 1) : base_cur;
+    const bool even = (pg & 1) == 0;
+    // prefetch next K into the other register buffer
+    if (pg + 1 < npages) {
+      if (even) load_k8(kB, k_cache, base_nxt, c.p_pos, c.half);
+      else      load_k8(kA, k_cache, base_nxt, c.p_pos, c.half);
     }
-    // ---- V page -> LDS (coalesced 16B per lane)
-    {
-      const u16* vsrc = v_cache + kv_base;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int off = (j * (int)WAVE + lane) * 8;  // 8 u16 per slot
-        *(u16x8*)(sm->vstage + off) = *(const u16x8*)(vsrc + off);
-      }
-    }
-
-    // ---- phase A: scores + online softmax per grouped head
-    for (int g = 0; g < G; ++g) {
-      const float* qv = sm->qs + g * DHEAD + half * 64;
-      float s = 0.f;
-#pragma unroll
-      for (int d = 0; d < 64; ++d) s = fmaf(kf[d], qv[d], s);
-      s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
-      if (p_pos >= valid) s = -1e30f;
-      const float tile_max = wave_max_f32(s);
-      const float m_new = fmaxf(m[g], tile_max);
-      const float alpha = __expf(m[g] - m_new);
-      const float p_val = __expf(s - m_new);
-      // each position is present in 2 lanes -> halve the wave sum
-      const float tile_sum = wave_sum_f32(p_val) * 0.5f;
-      lsum[g] = lsum[g] * alpha + tile_sum;
-      acc0[g] *= alpha; acc1[g] *= alpha;
-      m[g] = m_new;
-      if (half == 0) sm->p[g * BS + p_pos] = p_val;
-    }
-
-    // ---- phase B: PV, dim-parallel (lane owns dims 2*lane, 2*lane+1)
-    const int d0 = lane * 2;
-    for (int pos = 0; pos < valid; ++pos) {
-      u16x2 v2 = *(const u16x2*)(sm->vstage + pos * DHEAD + d0);
-      const float v0 = bf2f(v2[0]), v1 = bf2f(v2[1]);
-#pragma unroll
-      for (int g = 0; g < MAXG; ++g) {
-        if (g >= G) break;
-        const float pv = sm->p[g * BS + pos];
-        acc0[g] = fmaf(pv, v0, acc0[g]);
-        acc1[g] = fmaf(pv, v1, acc1[g]);
-      }
-    }
+    // issue V loads now; stage to LDS after phase A (latency hidden)
+    load_v8(vbuf, v_cache, base_cur, lane);
+    c.valid = min(BS, L - pg * BS);
+    if (even) phase_a(sm, st, kA, c);
+    else      phase_a(sm, st, kB, c);
+    stage_v(sm, vbuf, lane);
+    phase_b(sm, st, c);
   }
 
-  // ---- epilogue
   for (int g = 0; g < G; ++g) {
-    const float inv = 1.0f / lsum[g];
+    const float inv = 1.0f / st.lsum[g];
     u16x2 o;
-    o[0] = f2bf(acc0[g] * inv);
-    o[1] = f2bf(acc1[g] * inv);
+    o[0] = f2bf(st.acc0[g] * inv);
+    o[1] = f2bf(st.acc1[g] * inv);
     *(u16x2*)(out + ((long)sd * Hq + kh * G + g) * DHEAD + lane * 2) = o;
   }
 }
