@@ -43,6 +43,9 @@ def build_args():
                    help="measure pure decode of the initial batch (no new rows)")
     p.add_argument("--tune", action="store_true",
                    help="run TunableOp GEMM tuning and write the results file")
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree (ranks per engine replica; "
+                        "dp = world_size // tp)")
     return p.parse_args()
 
 
@@ -84,6 +87,8 @@ def main():
     else:
         spec = get_model_spec(args.model)
 
+    tp = max(1, args.tp)
+    dp_idx = rank // tp
     cfg = EngineConfig(
         spec=spec,
         device=device,
@@ -91,14 +96,16 @@ def main():
         max_num_seqs=args.batch,
         max_tokens_per_step=args.tokens_per_step,
         num_kv_blocks=args.kv_blocks if device != "cpu" else 512,
-        seed=rank,
+        seed=dp_idx,  # identical within a TP group (lockstep), unique per replica
+        tp_size=tp,
     )
     t_init0 = time.time()
     eng = LLMEngine(cfg)
     init_s = time.time() - t_init0
 
-    # synthetic row stream: prompts of prompt_len byte-tokens, seeded per rank
-    rng = np.random.default_rng(1234 + rank)
+    # synthetic row stream: prompts of prompt_len byte-tokens, seeded per
+    # DP replica (identical within a TP group so all its ranks run lockstep)
+    rng = np.random.default_rng(1234 + dp_idx)
 
     def make_prompt():
         body = rng.integers(3, 259, size=args.prompt_len - 1).tolist()
@@ -134,11 +141,13 @@ def main():
     t0 = time.time()
     out_tokens = 0
     sched_tokens = 0
+    count_me = (rank % tp) == 0  # one counter per TP replica
     for _ in range(args.steps):
         refill()
         stats = eng.step()
-        out_tokens += stats.output_tokens
-        sched_tokens += stats.scheduled_tokens
+        if count_me:
+            out_tokens += stats.output_tokens
+            sched_tokens += stats.scheduled_tokens
     if have_gpu:
         torch.cuda.synchronize()
     t1 = time.time()
@@ -173,9 +182,10 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": spec.name,
-                "global_batch": args.batch * n_gpus,
+                "global_batch": args.batch * max(1, n_gpus // tp),
                 "seq_len": args.prompt_len + args.max_new,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"dp{n_gpus}" if tp == 1
+                                else f"dp{max(1, n_gpus // tp)}xtp{tp}"),
                 "prompt_len": args.prompt_len,
                 "max_new_tokens": args.max_new,
                 "rows_per_hour": round(value / max(1, args.max_new) * 3600, 1),

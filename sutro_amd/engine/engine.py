@@ -67,9 +67,14 @@ class LLMEngine:
         self.device = cfg.device
         self.tokenizer = get_tokenizer()
         _setup_tunableop(cfg)
+        from ..parallel.tp import TPContext
+
+        self.tp = (TPContext.from_world(cfg.tp_size) if cfg.tp_size > 1
+                   else TPContext())
         if model is None:
             with torch.device(cfg.device):
-                model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len)
+                model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len,
+                                   self.tp)
             model.init_random_weights(cfg.seed)
         self.model = model.to(cfg.device).eval()
 
@@ -108,7 +113,14 @@ class LLMEngine:
                 and not self.spec.embedding):
             from .graph_runner import DecodeGraphRunner
 
-            self.graph_runner = DecodeGraphRunner(self)
+            try:
+                self.graph_runner = DecodeGraphRunner(self)
+            except Exception as e:
+                import warnings
+
+                warnings.warn(f"hipGraph decode capture failed ({e}); "
+                              f"running decode eagerly")
+                self.graph_runner = None
 
         self._next_req_id = 0
         self._fsms: Dict[int, GuidedFSM] = {}

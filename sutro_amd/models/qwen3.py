@@ -19,7 +19,22 @@ from .. import ops
 from ..engine.batch import ForwardBatch
 from ..engine.kv_cache import PagedKVCache
 from ..ops import torch_ref
+from ..parallel.tp import TPContext
 from .registry import ModelSpec
+
+
+def _mark_shard(p: nn.Parameter, full_shape, dim: int, tp: TPContext,
+                row_sections=None) -> None:
+    """Record how a parameter shards so init can slice a deterministic full
+    tensor (keeps TP=k numerically consistent with TP=1).
+
+    row_sections: for merged projections (qkv, gate_up): list of
+    (full_start, full_rows) sections; each rank takes its slice of each."""
+    p._tp_full_shape = tuple(full_shape)
+    p._tp_dim = dim
+    p._tp_rank = tp.rank
+    p._tp_size = tp.size
+    p._tp_sections = row_sections
 
 
 class RMSNorm(nn.Module):
@@ -34,20 +49,32 @@ class RMSNorm(nn.Module):
 
 class Qwen3Attention(nn.Module):
     def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int,
-                 tp_size: int = 1):
+                 tp: Optional[TPContext] = None):
         super().__init__()
-        assert spec.num_heads % tp_size == 0
+        tp = tp or TPContext()
+        self.tp = tp
+        assert spec.num_heads % tp.size == 0, "num_heads must divide tp_size"
+        assert spec.num_kv_heads % tp.size == 0 or tp.size == 1, \
+            "num_kv_heads must divide tp_size"
         self.layer_idx = layer_idx
-        self.num_heads = spec.num_heads // tp_size
-        self.num_kv_heads = max(1, spec.num_kv_heads // tp_size)
+        self.num_heads = spec.num_heads // tp.size
+        self.num_kv_heads = max(1, spec.num_kv_heads // tp.size)
         self.head_dim = spec.head_dim
         self.q_size = self.num_heads * self.head_dim
         self.kv_size = self.num_kv_heads * self.head_dim
         self.scale = 1.0 / math.sqrt(self.head_dim)
         h = spec.hidden_size
+        qs_full = spec.num_heads * spec.head_dim
+        kv_full = spec.num_kv_heads * spec.head_dim
+        # merged qkv projection, column-parallel per q/k/v section
         self.qkv_proj = nn.Linear(h, self.q_size + 2 * self.kv_size, bias=False,
                                   dtype=dtype)
+        _mark_shard(self.qkv_proj.weight, (qs_full + 2 * kv_full, h), 0, tp,
+                    row_sections=[(0, qs_full), (qs_full, kv_full),
+                                  (qs_full + kv_full, kv_full)])
+        # output projection, row-parallel (one all-reduce per attention block)
         self.o_proj = nn.Linear(self.q_size, h, bias=False, dtype=dtype)
+        _mark_shard(self.o_proj.weight, (h, qs_full), 1, tp)
         self.qk_norm = spec.qk_norm
         if spec.qk_norm:
             self.q_norm = RMSNorm(self.head_dim, spec.rms_eps, dtype)
@@ -70,31 +97,47 @@ class Qwen3Attention(nn.Module):
                                 fb.seq_lens, fb.query_start_locs, self.scale,
                                 fb.num_decodes_tail, fb.tile_seq, fb.tile_q0,
                                 fb.prefill_token_count)
-        return self.o_proj(o.view(T, self.q_size))
+        return self.tp.all_reduce(self.o_proj(o.view(T, self.q_size)))
 
 
 class Qwen3MLP(nn.Module):
-    def __init__(self, hidden: int, intermediate: int, dtype: torch.dtype):
+    def __init__(self, hidden: int, intermediate: int, dtype: torch.dtype,
+                 tp: Optional[TPContext] = None):
         super().__init__()
-        self.gate_up_proj = nn.Linear(hidden, 2 * intermediate, bias=False, dtype=dtype)
-        self.down_proj = nn.Linear(intermediate, hidden, bias=False, dtype=dtype)
+        tp = tp or TPContext()
+        self.tp = tp
+        assert intermediate % tp.size == 0
+        inter_l = intermediate // tp.size
+        self.gate_up_proj = nn.Linear(hidden, 2 * inter_l, bias=False, dtype=dtype)
+        _mark_shard(self.gate_up_proj.weight, (2 * intermediate, hidden), 0, tp,
+                    row_sections=[(0, intermediate), (intermediate, intermediate)])
+        self.down_proj = nn.Linear(inter_l, hidden, bias=False, dtype=dtype)
+        _mark_shard(self.down_proj.weight, (hidden, intermediate), 1, tp)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(ops.silu_mul(self.gate_up_proj(x)))
+        return self.tp.all_reduce(self.down_proj(ops.silu_mul(self.gate_up_proj(x))))
 
 
 class Qwen3MoE(nn.Module):
     """Top-k routed SwiGLU experts. Round-1 execution: sort tokens by expert and
     run per-expert GEMMs on contiguous segments (grouped-GEMM kernel later)."""
 
-    def __init__(self, spec: ModelSpec, dtype: torch.dtype):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype,
+                 tp: Optional[TPContext] = None):
         super().__init__()
+        tp = tp or TPContext()
+        self.tp = tp
         self.num_experts = spec.num_experts
         self.top_k = spec.experts_per_token
         h, m = spec.hidden_size, spec.moe_intermediate_size
+        assert m % tp.size == 0
+        m_l = m // tp.size
         self.router = nn.Linear(h, spec.num_experts, bias=False, dtype=dtype)
-        self.gate_up = nn.Parameter(torch.empty(spec.num_experts, h, 2 * m, dtype=dtype))
-        self.down = nn.Parameter(torch.empty(spec.num_experts, m, h, dtype=dtype))
+        self.gate_up = nn.Parameter(torch.empty(spec.num_experts, h, 2 * m_l, dtype=dtype))
+        _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 2, tp,
+                    row_sections=[(0, m), (m, m)])
+        self.down = nn.Parameter(torch.empty(spec.num_experts, m_l, h, dtype=dtype))
+        _mark_shard(self.down, (spec.num_experts, m, h), 1, tp)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, h = x.shape
@@ -121,19 +164,20 @@ class Qwen3MoE(nn.Module):
         out = torch.zeros_like(x, dtype=torch.float32)
         w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
         out.index_add_(0, sorted_tok, out_sorted.float() * w_sorted)
-        return out.to(x.dtype)
+        return self.tp.all_reduce(out.to(x.dtype))
 
 
 class Qwen3Block(nn.Module):
-    def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int,
+                 tp: Optional[TPContext] = None):
         super().__init__()
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
-        self.self_attn = Qwen3Attention(spec, dtype, layer_idx)
+        self.self_attn = Qwen3Attention(spec, dtype, layer_idx, tp)
         self.post_attention_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
         if spec.num_experts > 0:
-            self.mlp = Qwen3MoE(spec, dtype)
+            self.mlp = Qwen3MoE(spec, dtype, tp)
         else:
-            self.mlp = Qwen3MLP(spec.hidden_size, spec.intermediate_size, dtype)
+            self.mlp = Qwen3MLP(spec.hidden_size, spec.intermediate_size, dtype, tp)
 
     def forward(self, x, residual, fb: ForwardBatch, kv, cos_sin):
         if residual is None:
@@ -155,12 +199,14 @@ class Qwen3Model(nn.Module):
     """Full decoder. For embedding specs there is no lm_head; for generative
     specs `compute_logits` projects selected rows through it."""
 
-    def __init__(self, spec: ModelSpec, dtype: torch.dtype, max_len: int):
+    def __init__(self, spec: ModelSpec, dtype: torch.dtype, max_len: int,
+                 tp: Optional[TPContext] = None):
         super().__init__()
         self.spec = spec
+        self.tp = tp or TPContext()
         self.embed_tokens = nn.Embedding(spec.vocab_size, spec.hidden_size, dtype=dtype)
         self.layers = nn.ModuleList(
-            [Qwen3Block(spec, dtype, i) for i in range(spec.num_layers)]
+            [Qwen3Block(spec, dtype, i, self.tp) for i in range(spec.num_layers)]
         )
         self.norm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
         if not spec.embedding:
@@ -195,14 +241,26 @@ class Qwen3Model(nn.Module):
         import zlib
 
         for name, p in sorted(self.named_parameters()):
-            if p.dim() >= 2:
-                fan_in = p.shape[-1] if "embed" not in name else p.shape[0]
-                std = 0.02 if "embed" in name else (1.0 / math.sqrt(fan_in))
-                dev = p.device
-                gen = torch.Generator(device=dev)
-                gen.manual_seed(seed * 1000003 + zlib.crc32(name.encode()))
-                r = torch.randn(p.shape, generator=gen, dtype=torch.float32,
-                                device=dev)
-                p.copy_((r * std).to(p.dtype))
-            else:
+            if p.dim() < 2:
                 p.fill_(1.0)  # norm weights
+                continue
+            full_shape = getattr(p, "_tp_full_shape", tuple(p.shape))
+            fan_in = full_shape[-1] if "embed" not in name else full_shape[0]
+            std = 0.02 if "embed" in name else (1.0 / math.sqrt(fan_in))
+            dev = p.device
+            gen = torch.Generator(device=dev)
+            gen.manual_seed(seed * 1000003 + zlib.crc32(name.encode()))
+            full = torch.randn(full_shape, generator=gen, dtype=torch.float32,
+                               device=dev) * std
+            tp_size = getattr(p, "_tp_size", 1)
+            if tp_size > 1:
+                dim = p._tp_dim
+                rank = p._tp_rank
+                sections = p._tp_sections or [(0, full_shape[dim])]
+                parts = []
+                for start, length in sections:
+                    per = length // tp_size
+                    a = start + rank * per
+                    parts.append(full.narrow(dim, a, per))
+                full = torch.cat(parts, dim=dim)
+            p.copy_(full.to(p.dtype))

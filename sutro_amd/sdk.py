@@ -205,8 +205,17 @@ class Sutro(ObservabilityMixin, EmbeddingTemplates, ClassificationTemplates,
                                         timeout=600)
                 elif method.upper() == "POST":
                     if files:
-                        resp = requests.post(url, headers=headers, data=payload,
-                                             files=files, timeout=600)
+                        import base64
+
+                        body = dict(payload or {})
+                        body["files"] = {
+                            name: base64.b64encode(
+                                f.read() if hasattr(f, "read") else f
+                            ).decode()
+                            for name, f in files.items()
+                        }
+                        resp = requests.post(url, headers=headers, json=body,
+                                             timeout=600)
                     else:
                         resp = requests.post(url, headers=headers, json=payload,
                                              stream=stream, timeout=600)
