@@ -140,6 +140,12 @@ class Qwen3MoE(nn.Module):
         _mark_shard(self.down, (spec.num_experts, m, h), 1, tp)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.is_cuda:
+            return self.tp.all_reduce(self._forward_batched(x))
+        return self.tp.all_reduce(self._forward_loop(x))
+
+    def _forward_loop(self, x: torch.Tensor) -> torch.Tensor:
+        """Exact segment-per-expert execution (CPU reference path)."""
         T, h = x.shape
         weights, idx = torch_ref.topk_softmax_router(self.router(x), self.top_k)
         flat_expert = idx.reshape(-1)                      # [T*k]
@@ -160,11 +166,54 @@ class Qwen3MoE(nn.Module):
             act = ops.silu_mul(seg @ self.gate_up[e])
             out_sorted[start:start + n] = act @ self.down[e]
             start += n
-        # scatter-add back with routing weights
         out = torch.zeros_like(x, dtype=torch.float32)
         w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
         out.index_add_(0, sorted_tok, out_sorted.float() * w_sorted)
-        return self.tp.all_reduce(out.to(x.dtype))
+        return out.to(x.dtype)
+
+    def _forward_batched(self, x: torch.Tensor,
+                         capacity_factor: float = 2.0) -> torch.Tensor:
+        """Capacity-padded batched-GEMM execution (GPU hot path).
+
+        Tokens are scattered into per-expert slots of a fixed-capacity buffer
+        and run through two hipBLASLt batched GEMMs. Shapes are static in T,
+        with no host synchronization, so decode steps stay hipGraph-capturable.
+        Assignments beyond an expert's capacity (cap = T*top_k/E *
+        capacity_factor) are dropped and their routing weight renormalized
+        away — with top-k renormalized router weights this is the standard
+        capacity-factor semantics."""
+        import math as _math
+
+        T, h = x.shape
+        E, k = self.num_experts, self.top_k
+        cap = max(8, int(_math.ceil(T * k / E * capacity_factor)))
+        cap = min(cap, T)  # an expert can't hold more than every token once
+        weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
+        flat_e = idx.reshape(-1)                          # [T*k]
+        flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
+        order = torch.argsort(flat_e, stable=True)
+        sorted_e = flat_e[order]
+        sorted_tok = flat_tok[order]
+        # rank of each assignment within its expert segment
+        first = torch.searchsorted(sorted_e, sorted_e, side="left")
+        pos = torch.arange(T * k, device=x.device) - first
+        valid = pos < cap
+        # overflow assignments land in a trash row (static shapes everywhere:
+        # boolean-mask indexing would break hipGraph capture)
+        trash = E * cap
+        slot = torch.where(valid, sorted_e * cap + pos,
+                           torch.full_like(pos, trash))
+        buf = x.new_zeros(E * cap + 1, h)
+        buf.index_put_((slot,), x[sorted_tok])
+        act = ops.silu_mul(
+            torch.bmm(buf[:E * cap].view(E, cap, h), self.gate_up)
+            .view(E * cap, -1))
+        out_buf = torch.bmm(act.view(E, cap, -1), self.down).view(E * cap, h)
+        gathered = out_buf[slot.clamp(max=E * cap - 1)].float()
+        out = torch.zeros(T, h, dtype=torch.float32, device=x.device)
+        w_sorted = weights.reshape(-1)[order] * valid.float()
+        out.index_add_(0, sorted_tok, gathered * w_sorted.unsqueeze(-1))
+        return out.to(x.dtype)
 
 
 class Qwen3Block(nn.Module):

@@ -49,7 +49,7 @@ class FunctionStore:
         fn = self.get(name)
         if fn is None:
             raise KeyError(f"unknown function {name!r}")
-        worker = job_service._get_worker(fn["model"])
+        worker = job_service._get_local_worker(fn["model"])
         eng = worker.engine
         from ..engine.request import SamplingParams
         from ..engine.tokenizer import get_tokenizer

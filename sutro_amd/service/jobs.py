@@ -328,6 +328,14 @@ class JobService:
             raise ValueError(
                 f"job exceeds row quota for priority {priority}: "
                 f"{len(inputs)} > {q['row_quota']}")
+        # token quota: byte tokenizer => input tokens ~= utf-8 bytes per row
+        est_tokens = sum(
+            len(r.encode() if isinstance(r, str) else json.dumps(r).encode())
+            for r in inputs)
+        if est_tokens > q["token_quota"]:
+            raise ValueError(
+                f"job exceeds token quota for priority {priority}: "
+                f"~{est_tokens} > {q['token_quota']}")
         job = JobRecord(
             job_id=f"job-{uuid.uuid4().hex[:12]}",
             model=model,
@@ -390,12 +398,43 @@ class JobService:
         id_values = df[id_column].tolist() if id_column else None
         return rows, id_values
 
-    def _get_worker(self, model: str) -> EngineWorker:
+    def _num_dp_workers(self) -> int:
+        env = os.environ.get("SUTRO_AMD_NUM_WORKERS")
+        if env is not None:
+            return max(1, int(env))
+        try:
+            import torch
+
+            n = torch.cuda.device_count()
+        except Exception:
+            n = 0
+        return n if n > 1 else 1
+
+    def _get_worker(self, model: str):
         with self._lock:
             w = self.workers.get(model)
             if w is None:
-                w = EngineWorker(model, self.device, self.engine_kwargs)
+                n = self._num_dp_workers()
+                if n > 1:
+                    from .dp_pool import MultiProcEngineWorker
+
+                    w = MultiProcEngineWorker(model, n, self.device,
+                                              self.engine_kwargs)
+                else:
+                    w = EngineWorker(model, self.device, self.engine_kwargs)
                 self.workers[model] = w
+            return w
+
+    def _get_local_worker(self, model: str) -> EngineWorker:
+        """Always an in-process engine (online Functions need direct access)."""
+        key = f"local:{model}"
+        with self._lock:
+            w = self.workers.get(key)
+            if w is None or not isinstance(w, EngineWorker):
+                w = self.workers.get(model)
+                if not isinstance(w, EngineWorker):
+                    w = EngineWorker(model, self.device, self.engine_kwargs)
+                    self.workers[key] = w
             return w
 
     # ---- lifecycle queries ----
