@@ -142,12 +142,14 @@ def main():
     out_tokens = 0
     sched_tokens = 0
     count_me = (rank % tp) == 0  # one counter per TP replica
+    rows_done = 0
     for _ in range(args.steps):
         refill()
         stats = eng.step()
         if count_me:
             out_tokens += stats.output_tokens
             sched_tokens += stats.scheduled_tokens
+            rows_done += len(stats.finished)
     if have_gpu:
         torch.cuda.synchronize()
     t1 = time.time()
@@ -165,10 +167,21 @@ def main():
         out_tokens, sched_tokens = float(tt[0].item()), float(tt[1].item())
 
     n_gpus = world if world > 1 else args.gpus
-    value = out_tokens / elapsed if elapsed > 0 else 0.0
+    if spec.embedding:
+        # embedding jobs have no decode; throughput = rows (and prompt tokens)
+        if dist is not None:
+            tr = torch.tensor([float(rows_done)],
+                              device=device if have_gpu else "cpu")
+            dist.all_reduce(tr)
+            rows_done = float(tr.item())
+        value = rows_done / elapsed if elapsed > 0 else 0.0
+        metric = "embedding_rows_per_sec"
+    else:
+        value = out_tokens / elapsed if elapsed > 0 else 0.0
+        metric = "output_tokens_per_sec"
     if rank == 0:
         result = {
-            "metric": "output_tokens_per_sec",
+            "metric": metric,
             "value": round(value, 2),
             "unit": "tokens/s",
             "n_gpus": n_gpus,
