@@ -344,18 +344,38 @@ class LLMEngine:
 
     def _embedding_update(self, sb: ScheduledBatch, fb: ForwardBatch,
                           hidden: torch.Tensor, stats: StepStats) -> None:
-        qlocs = fb.query_start_locs
+        # batched per-sequence chunk sums (no per-seq host sync)
+        S = len(sb.reqs)
+        counts = torch.tensor(sb.num_new_tokens, device=hidden.device)
+        seg = torch.repeat_interleave(
+            torch.arange(S, device=hidden.device), counts)
+        sums = torch.zeros(S, hidden.shape[-1], dtype=torch.float32,
+                           device=hidden.device)
+        sums.index_add_(0, seg, hidden.float())
+        done_rows: List[int] = []
+        done_reqs: List[Request] = []
         for s, (req, c) in enumerate(zip(sb.reqs, sb.num_new_tokens)):
-            a, b = int(qlocs[s]), int(qlocs[s + 1])
-            chunk_sum = hidden[a:b].float().sum(dim=0)
-            if req.req_id in self._emb_sums:
-                self._emb_sums[req.req_id] += chunk_sum
-            else:
-                self._emb_sums[req.req_id] = chunk_sum
-            if req.num_computed_tokens + c == req.num_prompt_tokens:
-                vec = self._emb_sums.pop(req.req_id) / max(1, req.num_prompt_tokens)
-                vec = vec / (vec.norm() + 1e-12)
-                self.embeddings[req.req_id] = vec.cpu().numpy()
+            completing = req.num_computed_tokens + c == req.num_prompt_tokens
+            if req.req_id in self._emb_sums or not completing:
+                if req.req_id in self._emb_sums:
+                    self._emb_sums[req.req_id] += sums[s]
+                else:
+                    self._emb_sums[req.req_id] = sums[s].clone()
+            if completing:
+                done_rows.append(s)
+                done_reqs.append(req)
+        if done_reqs:
+            vecs = []
+            for s, req in zip(done_rows, done_reqs):
+                v = self._emb_sums.pop(req.req_id, None)
+                if v is None:
+                    v = sums[s]
+                vecs.append(v / max(1, req.num_prompt_tokens))
+            mat = torch.stack(vecs)
+            mat = mat / (mat.norm(dim=-1, keepdim=True) + 1e-12)
+            mat_cpu = mat.cpu().numpy()  # ONE sync for the whole step
+            for i, req in enumerate(done_reqs):
+                self.embeddings[req.req_id] = mat_cpu[i]
                 self.scheduler.finish(req, FinishReason.STOP)
                 stats.finished.append(req)
 
