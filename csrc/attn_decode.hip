@@ -1,39 +1,39 @@
 // Paged decode attention (one new token per sequence), GQA, head_dim 128.
 //
-// Flash-decode mapping: ONE BLOCK (4 waves) per (decode_seq, kv_head). The
-// sequence's KV pages (page == KV block, 32 tokens) are strided across the 4
-// waves, each running its own online softmax; a final LDS merge combines the
-// per-wave (m, l, acc) — 4x the memory-level parallelism per sequence of a
-// single-wave walk. Within a wave the page loop is software-pipelined:
-//   - K page p+4 prefetched into a second register buffer (static buffers per
-//     §5.4 rule 20) while page p computes;
-//   - V loaded to registers at iteration start, staged to LDS after phase A
-//     (T14 issue-early/write-late).
-// Phase A scores 2 lanes/position (64 dims each) for the G grouped q-heads
-// with f32 FMA; phase B accumulates P*V dim-parallel (2 dims/lane).
+// Mapping: ONE WAVE per (decode_seq, kv_head). The wave walks the sequence's
+// KV pages (page == KV block, block_size 32), software-pipelined:
+//   - K page p+1 is prefetched into a second register buffer while page p
+//     computes (2x-unrolled loop; static buffers per §5.4 rule 20 — a
+//     runtime-indexed buffer array would spill to scratch).
+//   - V page p is loaded into registers at iteration start and only staged to
+//     LDS after phase A (T14 issue-early/write-late: HBM latency hides under
+//     the score computation).
+// Per page: phase A scores 2 lanes/position (64 dims each) against the G
+// grouped q-heads with f32 FMA + online softmax; phase B accumulates P*V
+// dim-parallel (2 dims/lane) from LDS with broadcast P reads.
 // Decode is KV-bandwidth-bound (~8 flop/byte); no MFMA needed.
+// No cross-wave barriers: waves in a block serve independent sequences.
 #include "common.h"
 
-#define BS 32
+#define BS 32       // KV page size (tokens) == EngineConfig.kv_block_size
 #define DHEAD 128
-#define MAXG 8
-#define DEC_WAVES 4
+#define MAXG 8      // max grouped q-heads per kv head handled per wave
 
-struct DecodeWaveSmem {
+struct DecodeSmem {
+  float qs[MAXG * DHEAD];   // Q rows, pre-scaled, f32
   u16 vstage[BS * DHEAD];   // V page, row-major bf16
   float p[MAXG * BS];       // softmax weights for current page
 };
 
-struct DecodeBlockSmem {
-  float qs[MAXG * DHEAD];                       // shared, pre-scaled
-  float comb_m[DEC_WAVES - 1][MAXG];
-  float comb_l[DEC_WAVES - 1][MAXG];
-  float comb_acc[DEC_WAVES - 1][MAXG][DHEAD];   // waves 1..3 dump here
-  DecodeWaveSmem wv[DEC_WAVES];
-};
-
-struct SoftmaxState {
-  float m[MAXG], lsum[MAXG], acc0[MAXG], acc1[MAXG];
+struct DecodeCtx {
+  const u16* k_cache;
+  const u16* v_cache;
+  long kv_base;     // element offset of (block, kv_head) slab, set per page
+  int p_pos;        // position this lane scores (phase A): lane>>1
+  int half;         // which 64-dim half: lane&1
+  int lane;
+  int valid;        // valid positions in current page
+  int G;
 };
 
 __device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
@@ -46,28 +46,34 @@ __device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
 __device__ __forceinline__ void load_v8(u16x8* vreg, const u16* v_cache,
                                         long kv_base, int lane) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
-    vreg[j] = *(const u16x8*)(v_cache + kv_base + (j * (int)WAVE + lane) * 8);
+  for (int j = 0; j < 8; ++j) {
+    const int off = (j * (int)WAVE + lane) * 8;
+    vreg[j] = *(const u16x8*)(v_cache + kv_base + off);
+  }
 }
 
-__device__ __forceinline__ void stage_v(DecodeWaveSmem* sm, const u16x8* vreg,
+__device__ __forceinline__ void stage_v(DecodeSmem* sm, const u16x8* vreg,
                                         int lane) {
 #pragma unroll
   for (int j = 0; j < 8; ++j)
     *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = vreg[j];
 }
 
-__device__ __forceinline__ void phase_a(const float* qs, DecodeWaveSmem* sm,
-                                        SoftmaxState& st, const u16x8* kreg,
-                                        int p_pos, int half, int valid, int G) {
+struct SoftmaxState {
+  float m[MAXG], lsum[MAXG], acc0[MAXG], acc1[MAXG];
+};
+
+// phase A: scores for page from K regs + online-softmax update + P -> LDS
+__device__ __forceinline__ void phase_a(DecodeSmem* sm, SoftmaxState& st,
+                                        const u16x8* kreg, const DecodeCtx& c) {
   float kf[64];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
 #pragma unroll
     for (int t = 0; t < 8; ++t) kf[j * 8 + t] = bf2f(kreg[j][t]);
   }
-  for (int g = 0; g < G; ++g) {
-    const float* qv = qs + g * DHEAD + half * 64;
+  for (int g = 0; g < c.G; ++g) {
+    const float* qv = sm->qs + g * DHEAD + c.half * 64;
     float s = 0.f;
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
@@ -77,31 +83,32 @@ __device__ __forceinline__ void phase_a(const float* qs, DecodeWaveSmem* sm,
       s = fmaf(kf[j * 4 + 2], q4[2], s);
       s = fmaf(kf[j * 4 + 3], q4[3], s);
     }
-    s += __shfl_xor(s, 1, 64);
-    if (p_pos >= valid) s = -1e30f;
+    s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
+    if (c.p_pos >= c.valid) s = -1e30f;
     const float tile_max = wave_max_f32(s);
     const float m_new = fmaxf(st.m[g], tile_max);
     const float alpha = __expf(st.m[g] - m_new);
     const float p_val = __expf(s - m_new);
-    const float tile_sum = wave_sum_f32(p_val) * 0.5f;
+    const float tile_sum = wave_sum_f32(p_val) * 0.5f;  // each pos in 2 lanes
     st.lsum[g] = st.lsum[g] * alpha + tile_sum;
     st.acc0[g] *= alpha;
     st.acc1[g] *= alpha;
     st.m[g] = m_new;
-    if (half == 0) sm->p[g * BS + p_pos] = p_val;
+    if (c.half == 0) sm->p[g * BS + c.p_pos] = p_val;
   }
 }
 
-__device__ __forceinline__ void phase_b(DecodeWaveSmem* sm, SoftmaxState& st,
-                                        int lane, int valid, int G) {
-  const int d0 = lane * 2;
-  for (int pos = 0; pos < valid; ++pos) {
+// phase B: PV accumulate, lane owns dims (2*lane, 2*lane+1)
+__device__ __forceinline__ void phase_b(DecodeSmem* sm, SoftmaxState& st,
+                                        const DecodeCtx& c) {
+  const int d0 = c.lane * 2;
+  for (int pos = 0; pos < c.valid; ++pos) {
     u16x2 v2 = *(const u16x2*)(sm->vstage + pos * DHEAD + d0);
     const float v0 = bf2f(v2[0]), v1 = bf2f(v2[1]);
-    const float* prow = sm->p + pos;
+    const float* prow = sm->p + pos;  // strided by BS per head
 #pragma unroll
     for (int g = 0; g < MAXG; ++g) {
-      if (g >= G) break;
+      if (g >= c.G) break;
       const float pv = prow[g * BS];
       st.acc0[g] = fmaf(pv, v0, st.acc0[g]);
       st.acc1[g] = fmaf(pv, v1, st.acc1[g]);
@@ -109,7 +116,7 @@ __device__ __forceinline__ void phase_b(DecodeWaveSmem* sm, SoftmaxState& st,
   }
 }
 
-__global__ __launch_bounds__(DEC_WAVES * WAVE) void attn_decode_kernel(
+__global__ __launch_bounds__(256) void attn_decode_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
     const u16* __restrict__ k_cache,  // [nb, Hk, BS, D]
@@ -119,7 +126,7 @@ __global__ __launch_bounds__(DEC_WAVES * WAVE) void attn_decode_kernel(
     int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  const long item = blockIdx.x;
+  const long item = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
   const int G = Hq / Hk;
   if (item >= (long)n_dec * Hk) return;
   const int sd = (int)(item / Hk);
@@ -130,18 +137,17 @@ __global__ __launch_bounds__(DEC_WAVES * WAVE) void attn_decode_kernel(
   const int* bt = block_tables + (long)sg * bt_stride;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DecodeBlockSmem* sm = (DecodeBlockSmem*)smem_raw;
-  DecodeWaveSmem* wsm = &sm->wv[wid];
+  DecodeSmem* sm = ((DecodeSmem*)smem_raw) + wid;
 
-  // cooperative Q stage (G heads x 128 dims), pre-scaled
-  for (int i = threadIdx.x; i < G * DHEAD; i += blockDim.x) {
-    const int g = i / DHEAD, d = i - g * DHEAD;
-    sm->qs[i] = bf2f(q[((long)sd * Hq + kh * G + g) * DHEAD + d]) * scale;
+  for (int g = 0; g < G; ++g) {
+    const u16* qrow = q + ((long)sd * Hq + kh * G + g) * DHEAD;
+    for (int d = lane; d < DHEAD; d += WAVE)
+      sm->qs[g * DHEAD + d] = bf2f(qrow[d]) * scale;
   }
-  __syncthreads();
 
-  const int p_pos = lane >> 1;
-  const int half = lane & 1;
+  DecodeCtx c;
+  c.k_cache = k_cache; c.v_cache = v_cache;
+  c.p_pos = lane >> 1; c.half = lane & 1; c.lane = lane; c.G = G;
 
   SoftmaxState st;
 #pragma unroll
@@ -153,59 +159,33 @@ __global__ __launch_bounds__(DEC_WAVES * WAVE) void attn_decode_kernel(
     return (((long)bt[pg] * Hk + kh) * BS) * DHEAD;
   };
 
-  // pages wid, wid+4, wid+8, ... belong to this wave
   u16x8 kA[8], kB[8], vbuf[8];
-  if (wid < npages) load_k8(kA, k_cache, slab(wid), p_pos, half);
-  int iter = 0;
-  for (int pg = wid; pg < npages; pg += DEC_WAVES, ++iter) {
-    const long base_cur = slab(pg);
-    const bool even = (iter & 1) == 0;
-    if (pg + DEC_WAVES < npages) {
-      const long base_nxt = slab(pg + DEC_WAVES);
-      if (even) load_k8(kB, k_cache, base_nxt, p_pos, half);
-      else      load_k8(kA, k_cache, base_nxt, p_pos, half);
-    }
-    load_v8(vbuf, v_cache, base_cur, lane);
-    const int valid = min(BS, L - pg * BS);
-    if (even) phase_a(sm->qs, wsm, st, kA, p_pos, half, valid, G);
-    else      phase_a(sm->qs, wsm, st, kB, p_pos, half, valid, G);
-    stage_v(wsm, vbuf, lane);
-    phase_b(wsm, st, lane, valid, G);
-  }
+  long base0 = slab(0);
+  load_k8(kA, k_cache, base0, c.p_pos, c.half);
 
-  // ---- cross-wave combine (uniform control flow: every wave hits these)
-  if (wid > 0) {
-    for (int g = 0; g < G; ++g) {
-      if (lane == 0) {
-        sm->comb_m[wid - 1][g] = st.m[g];
-        sm->comb_l[wid - 1][g] = st.lsum[g];
-      }
-      sm->comb_acc[wid - 1][g][lane * 2] = st.acc0[g];
-      sm->comb_acc[wid - 1][g][lane * 2 + 1] = st.acc1[g];
+  for (int pg = 0; pg < npages; ++pg) {
+    const long base_cur = slab(pg);
+    const long base_nxt = (pg + 1 < npages) ? slab(pg + 1) : base_cur;
+    const bool even = (pg & 1) == 0;
+    // prefetch next K into the other register buffer
+    if (pg + 1 < npages) {
+      if (even) load_k8(kB, k_cache, base_nxt, c.p_pos, c.half);
+      else      load_k8(kA, k_cache, base_nxt, c.p_pos, c.half);
     }
+    // issue V loads now; stage to LDS after phase A (latency hidden)
+    load_v8(vbuf, v_cache, base_cur, lane);
+    c.valid = min(BS, L - pg * BS);
+    if (even) phase_a(sm, st, kA, c);
+    else      phase_a(sm, st, kB, c);
+    stage_v(sm, vbuf, lane);
+    phase_b(sm, st, c);
   }
-  __syncthreads();
-  if (wid != 0) return;
 
   for (int g = 0; g < G; ++g) {
-    float m_star = st.m[g];
-#pragma unroll
-    for (int w = 0; w < DEC_WAVES - 1; ++w)
-      m_star = fmaxf(m_star, sm->comb_m[w][g]);
-    float scale0 = __expf(st.m[g] - m_star);
-    float acc0 = st.acc0[g] * scale0, acc1 = st.acc1[g] * scale0;
-    float l = st.lsum[g] * scale0;
-#pragma unroll
-    for (int w = 0; w < DEC_WAVES - 1; ++w) {
-      const float f = __expf(sm->comb_m[w][g] - m_star);
-      acc0 = fmaf(sm->comb_acc[w][g][lane * 2], f, acc0);
-      acc1 = fmaf(sm->comb_acc[w][g][lane * 2 + 1], f, acc1);
-      l = fmaf(sm->comb_l[w][g], f, l);
-    }
-    const float inv = 1.0f / l;
+    const float inv = 1.0f / st.lsum[g];
     u16x2 o;
-    o[0] = f2bf(acc0 * inv);
-    o[1] = f2bf(acc1 * inv);
+    o[0] = f2bf(st.acc0[g] * inv);
+    o[1] = f2bf(st.acc1[g] * inv);
     *(u16x2*)(out + ((long)sd * Hq + kh * G + g) * DHEAD + lane * 2) = o;
   }
 }
@@ -216,10 +196,13 @@ extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   int Hq, int Hk, int seq_offset, float scale,
                                   hipStream_t s) {
   if (n_dec == 0) return;
-  const long blocks = (long)n_dec * Hk;
-  const size_t smem = sizeof(DecodeBlockSmem);
+  const int waves_per_block = 4;
+  const long items = (long)n_dec * Hk;
+  const long blocks = (items + waves_per_block - 1) / waves_per_block;
+  const size_t smem = sizeof(DecodeSmem) * waves_per_block;
   hipLaunchKernelGGL(attn_decode_kernel, dim3((unsigned)blocks),
-                     dim3(DEC_WAVES * WAVE), smem, s, (u16*)out, (const u16*)q,
-                     (const u16*)k_cache, (const u16*)v_cache, block_tables,
-                     seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+                     dim3(waves_per_block * WAVE), smem, s, (u16*)out,
+                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
+                     block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
+                     seq_offset, scale);
 }
