@@ -33,6 +33,9 @@ class EngineConfig:
     # tensor parallelism (process group set up by the caller)
     tp_size: int = 1
     tp_rank: int = 0
+    # MoE layers: shard EXPERTS across the tp group (expert parallelism)
+    # instead of sharding every expert's intermediate dim
+    moe_ep: bool = False
 
     def __post_init__(self) -> None:
         if self.device == "cpu":

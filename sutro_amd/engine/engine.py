@@ -74,7 +74,7 @@ class LLMEngine:
         if model is None:
             with torch.device(cfg.device):
                 model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len,
-                                   self.tp)
+                                   self.tp, moe_ep=cfg.moe_ep)
             model.init_random_weights(cfg.seed)
         self.model = model.to(cfg.device).eval()
 

@@ -123,21 +123,45 @@ class Qwen3MoE(nn.Module):
     run per-expert GEMMs on contiguous segments (grouped-GEMM kernel later)."""
 
     def __init__(self, spec: ModelSpec, dtype: torch.dtype,
-                 tp: Optional[TPContext] = None):
+                 tp: Optional[TPContext] = None, ep: bool = False):
+        """Two parallelization modes over the same group:
+        - tp (default): every rank holds ALL experts with the intermediate
+          dim sharded (like the dense MLP).
+        - ep: EXPERTS are sharded across ranks (E/size each, full width);
+          each rank computes the assignments routed to its local experts and
+          the per-block all-reduce combines — the natural expert-parallel
+          form for lockstep TP engines on the fully-connected xGMI mesh
+          (token-sharded all-to-all dispatch applies only when ranks hold
+          different tokens, which lockstep groups do not)."""
         super().__init__()
         tp = tp or TPContext()
         self.tp = tp
+        self.ep = ep and tp.size > 1
         self.num_experts = spec.num_experts
         self.top_k = spec.experts_per_token
         h, m = spec.hidden_size, spec.moe_intermediate_size
-        assert m % tp.size == 0
-        m_l = m // tp.size
         self.router = nn.Linear(h, spec.num_experts, bias=False, dtype=dtype)
-        self.gate_up = nn.Parameter(torch.empty(spec.num_experts, h, 2 * m_l, dtype=dtype))
-        _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 2, tp,
-                    row_sections=[(0, m), (m, m)])
-        self.down = nn.Parameter(torch.empty(spec.num_experts, m_l, h, dtype=dtype))
-        _mark_shard(self.down, (spec.num_experts, m, h), 1, tp)
+        if self.ep:
+            assert spec.num_experts % tp.size == 0
+            e_l = spec.num_experts // tp.size
+            self.experts_per_rank = e_l
+            self.expert_base = tp.rank * e_l
+            self.gate_up = nn.Parameter(torch.empty(e_l, h, 2 * m, dtype=dtype))
+            _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 0, tp)
+            self.down = nn.Parameter(torch.empty(e_l, m, h, dtype=dtype))
+            _mark_shard(self.down, (spec.num_experts, m, h), 0, tp)
+        else:
+            assert m % tp.size == 0
+            m_l = m // tp.size
+            self.experts_per_rank = spec.num_experts
+            self.expert_base = 0
+            self.gate_up = nn.Parameter(
+                torch.empty(spec.num_experts, h, 2 * m_l, dtype=dtype))
+            _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 2, tp,
+                        row_sections=[(0, m), (m, m)])
+            self.down = nn.Parameter(
+                torch.empty(spec.num_experts, m_l, h, dtype=dtype))
+            _mark_shard(self.down, (spec.num_experts, m, h), 1, tp)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.is_cuda:
@@ -158,13 +182,17 @@ class Qwen3MoE(nn.Module):
         counts = torch.bincount(sorted_expert, minlength=self.num_experts)
         start = 0
         counts_l = counts.tolist()
+        base, e_l = self.expert_base, self.experts_per_rank
         for e in range(self.num_experts):
             n = counts_l[e]
             if n == 0:
                 continue
-            seg = gathered[start:start + n]
-            act = ops.silu_mul(seg @ self.gate_up[e])
-            out_sorted[start:start + n] = act @ self.down[e]
+            if base <= e < base + e_l:  # EP: other ranks own the rest
+                seg = gathered[start:start + n]
+                act = ops.silu_mul(seg @ self.gate_up[e - base])
+                out_sorted[start:start + n] = act @ self.down[e - base]
+            else:
+                out_sorted[start:start + n] = 0
             start += n
         out = torch.zeros_like(x, dtype=torch.float32)
         w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
@@ -185,8 +213,9 @@ class Qwen3MoE(nn.Module):
         import math as _math
 
         T, h = x.shape
-        E, k = self.num_experts, self.top_k
-        cap = max(8, int(_math.ceil(T * k / E * capacity_factor)))
+        k = self.top_k
+        E_l, base = self.experts_per_rank, self.expert_base
+        cap = max(8, int(_math.ceil(T * k / self.num_experts * capacity_factor)))
         cap = min(cap, T)  # an expert can't hold more than every token once
         weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
         flat_e = idx.reshape(-1)                          # [T*k]
@@ -197,19 +226,20 @@ class Qwen3MoE(nn.Module):
         # rank of each assignment within its expert segment
         first = torch.searchsorted(sorted_e, sorted_e, side="left")
         pos = torch.arange(T * k, device=x.device) - first
-        valid = pos < cap
-        # overflow assignments land in a trash row (static shapes everywhere:
+        e_loc = sorted_e - base
+        valid = (pos < cap) & (e_loc >= 0) & (e_loc < E_l)  # EP: local only
+        # overflow/non-local assignments land in a trash row (static shapes:
         # boolean-mask indexing would break hipGraph capture)
-        trash = E * cap
-        slot = torch.where(valid, sorted_e * cap + pos,
+        trash = E_l * cap
+        slot = torch.where(valid, e_loc * cap + pos,
                            torch.full_like(pos, trash))
-        buf = x.new_zeros(E * cap + 1, h)
+        buf = x.new_zeros(E_l * cap + 1, h)
         buf.index_put_((slot,), x[sorted_tok])
         act = ops.silu_mul(
-            torch.bmm(buf[:E * cap].view(E, cap, h), self.gate_up)
-            .view(E * cap, -1))
-        out_buf = torch.bmm(act.view(E, cap, -1), self.down).view(E * cap, h)
-        gathered = out_buf[slot.clamp(max=E * cap - 1)].float()
+            torch.bmm(buf[:E_l * cap].view(E_l, cap, h), self.gate_up)
+            .view(E_l * cap, -1))
+        out_buf = torch.bmm(act.view(E_l, cap, -1), self.down).view(E_l * cap, h)
+        gathered = out_buf[slot.clamp(max=E_l * cap - 1)].float()
         out = torch.zeros(T, h, dtype=torch.float32, device=x.device)
         w_sorted = weights.reshape(-1)[order] * valid.float()
         out.index_add_(0, sorted_tok, gathered * w_sorted.unsqueeze(-1))
@@ -218,13 +248,13 @@ class Qwen3MoE(nn.Module):
 
 class Qwen3Block(nn.Module):
     def __init__(self, spec: ModelSpec, dtype: torch.dtype, layer_idx: int,
-                 tp: Optional[TPContext] = None):
+                 tp: Optional[TPContext] = None, moe_ep: bool = False):
         super().__init__()
         self.input_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
         self.self_attn = Qwen3Attention(spec, dtype, layer_idx, tp)
         self.post_attention_layernorm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
         if spec.num_experts > 0:
-            self.mlp = Qwen3MoE(spec, dtype, tp)
+            self.mlp = Qwen3MoE(spec, dtype, tp, ep=moe_ep)
         else:
             self.mlp = Qwen3MLP(spec.hidden_size, spec.intermediate_size, dtype, tp)
 
@@ -249,13 +279,14 @@ class Qwen3Model(nn.Module):
     specs `compute_logits` projects selected rows through it."""
 
     def __init__(self, spec: ModelSpec, dtype: torch.dtype, max_len: int,
-                 tp: Optional[TPContext] = None):
+                 tp: Optional[TPContext] = None, moe_ep: bool = False):
         super().__init__()
         self.spec = spec
         self.tp = tp or TPContext()
         self.embed_tokens = nn.Embedding(spec.vocab_size, spec.hidden_size, dtype=dtype)
         self.layers = nn.ModuleList(
-            [Qwen3Block(spec, dtype, i, self.tp) for i in range(spec.num_layers)]
+            [Qwen3Block(spec, dtype, i, self.tp, moe_ep)
+             for i in range(spec.num_layers)]
         )
         self.norm = RMSNorm(spec.hidden_size, spec.rms_eps, dtype)
         if not spec.embedding:

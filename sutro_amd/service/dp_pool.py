@@ -1,17 +1,24 @@
-"""Data-parallel engine pool: one engine worker PROCESS per GPU.
+"""Data-parallel / tensor-parallel engine pool: one engine process per GPU.
 
-Batch jobs are embarrassingly parallel over rows; on an 8-GPU MI355X node the
-job service shards each job's rows across N worker processes (one per GPU,
-`cuda:i` each), collects per-row results over queues, and merges them
-input-ordered into the JobRecord. No collectives on this path — row scatter
-and result gather are host-side and tiny next to the compute.
+Topology: `world` worker processes partition into `world // tp` replicas of
+`tp` ranks each (tp from the model's recommended_tp, capped by GPU count, or
+SUTRO_AMD_TP). Job rows shard across replicas (DP); within a replica all tp
+ranks run the engine in lockstep over RCCL (gloo on CPU).
 
-Workers are plain `spawn` processes running a continuous-batching engine loop;
-the same pool runs on CPU (dev/tests) with device="cpu".
+Lockstep admission: only a replica's lead rank reads its control queue; each
+loop iteration it broadcasts the drained messages to its tp group
+(`broadcast_object_list`), so every rank admits identical requests in
+identical order and the schedulers never diverge across collectives.
+
+Row results flow back over a queue from each replica's lead rank and merge
+input-ordered into the JobRecord. On an 8-GPU MI355X node this gives e.g.
+dp2 x tp4 for Qwen3-32B with zero host-side coordination on the hot path.
 """
 
 from __future__ import annotations
 
+import json
+import math
 import os
 import queue as queue_mod
 import threading
@@ -21,25 +28,43 @@ from typing import Any, Dict, List, Optional
 from ..interfaces import JobStatus
 
 
-def _worker_main(rank: int, model: str, device: str, engine_kwargs: dict,
-                 in_q, out_q) -> None:
+def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
+                 device: str, engine_kwargs: dict, in_q, out_q) -> None:
     try:
+        import torch
+
+        from ..engine.engine import LLMEngine
         from ..engine.request import SamplingParams
         from ..engine.tokenizer import get_tokenizer
         from .models_map import resolve_engine_config
-        from ..engine.engine import LLMEngine
 
         if device == "auto":
-            import torch
-
             device = f"cuda:{rank}" if torch.cuda.is_available() else "cpu"
         elif device.startswith("cuda"):
             device = f"cuda:{rank}"
-        cfg = resolve_engine_config(model, device=device, **(engine_kwargs or {}))
+        if device.startswith("cuda"):
+            torch.cuda.set_device(int(device.split(":")[1]))
+
+        group = None
+        if world > 1 and tp > 1:
+            import torch.distributed as dist
+
+            os.environ["MASTER_ADDR"] = "127.0.0.1"
+            os.environ["MASTER_PORT"] = str(port)
+            backend = "nccl" if device.startswith("cuda") else "gloo"
+            dist.init_process_group(backend, rank=rank, world_size=world)
+
+        kwargs = dict(engine_kwargs or {})
+        kwargs["tp_size"] = tp
+        cfg = resolve_engine_config(model, device=device, **kwargs)
         eng = LLMEngine(cfg)
+        if tp > 1:
+            group = eng.tp.group
         tok = get_tokenizer()
         spec = eng.spec
-        req_meta: Dict[int, tuple] = {}   # req_id -> (job_id, row_idx, auto_reasoning)
+        tp_rank = rank % tp
+        lead = tp_rank == 0
+        req_meta: Dict[int, tuple] = {}
         fsm_cache: Dict[str, Optional[int]] = {}
         cancelled: set = set()
         out_q.put(("ready", rank, None, None))
@@ -47,18 +72,15 @@ def _worker_main(rank: int, model: str, device: str, engine_kwargs: dict,
         def admit(msg):
             _, job_id, rows, opts = msg
             schema = opts.get("json_schema")
-            auto_reasoning = False
             if schema is None and spec.reasoning and not spec.embedding:
                 schema = {"type": "object", "properties": {
                     "reasoning_content": {"type": "string", "maxLength": 512},
                     "content": {"type": "string", "maxLength": 512}}}
-                auto_reasoning = True
-            key = f"{job_id}"
-            if key not in fsm_cache:
-                fsm_cache[key] = (eng.register_fsm(schema)
-                                  if schema is not None and not spec.embedding
-                                  else None)
-            fsm_id = fsm_cache[key]
+            if job_id not in fsm_cache:
+                fsm_cache[job_id] = (eng.register_fsm(schema)
+                                     if schema is not None and not spec.embedding
+                                     else None)
+            fsm_id = fsm_cache[job_id]
             default_max = 1024 if schema is not None else cfg.default_max_new_tokens
             for row_idx, text in rows:
                 ids = tok.render_prompt(text, opts.get("system_prompt"))
@@ -70,37 +92,58 @@ def _worker_main(rank: int, model: str, device: str, engine_kwargs: dict,
                                       priority=opts.get("priority", 0),
                                       arrival_idx=row_idx,
                                       truncate=opts.get("truncate_rows", True))
-                req_meta[req.req_id] = (job_id, row_idx, auto_reasoning)
+                req_meta[req.req_id] = (job_id, row_idx)
 
-        while True:
-            # drain control queue
+        def handle(msg) -> bool:
+            """Apply one control message; returns False on shutdown."""
+            if msg[0] == "shutdown":
+                return False
+            if msg[0] == "cancel":
+                cancelled.add(msg[1])
+                for req_id, (jid, _) in list(req_meta.items()):
+                    if jid == msg[1]:
+                        r = _find_req(eng, req_id)
+                        if r is not None:
+                            eng.abort_request(r)
+                        req_meta.pop(req_id, None)
+            elif msg[0] == "run_rows" and msg[1] not in cancelled:
+                admit(msg)
+            return True
+
+        def drain_lead() -> List[Any]:
+            msgs = []
             try:
                 while True:
-                    msg = in_q.get_nowait()
-                    if msg[0] == "shutdown":
-                        return
-                    if msg[0] == "cancel":
-                        cancelled.add(msg[1])
-                        for req_id, (jid, _, _) in list(req_meta.items()):
-                            if jid == msg[1]:
-                                r = _find_req(eng, req_id)
-                                if r is not None:
-                                    eng.abort_request(r)
-                                req_meta.pop(req_id, None)
-                    elif msg[0] == "run_rows":
-                        if msg[1] not in cancelled:
-                            admit(msg)
+                    msgs.append(in_q.get_nowait())
             except queue_mod.Empty:
                 pass
+            return msgs
+
+        running = True
+        while running:
+            if tp > 1:
+                import torch.distributed as dist
+
+                box = [drain_lead() if lead else None]
+                dist.broadcast_object_list(box, src=(rank // tp) * tp,
+                                           group=group)
+                msgs = box[0] or []
+            else:
+                msgs = drain_lead()
+            for m in msgs:
+                if not handle(m):
+                    running = False
+            if not running:
+                break
             if eng.has_work():
                 stats = eng.step()
+                if not lead:
+                    continue
                 for req in stats.finished:
                     meta = req_meta.pop(req.req_id, None)
                     if meta is None:
                         continue
-                    job_id, row_idx, _auto = meta
-                    import math
-
+                    job_id, row_idx = meta
                     if spec.embedding:
                         emb = eng.embeddings.pop(req.req_id, None)
                         payload = {"emb": emb.tolist() if emb is not None else None,
@@ -120,16 +163,7 @@ def _worker_main(rank: int, model: str, device: str, engine_kwargs: dict,
                         }
                     out_q.put(("row_done", rank, job_id, (row_idx, payload)))
             else:
-                try:
-                    msg = in_q.get(timeout=0.05)
-                except queue_mod.Empty:
-                    continue
-                if msg[0] == "shutdown":
-                    return
-                if msg[0] == "run_rows" and msg[1] not in cancelled:
-                    admit(msg)
-                elif msg[0] == "cancel":
-                    cancelled.add(msg[1])
+                time.sleep(0.02 if tp > 1 else 0.05)
     except Exception as e:  # pragma: no cover
         out_q.put(("worker_error", rank, None, f"{type(e).__name__}: {e}"))
 
@@ -144,37 +178,42 @@ def _find_req(eng, req_id):
 
 
 class MultiProcEngineWorker:
-    """EngineWorker-compatible facade over N engine processes (DP row shard)."""
+    """EngineWorker-compatible facade over `world` engine processes
+    (`world // tp` DP replicas of `tp` lockstep ranks)."""
 
     def __init__(self, model: str, n_workers: int, device: str = "auto",
-                 engine_kwargs: Optional[dict] = None):
+                 engine_kwargs: Optional[dict] = None, tp: int = 1):
         import torch.multiprocessing as mp
 
         from ..models.registry import get_model_spec
 
         self.model = model
         self.spec = get_model_spec(model)
-        self.n = n_workers
+        assert n_workers % tp == 0, "worker count must be a multiple of tp"
+        self.world = n_workers
+        self.tp = tp
+        self.n_replicas = n_workers // tp
+        port = 29700 + (os.getpid() % 200)
         ctx = mp.get_context("spawn")
         self.in_qs = [ctx.Queue() for _ in range(n_workers)]
         self.out_q = ctx.Queue()
+        kwargs = dict(engine_kwargs or {})
         self.procs = [
             ctx.Process(target=_worker_main,
-                        args=(r, model, device, engine_kwargs or {},
+                        args=(r, n_workers, tp, port, model, device, kwargs,
                               self.in_qs[r], self.out_q), daemon=True)
             for r in range(n_workers)
         ]
         for p in self.procs:
             p.start()
         ready = 0
-        deadline = time.time() + 600
-        while ready < n_workers and time.time() < deadline:
-            kind, rank, _, info = self.out_q.get(timeout=600)
+        while ready < n_workers:
+            kind, rank, _, info = self.out_q.get(timeout=900)
             if kind == "ready":
                 ready += 1
             elif kind == "worker_error":
                 raise RuntimeError(f"engine worker {rank} failed: {info}")
-        self._jobs: Dict[str, tuple] = {}  # job_id -> (JobRecord, service, t0)
+        self._jobs: Dict[str, tuple] = {}
         self._collector = threading.Thread(target=self._collect, daemon=True)
         self._collector.start()
 
@@ -191,30 +230,30 @@ class MultiProcEngineWorker:
             "truncate_rows": job.truncate_rows,
             "priority": job.priority,
         }
-        # contiguous row shards, one per worker
-        rows = [(i, r if isinstance(r, str) else __import__("json").dumps(r))
+        rows = [(i, r if isinstance(r, str) else json.dumps(r))
                 for i, r in enumerate(job.inputs)]
-        per = (len(rows) + self.n - 1) // self.n
-        for w in range(self.n):
-            shard = rows[w * per:(w + 1) * per]
+        per = (len(rows) + self.n_replicas - 1) // self.n_replicas
+        for rep in range(self.n_replicas):
+            shard = rows[rep * per:(rep + 1) * per]
             if shard:
-                self.in_qs[w].put(("run_rows", job.job_id, shard, opts))
+                # the replica's LEAD rank gets the rows; it broadcasts to its
+                # tp group before admission
+                self.in_qs[rep * self.tp].put(("run_rows", job.job_id, shard,
+                                               opts))
 
     def cancel_job(self, job_id: str) -> None:
-        for q in self.in_qs:
-            q.put(("cancel", job_id))
+        for rep in range(self.n_replicas):
+            self.in_qs[rep * self.tp].put(("cancel", job_id))
 
     def shutdown(self) -> None:
-        for q in self.in_qs:
-            q.put(("shutdown",))
+        for rep in range(self.n_replicas):
+            self.in_qs[rep * self.tp].put(("shutdown",))
         for p in self.procs:
-            p.join(timeout=10)
+            p.join(timeout=15)
             if p.is_alive():
                 p.terminate()
 
     def _collect(self) -> None:
-        import math
-
         while True:
             try:
                 kind, rank, job_id, data = self.out_q.get(timeout=0.5)
@@ -240,10 +279,8 @@ class MultiProcEngineWorker:
                 else:
                     text = payload["output"]
                     if self.spec.reasoning and job.json_schema is not None:
-                        import json as _json
-
-                        text = _json.dumps({"content": text,
-                                            "reasoning_content": ""})
+                        text = json.dumps({"content": text,
+                                           "reasoning_content": ""})
                     job.outputs[row_idx] = text
                     job.cumulative_logprobs[row_idx] = payload.get(
                         "cumulative_logprob")

@@ -418,8 +418,14 @@ class JobService:
                 if n > 1:
                     from .dp_pool import MultiProcEngineWorker
 
+                    spec = get_model_spec(model)
+                    tp_env = os.environ.get("SUTRO_AMD_TP")
+                    tp = int(tp_env) if tp_env else min(spec.recommended_tp, n)
+                    tp = max(1, tp)
+                    while n % tp:
+                        tp -= 1
                     w = MultiProcEngineWorker(model, n, self.device,
-                                              self.engine_kwargs)
+                                              self.engine_kwargs, tp=tp)
                 else:
                     w = EngineWorker(model, self.device, self.engine_kwargs)
                 self.workers[model] = w

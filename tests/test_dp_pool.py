@@ -73,3 +73,28 @@ def test_dp_guided_job(dp_service):
     results = dp_service.job_results(job_id)["results"]
     for o in results["outputs"]:
         assert json.loads(o)["label"] in ("A", "B")
+
+
+def test_dp_pool_with_tp_groups(sutro_home, monkeypatch):
+    """2 workers forming ONE tp=2 replica over gloo: lockstep admission via
+    group broadcast; results complete and ordered."""
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "2")
+    monkeypatch.setenv("SUTRO_AMD_TP", "2")
+    from sutro_amd.service.jobs import JobService
+
+    svc = JobService(home=sutro_home + "-tp", device="cpu",
+                     engine_kwargs={"num_kv_blocks": 128, "max_model_len": 1024})
+    try:
+        out = svc.submit_job({
+            "model": "qwen-3.5-2b", "inputs": ["alpha", "beta", "gamma"],
+            "sampling_params": {"max_tokens": 6, "temperature": 0.7},
+        })
+        job_id = out["results"]
+        assert _await(svc, job_id) == "SUCCEEDED"
+        w = svc.workers["qwen-3.5-2b"]
+        assert w.tp == 2 and w.n_replicas == 1
+        results = svc.job_results(job_id)["results"]
+        assert len(results["outputs"]) == 3
+        assert all(o is not None for o in results["outputs"])
+    finally:
+        svc.shutdown()
