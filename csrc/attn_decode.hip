@@ -16,13 +16,13 @@
 #include "common.h"
 
 #define BS 32       // KV page size (tokens) == EngineConfig.kv_block_size
-#define DHEAD 128
 #define MAXG 8      // max grouped q-heads per kv head handled per wave
 
+template <int D>
 struct DecodeSmem {
-  float qs[MAXG * DHEAD];   // Q rows, pre-scaled, f32
-  u16 vstage[BS * DHEAD];   // V page, row-major bf16
-  float p[MAXG * BS];       // softmax weights for current page
+  float qs[MAXG * D];   // Q rows, pre-scaled, f32
+  u16 vstage[BS * D];   // V page, row-major bf16
+  float p[MAXG * BS];   // softmax weights for current page
 };
 
 struct DecodeCtx {
@@ -36,26 +36,29 @@ struct DecodeCtx {
   int G;
 };
 
+template <int D>
 __device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
                                         long kv_base, int p_pos, int half) {
-  const u16* krow = k_cache + kv_base + (long)p_pos * DHEAD + half * 64;
+  const u16* krow = k_cache + kv_base + (long)p_pos * D + half * (D / 2);
 #pragma unroll
-  for (int j = 0; j < 8; ++j) kreg[j] = *(const u16x8*)(krow + j * 8);
+  for (int j = 0; j < D / 16; ++j) kreg[j] = *(const u16x8*)(krow + j * 8);
 }
 
+template <int D>
 __device__ __forceinline__ void load_v8(u16x8* vreg, const u16* v_cache,
                                         long kv_base, int lane) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
+  for (int j = 0; j < D / 16; ++j) {
     const int off = (j * (int)WAVE + lane) * 8;
     vreg[j] = *(const u16x8*)(v_cache + kv_base + off);
   }
 }
 
-__device__ __forceinline__ void stage_v(DecodeSmem* sm, const u16x8* vreg,
+template <int D>
+__device__ __forceinline__ void stage_v(DecodeSmem<D>* sm, const u16x8* vreg,
                                         int lane) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j)
+  for (int j = 0; j < D / 16; ++j)
     *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = vreg[j];
 }
 
@@ -64,19 +67,20 @@ struct SoftmaxState {
 };
 
 // phase A: scores for page from K regs + online-softmax update + P -> LDS
-__device__ __forceinline__ void phase_a(DecodeSmem* sm, SoftmaxState& st,
+template <int D>
+__device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState& st,
                                         const u16x8* kreg, const DecodeCtx& c) {
-  float kf[64];
+  float kf[D / 2];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
+  for (int j = 0; j < D / 16; ++j) {
 #pragma unroll
     for (int t = 0; t < 8; ++t) kf[j * 8 + t] = bf2f(kreg[j][t]);
   }
   for (int g = 0; g < c.G; ++g) {
-    const float* qv = sm->qs + g * DHEAD + c.half * 64;
+    const float* qv = sm->qs + g * D + c.half * (D / 2);
     float s = 0.f;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) {
+    for (int j = 0; j < D / 8; ++j) {
       f32x4 q4 = *(const f32x4*)(qv + j * 4);
       s = fmaf(kf[j * 4 + 0], q4[0], s);
       s = fmaf(kf[j * 4 + 1], q4[1], s);
@@ -98,13 +102,20 @@ __device__ __forceinline__ void phase_a(DecodeSmem* sm, SoftmaxState& st,
   }
 }
 
-// phase B: PV accumulate, lane owns dims (2*lane, 2*lane+1)
-__device__ __forceinline__ void phase_b(DecodeSmem* sm, SoftmaxState& st,
+// phase B: PV accumulate; lane owns D/64 dims (2 at D=128, 1 at D=64)
+template <int D>
+__device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState& st,
                                         const DecodeCtx& c) {
-  const int d0 = c.lane * 2;
+  const int dpl = D / (int)WAVE;  // dims per lane
+  const int d0 = c.lane * dpl;
   for (int pos = 0; pos < c.valid; ++pos) {
-    u16x2 v2 = *(const u16x2*)(sm->vstage + pos * DHEAD + d0);
-    const float v0 = bf2f(v2[0]), v1 = bf2f(v2[1]);
+    float v0, v1;
+    if (dpl == 2) {
+      u16x2 v2 = *(const u16x2*)(sm->vstage + pos * D + d0);
+      v0 = bf2f(v2[0]); v1 = bf2f(v2[1]);
+    } else {
+      v0 = bf2f(sm->vstage[pos * D + d0]); v1 = 0.f;
+    }
     const float* prow = sm->p + pos;  // strided by BS per head
 #pragma unroll
     for (int g = 0; g < MAXG; ++g) {
@@ -116,6 +127,7 @@ __device__ __forceinline__ void phase_b(DecodeSmem* sm, SoftmaxState& st,
   }
 }
 
+template <int D>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
@@ -137,12 +149,12 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   const int* bt = block_tables + (long)sg * bt_stride;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DecodeSmem* sm = ((DecodeSmem*)smem_raw) + wid;
+  DecodeSmem<D>* sm = ((DecodeSmem<D>*)smem_raw) + wid;
 
   for (int g = 0; g < G; ++g) {
-    const u16* qrow = q + ((long)sd * Hq + kh * G + g) * DHEAD;
-    for (int d = lane; d < DHEAD; d += WAVE)
-      sm->qs[g * DHEAD + d] = bf2f(qrow[d]) * scale;
+    const u16* qrow = q + ((long)sd * Hq + kh * G + g) * D;
+    for (int d = lane; d < D; d += WAVE)
+      sm->qs[g * D + d] = bf2f(qrow[d]) * scale;
   }
 
   DecodeCtx c;
@@ -156,12 +168,12 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   }
 
   auto slab = [&](int pg) {
-    return (((long)bt[pg] * Hk + kh) * BS) * DHEAD;
+    return (((long)bt[pg] * Hk + kh) * BS) * D;
   };
 
-  u16x8 kA[8], kB[8], vbuf[8];
+  u16x8 kA[D / 16], kB[D / 16], vbuf[D / 16];
   long base0 = slab(0);
-  load_k8(kA, k_cache, base0, c.p_pos, c.half);
+  load_k8<D>(kA, k_cache, base0, c.p_pos, c.half);
 
   for (int pg = 0; pg < npages; ++pg) {
     const long base_cur = slab(pg);
@@ -169,40 +181,53 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     const bool even = (pg & 1) == 0;
     // prefetch next K into the other register buffer
     if (pg + 1 < npages) {
-      if (even) load_k8(kB, k_cache, base_nxt, c.p_pos, c.half);
-      else      load_k8(kA, k_cache, base_nxt, c.p_pos, c.half);
+      if (even) load_k8<D>(kB, k_cache, base_nxt, c.p_pos, c.half);
+      else      load_k8<D>(kA, k_cache, base_nxt, c.p_pos, c.half);
     }
     // issue V loads now; stage to LDS after phase A (latency hidden)
-    load_v8(vbuf, v_cache, base_cur, lane);
+    load_v8<D>(vbuf, v_cache, base_cur, lane);
     c.valid = min(BS, L - pg * BS);
-    if (even) phase_a(sm, st, kA, c);
-    else      phase_a(sm, st, kB, c);
-    stage_v(sm, vbuf, lane);
-    phase_b(sm, st, c);
+    if (even) phase_a<D>(sm, st, kA, c);
+    else      phase_a<D>(sm, st, kB, c);
+    stage_v<D>(sm, vbuf, lane);
+    phase_b<D>(sm, st, c);
   }
 
   for (int g = 0; g < G; ++g) {
     const float inv = 1.0f / st.lsum[g];
-    u16x2 o;
-    o[0] = f2bf(st.acc0[g] * inv);
-    o[1] = f2bf(st.acc1[g] * inv);
-    *(u16x2*)(out + ((long)sd * Hq + kh * G + g) * DHEAD + lane * 2) = o;
+    if (D / (int)WAVE == 2) {
+      u16x2 o;
+      o[0] = f2bf(st.acc0[g] * inv);
+      o[1] = f2bf(st.acc1[g] * inv);
+      *(u16x2*)(out + ((long)sd * Hq + kh * G + g) * D + lane * 2) = o;
+    } else {
+      out[((long)sd * Hq + kh * G + g) * D + lane] = f2bf(st.acc0[g] * inv);
+    }
   }
 }
 
 extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   const void* v_cache, const int* block_tables,
                                   const int* seq_lens, int bt_stride, int n_dec,
-                                  int Hq, int Hk, int seq_offset, float scale,
-                                  hipStream_t s) {
+                                  int Hq, int Hk, int head_dim, int seq_offset,
+                                  float scale, hipStream_t s) {
   if (n_dec == 0) return;
   const int waves_per_block = 4;
   const long items = (long)n_dec * Hk;
   const long blocks = (items + waves_per_block - 1) / waves_per_block;
-  const size_t smem = sizeof(DecodeSmem) * waves_per_block;
-  hipLaunchKernelGGL(attn_decode_kernel, dim3((unsigned)blocks),
-                     dim3(waves_per_block * WAVE), smem, s, (u16*)out,
-                     (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
-                     block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
-                     seq_offset, scale);
+  if (head_dim == 128) {
+    const size_t smem = sizeof(DecodeSmem<128>) * waves_per_block;
+    hipLaunchKernelGGL(attn_decode_kernel<128>, dim3((unsigned)blocks),
+                       dim3(waves_per_block * WAVE), smem, s, (u16*)out,
+                       (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
+                       block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
+                       seq_offset, scale);
+  } else {
+    const size_t smem = sizeof(DecodeSmem<64>) * waves_per_block;
+    hipLaunchKernelGGL(attn_decode_kernel<64>, dim3((unsigned)blocks),
+                       dim3(waves_per_block * WAVE), smem, s, (u16*)out,
+                       (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
+                       block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
+                       seq_offset, scale);
+  }
 }

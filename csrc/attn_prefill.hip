@@ -17,7 +17,6 @@
 #include "common.h"
 
 #define BS 32
-#define DHEAD 128
 #define QTILE 32
 #define VT_PAD 40  // padded row length (elems) of the transposed V tile
 
@@ -31,20 +30,25 @@ __device__ __forceinline__ int d_row(int r, int hi) {
   return (r & 3) + 8 * (r >> 2) + 4 * hi;
 }
 
-// K tile XOR swizzle: row-major [32][128] bf16 would put a whole 16-lane
-// ds_read_b128 group on one 16B slot (16-way conflict). byte ^= (row&15)<<4.
+// K tile XOR swizzle: a row-major [32][D] bf16 tile puts a whole 16-lane
+// ds_read_b128 group on few 16B slots. byte ^= (row&mask)<<4, mask chosen so
+// the swizzle stays inside the row (D=128: 15, D=64: 7).
+template <int D>
 __device__ __forceinline__ int k_swz(int row, int byte_in_row) {
-  return row * 256 + (byte_in_row ^ ((row & 15) << 4));
+  constexpr int mask = (D == 128) ? 15 : 7;
+  return row * (2 * D) + (byte_in_row ^ ((row & mask) << 4));
 }
 
+template <int D>
 struct PrefillSmem {
-  u16 ktile[BS * DHEAD];        // swizzled K page (8 KB)
-  u16 vt[DHEAD * VT_PAD];       // V^T, padded rows (10 KB)
+  u16 ktile[BS * D];        // swizzled K page
+  u16 vt[D * VT_PAD];       // V^T, padded rows
   // per-wave regions follow (P tile + softmax broadcast), carved at runtime
 };
 
 #define PWAVE_ELEMS (QTILE * VT_PAD)  // P tile per wave (bf16)
 
+template <int D>
 __global__ void attn_prefill_kernel(
     u16* __restrict__ out,            // [T, Hq, D]
     const u16* __restrict__ q,        // [T, Hq, D]
@@ -75,9 +79,9 @@ __global__ void attn_prefill_kernel(
   const int ntiles_kv = (kv_end + BS - 1) / BS;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  PrefillSmem* sm = (PrefillSmem*)smem_raw;
-  u16* p_lds = (u16*)(smem_raw + sizeof(PrefillSmem)) + (long)w * PWAVE_ELEMS;
-  float* stat_lds = (float*)((u16*)(smem_raw + sizeof(PrefillSmem)) +
+  PrefillSmem<D>* sm = (PrefillSmem<D>*)smem_raw;
+  u16* p_lds = (u16*)(smem_raw + sizeof(PrefillSmem<D>)) + (long)w * PWAVE_ELEMS;
+  float* stat_lds = (float*)((u16*)(smem_raw + sizeof(PrefillSmem<D>)) +
                              (long)G * PWAVE_ELEMS) + w * 2 * QTILE;
   float* alpha_lds = stat_lds;           // [32]
   float* lsum_lds = stat_lds + QTILE;    // [32]
@@ -85,39 +89,40 @@ __global__ void attn_prefill_kernel(
   const int my_q = lane & 31;            // q row this lane's scores belong to
   const int hi = lane >> 5;
 
-  // ---- Q fragments (B-operand), loaded once: 8 k-steps x 8 bf16
-  bf16frag qb[8];
+  // ---- Q fragments (B-operand), loaded once: D/16 k-steps x 8 bf16
+  bf16frag qb[D / 16];
   {
     const int qrow = row_base + (my_q < nq ? my_q : 0);
-    const u16* qptr = q + ((long)qrow * Hq + hq) * DHEAD + hi * 8;
+    const u16* qptr = q + ((long)qrow * Hq + hq) * D + hi * 8;
 #pragma unroll
-    for (int kk = 0; kk < 8; ++kk)
+    for (int kk = 0; kk < D / 16; ++kk)
       qb[kk] = *(const s16x8*)(qptr + kk * 16);
   }
 
-  f32x16 acc_o[4];  // PV accumulator, d-blocks of 32
+  f32x16 acc_o[D / 32];  // PV accumulator, d-blocks of 32
 #pragma unroll
-  for (int b = 0; b < 4; ++b) acc_o[b] = (f32x16)(0.f);
+  for (int b = 0; b < D / 32; ++b) acc_o[b] = (f32x16)(0.f);
   float m_run = -1e30f, l_run = 0.f;
 
   for (int kt = 0; kt < ntiles_kv; ++kt) {
     const int blk = block_tables[(long)s * bt_stride + kt];
-    const long kv_base = (((long)blk * Hk + kh) * BS) * DHEAD;
+    const long kv_base = (((long)blk * Hk + kh) * BS) * D;
     const int kv0 = kt * BS;
+    constexpr int SLOTS = D / 8;  // 16B chunks per row
 
     // ---- cooperative stage: K (swizzled) and V^T into LDS
     {
       const int tid = threadIdx.x, nthr = blockDim.x;
-      // K: 32 rows x 256B; thread moves 16B: item = row*16 + slot
-      for (int it = tid; it < BS * 16; it += nthr) {
-        const int row = it >> 4, slot = it & 15;
-        u16x8 kx = *(const u16x8*)(k_cache + kv_base + row * DHEAD + slot * 8);
-        *(u16x8*)((char*)sm->ktile + k_swz(row, slot * 16)) = kx;
+      // K: 32 rows x 2D bytes; thread moves 16B: item = row*SLOTS + slot
+      for (int it = tid; it < BS * SLOTS; it += nthr) {
+        const int row = it / SLOTS, slot = it % SLOTS;
+        u16x8 kx = *(const u16x8*)(k_cache + kv_base + row * D + slot * 8);
+        *(u16x8*)((char*)sm->ktile + k_swz<D>(row, slot * 16)) = kx;
       }
       // V^T: read V[kv][d0..d0+8), write 8 u16 at vt[d][kv]
-      for (int it = tid; it < BS * 16; it += nthr) {
-        const int kv = it >> 4, d0 = (it & 15) * 8;
-        u16x8 vx = *(const u16x8*)(v_cache + kv_base + kv * DHEAD + d0);
+      for (int it = tid; it < BS * SLOTS; it += nthr) {
+        const int kv = it / SLOTS, d0 = (it % SLOTS) * 8;
+        u16x8 vx = *(const u16x8*)(v_cache + kv_base + kv * D + d0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) sm->vt[(d0 + j) * VT_PAD + kv] = vx[j];
       }
@@ -127,11 +132,11 @@ __global__ void attn_prefill_kernel(
     // ---- QK^T: D1[kv][q] = sum_k K[kv][k] * Q^T[k][q]
     f32x16 d1 = (f32x16)(0.f);
 #pragma unroll
-    for (int kk = 0; kk < 8; ++kk) {
+    for (int kk = 0; kk < D / 16; ++kk) {
       // A = K frag: lane holds K[l%32][kk*16 + hi*8 + j]
       const int row = lane & 31;
       const int byte_in_row = (kk * 16 + hi * 8) * 2;
-      bf16frag ka = *(const s16x8*)((char*)sm->ktile + k_swz(row, byte_in_row));
+      bf16frag ka = *(const s16x8*)((char*)sm->ktile + k_swz<D>(row, byte_in_row));
       d1 = mfma32(ka, qb[kk], d1);
     }
 
@@ -172,7 +177,7 @@ __global__ void attn_prefill_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) al[r] = alpha_lds[d_row(r, hi)];
 #pragma unroll
-    for (int b = 0; b < 4; ++b)
+    for (int b = 0; b < D / 32; ++b)
 #pragma unroll
       for (int r = 0; r < 16; ++r) acc_o[b][r] *= al[r];
 
@@ -182,7 +187,7 @@ __global__ void attn_prefill_kernel(
     for (int kk = 0; kk < 2; ++kk)
       pa[kk] = *(const s16x8*)(p_lds + (lane & 31) * VT_PAD + kk * 16 + hi * 8);
 #pragma unroll
-    for (int b = 0; b < 4; ++b) {
+    for (int b = 0; b < D / 32; ++b) {
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
         // B = V^T frag: lane holds V[kk*16 + hi*8 + j][b*32 + l%32]
@@ -205,12 +210,12 @@ __global__ void attn_prefill_kernel(
   }
   const int d_col = lane & 31;
 #pragma unroll
-  for (int b = 0; b < 4; ++b) {
+  for (int b = 0; b < D / 32; ++b) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = d_row(r, hi);
       if (qrow < nq) {
-        out[((long)(row_base + qrow) * Hq + hq) * DHEAD + b * 32 + d_col] =
+        out[((long)(row_base + qrow) * Hq + hq) * D + b * 32 + d_col] =
             f2bf(acc_o[b][r] * inv[r]);
       }
     }
@@ -223,16 +228,28 @@ extern "C" void sutro_attn_prefill(void* out, const void* q,
                                    const int* seq_lens, const int* qlocs,
                                    const int* tile_seq, const int* tile_q0,
                                    int n_tiles, int bt_stride, int Hq, int Hk,
-                                   float scale, hipStream_t s) {
+                                   int head_dim, float scale, hipStream_t s) {
   if (n_tiles == 0) return;
   const int G = Hq / Hk;
-  const size_t smem = sizeof(PrefillSmem) +
-                      (size_t)G * PWAVE_ELEMS * sizeof(u16) +
-                      (size_t)G * 2 * QTILE * sizeof(float);
-  hipLaunchKernelGGL(attn_prefill_kernel, dim3(n_tiles, Hk), dim3(G * WAVE),
-                     smem, s, (u16*)out, (const u16*)q, (const u16*)k_cache,
-                     (const u16*)v_cache, block_tables, seq_lens, qlocs,
-                     tile_seq, tile_q0, bt_stride, Hq, Hk, scale);
+  if (head_dim == 128) {
+    const size_t smem = sizeof(PrefillSmem<128>) +
+                        (size_t)G * PWAVE_ELEMS * sizeof(u16) +
+                        (size_t)G * 2 * QTILE * sizeof(float);
+    hipLaunchKernelGGL(attn_prefill_kernel<128>, dim3(n_tiles, Hk),
+                       dim3(G * WAVE), smem, s, (u16*)out, (const u16*)q,
+                       (const u16*)k_cache, (const u16*)v_cache, block_tables,
+                       seq_lens, qlocs, tile_seq, tile_q0, bt_stride, Hq, Hk,
+                       scale);
+  } else {
+    const size_t smem = sizeof(PrefillSmem<64>) +
+                        (size_t)G * PWAVE_ELEMS * sizeof(u16) +
+                        (size_t)G * 2 * QTILE * sizeof(float);
+    hipLaunchKernelGGL(attn_prefill_kernel<64>, dim3(n_tiles, Hk),
+                       dim3(G * WAVE), smem, s, (u16*)out, (const u16*)q,
+                       (const u16*)k_cache, (const u16*)v_cache, block_tables,
+                       seq_lens, qlocs, tile_seq, tile_q0, bt_stride, Hq, Hk,
+                       scale);
+  }
 }
 
 // ---------------------------------------------------------------------------

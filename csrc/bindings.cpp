@@ -27,11 +27,12 @@ void sutro_rope_and_cache(void*, void*, const void*, const long*, const long*,
 void sutro_mean_pool_normalize(float*, const void*, const int*, int, int,
                                hipStream_t);
 void sutro_attn_decode(void*, const void*, const void*, const void*,
-                       const int*, const int*, int, int, int, int, int, float,
-                       hipStream_t);
+                       const int*, const int*, int, int, int, int, int, int,
+                       float, hipStream_t);
 void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         const int*, const int*, const int*, const int*,
-                        const int*, int, int, int, int, float, hipStream_t);
+                        const int*, int, int, int, int, int, float,
+                        hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
@@ -102,7 +103,8 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
   const int Hk = k_cache.size(1);
   const int bs = k_cache.size(2);
   const int S = seq_lens.size(0);
-  TORCH_CHECK(D == 128, "attention kernels support head_dim 128 (got ", D, ")");
+  TORCH_CHECK(D == 128 || D == 64,
+              "attention kernels support head_dim 64/128 (got ", D, ")");
   TORCH_CHECK(bs == 32, "attention kernels require kv_block_size 32");
   TORCH_CHECK(Hq % Hk == 0 && Hq / Hk <= 8,
               "GQA group size must divide and be <= 8");
@@ -113,7 +115,8 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                        v_cache.data_ptr(), block_tables.data_ptr<int>(),
                        seq_lens.data_ptr<int>(), qlocs.data_ptr<int>(),
                        tile_seq.data_ptr<int>(), tile_q0.data_ptr<int>(),
-                       n_tiles, bt_stride, Hq, Hk, (float)scale, cur_stream());
+                       n_tiles, bt_stride, Hq, Hk, D, (float)scale,
+                       cur_stream());
   }
   if (num_decodes > 0) {
     const long dec_off = prefill_token_count;  // decode rows are the tail
@@ -123,7 +126,7 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
     sutro_attn_decode(op + dec_off * Hq * D, qp + dec_off * Hq * D,
                       k_cache.data_ptr(), v_cache.data_ptr(),
                       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                      bt_stride, (int)num_decodes, Hq, Hk, seq_offset,
+                      bt_stride, (int)num_decodes, Hq, Hk, D, seq_offset,
                       (float)scale, cur_stream());
   }
 }

@@ -291,3 +291,33 @@ def test_engine_guided_json_gpu():
                         schema=schema)
     d = json.loads(outs[0])
     assert d["label"] in ("A", "B") and 0 <= d["score"] <= 9
+
+
+def test_attn_head_dim_64():
+    """gpt-oss/embeddinggemma class models: head_dim 64 paths."""
+    require_gpu()
+    # decode
+    q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(
+        [70, 33], [1, 1], Hq=8, Hk=2, D=64)
+    scale = 1.0 / math.sqrt(64)
+    got = ops.paged_attention(q, kc, vc, bt, sl, ql, scale,
+                              num_decodes_tail=2, prefill_token_count=0)
+    ref = R.paged_attention(q.float().cpu(), kc.float().cpu(),
+                            vc.float().cpu(), bt_c, sl_c, ql_c, scale)
+    assert torch.allclose(got.float().cpu(), ref, atol=3e-2, rtol=3e-2)
+    # prefill
+    q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(
+        [50, 40], [50, 40], Hq=8, Hk=2, D=64, seed=11)
+    tiles_s, tiles_q0 = [], []
+    for s, c in enumerate([50, 40]):
+        for j in range(0, c, 32):
+            tiles_s.append(s)
+            tiles_q0.append(j)
+    got = ops.paged_attention(
+        q, kc, vc, bt, sl, ql, scale, num_decodes_tail=0,
+        tile_seq=torch.tensor(tiles_s, dtype=torch.int32, device=DEV),
+        tile_q0=torch.tensor(tiles_q0, dtype=torch.int32, device=DEV),
+        prefill_token_count=90)
+    ref = R.paged_attention(q.float().cpu(), kc.float().cpu(),
+                            vc.float().cpu(), bt_c, sl_c, ql_c, scale)
+    assert torch.allclose(got.float().cpu(), ref, atol=5e-2, rtol=5e-2)
