@@ -38,10 +38,11 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
         from ..engine.tokenizer import get_tokenizer
         from .models_map import resolve_engine_config
 
+        n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
         if device == "auto":
-            device = f"cuda:{rank}" if torch.cuda.is_available() else "cpu"
-        elif device.startswith("cuda"):
-            device = f"cuda:{rank}"
+            device = f"cuda:{rank % n_gpus}" if n_gpus else "cpu"
+        elif device.startswith("cuda") and n_gpus:
+            device = f"cuda:{rank % n_gpus}"  # oversubscribe if fewer GPUs
         if device.startswith("cuda"):
             torch.cuda.set_device(int(device.split(":")[1]))
 

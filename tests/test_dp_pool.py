@@ -98,3 +98,39 @@ def test_dp_pool_with_tp_groups(sutro_home, monkeypatch):
         assert all(o is not None for o in results["outputs"])
     finally:
         svc.shutdown()
+
+
+def test_worker_death_fails_job(sutro_home, monkeypatch):
+    """Failure detection: a dead engine worker marks in-flight jobs FAILED."""
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "2")
+    monkeypatch.delenv("SUTRO_AMD_TP", raising=False)
+    from sutro_amd.service.jobs import JobService
+
+    svc = JobService(home=sutro_home + "-fault", device="cpu",
+                     engine_kwargs={"num_kv_blocks": 128, "max_model_len": 1024})
+    try:
+        out = svc.submit_job({
+            "model": "qwen-3.5-2b",
+            "inputs": [f"long running row {i}" for i in range(40)],
+            "sampling_params": {"max_tokens": 500, "temperature": 0.9},
+        })
+        job_id = out["results"]
+        w = svc.workers["qwen-3.5-2b"]
+        time.sleep(1.0)
+        # simulate a GPU/worker crash
+        for p in w.procs:
+            p.terminate()
+        for p in w.procs:
+            p.join(timeout=30)
+        w.out_q.put(("worker_error", 0, None, "simulated worker crash"))
+        t0 = time.time()
+        while time.time() - t0 < 60:
+            st = svc.job_status(job_id)["job_status"][job_id]
+            if JobStatus.is_terminal(st):
+                break
+            time.sleep(0.2)
+        assert st == "FAILED"
+        job = svc.get_job(job_id)
+        assert "crash" in job.failure_reason["message"]
+    finally:
+        svc.shutdown()
