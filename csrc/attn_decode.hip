@@ -28,16 +28,14 @@ struct DecodeSmem {
 struct DecodeCtx {
   const u16* k_cache;
   const u16* v_cache;
-  long kv_base;     // element offset of (block, kv_head) slab, set per page
   int p_pos;        // position this lane scores (phase A): lane>>1
   int half;         // which 64-dim half: lane&1
   int lane;
   int valid;        // valid positions in current page
-  int G;
 };
 
 template <int D>
-__device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
+__device__ __forceinline__ void load_k8(u16x8 (&kreg)[D / 16], const u16* k_cache,
                                         long kv_base, int p_pos, int half) {
   const u16* krow = k_cache + kv_base + (long)p_pos * D + half * (D / 2);
 #pragma unroll
@@ -45,7 +43,7 @@ __device__ __forceinline__ void load_k8(u16x8* kreg, const u16* k_cache,
 }
 
 template <int D>
-__device__ __forceinline__ void load_v8(u16x8* vreg, const u16* v_cache,
+__device__ __forceinline__ void load_v8(u16x8 (&vreg)[D / 16], const u16* v_cache,
                                         long kv_base, int lane) {
 #pragma unroll
   for (int j = 0; j < D / 16; ++j) {
@@ -55,37 +53,44 @@ __device__ __forceinline__ void load_v8(u16x8* vreg, const u16* v_cache,
 }
 
 template <int D>
-__device__ __forceinline__ void stage_v(DecodeSmem<D>* sm, const u16x8* vreg,
-                                        int lane) {
+__device__ __forceinline__ void stage_v(DecodeSmem<D>* sm,
+                                        const u16x8 (&vreg)[D / 16], int lane) {
 #pragma unroll
   for (int j = 0; j < D / 16; ++j)
     *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = vreg[j];
 }
 
+// G is a compile-time parameter everywhere the per-head state is indexed:
+// a runtime-G loop makes these arrays dynamically indexed and hipcc places
+// them in scratch (132 B/lane measured; §5.4 rule 20).
+template <int G>
 struct SoftmaxState {
-  float m[MAXG], lsum[MAXG], acc0[MAXG], acc1[MAXG];
+  float m[G], lsum[G], acc0[G], acc1[G];
 };
 
 // phase A: scores for page from K regs + online-softmax update + P -> LDS
-template <int D>
-__device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState& st,
-                                        const u16x8* kreg, const DecodeCtx& c) {
-  float kf[D / 2];
+template <int D, int G>
+__device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState<G>& st,
+                                        const u16x8 (&kreg)[D / 16],
+                                        const DecodeCtx& c) {
+  // convert K bf16->f32 inline per use: a staged float kf[D/2] array costs 64
+  // VGPRs and pushed the kernel to 215 VGPR + scratch spill (occupancy 4.5
+  // waves/CU measured); the extra v_cvt sits in idle VALU headroom.
 #pragma unroll
-  for (int j = 0; j < D / 16; ++j) {
-#pragma unroll
-    for (int t = 0; t < 8; ++t) kf[j * 8 + t] = bf2f(kreg[j][t]);
-  }
-  for (int g = 0; g < c.G; ++g) {
+  for (int g = 0; g < G; ++g) {
     const float* qv = sm->qs + g * D + c.half * (D / 2);
     float s = 0.f;
 #pragma unroll
-    for (int j = 0; j < D / 8; ++j) {
-      f32x4 q4 = *(const f32x4*)(qv + j * 4);
-      s = fmaf(kf[j * 4 + 0], q4[0], s);
-      s = fmaf(kf[j * 4 + 1], q4[1], s);
-      s = fmaf(kf[j * 4 + 2], q4[2], s);
-      s = fmaf(kf[j * 4 + 3], q4[3], s);
+    for (int j = 0; j < D / 16; ++j) {
+      const u16x8 k8 = kreg[j];
+#pragma unroll
+      for (int t = 0; t < 8; t += 4) {
+        f32x4 q4 = *(const f32x4*)(qv + j * 8 + t);
+        s = fmaf(bf2f(k8[t + 0]), q4[0], s);
+        s = fmaf(bf2f(k8[t + 1]), q4[1], s);
+        s = fmaf(bf2f(k8[t + 2]), q4[2], s);
+        s = fmaf(bf2f(k8[t + 3]), q4[3], s);
+      }
     }
     s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
     if (c.p_pos >= c.valid) s = -1e30f;
@@ -103,8 +108,8 @@ __device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState& st,
 }
 
 // phase B: PV accumulate; lane owns D/64 dims (2 at D=128, 1 at D=64)
-template <int D>
-__device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState& st,
+template <int D, int G>
+__device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState<G>& st,
                                         const DecodeCtx& c) {
   const int dpl = D / (int)WAVE;  // dims per lane
   const int d0 = c.lane * dpl;
@@ -118,8 +123,7 @@ __device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState& st,
     }
     const float* prow = sm->p + pos;  // strided by BS per head
 #pragma unroll
-    for (int g = 0; g < MAXG; ++g) {
-      if (g >= c.G) break;
+    for (int g = 0; g < G; ++g) {
       const float pv = prow[g * BS];
       st.acc0[g] = fmaf(pv, v0, st.acc0[g]);
       st.acc1[g] = fmaf(pv, v1, st.acc1[g]);
@@ -127,7 +131,7 @@ __device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState& st,
   }
 }
 
-template <int D>
+template <int D, int G, bool PF>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
@@ -139,7 +143,6 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const long item = (long)blockIdx.x * (blockDim.x / WAVE) + wid;
-  const int G = Hq / Hk;
   if (item >= (long)n_dec * Hk) return;
   const int sd = (int)(item / Hk);
   const int kh = (int)(item - (long)sd * Hk);
@@ -151,6 +154,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   DecodeSmem<D>* sm = ((DecodeSmem<D>*)smem_raw) + wid;
 
+#pragma unroll
   for (int g = 0; g < G; ++g) {
     const u16* qrow = q + ((long)sd * Hq + kh * G + g) * D;
     for (int d = lane; d < D; d += WAVE)
@@ -159,11 +163,11 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
 
   DecodeCtx c;
   c.k_cache = k_cache; c.v_cache = v_cache;
-  c.p_pos = lane >> 1; c.half = lane & 1; c.lane = lane; c.G = G;
+  c.p_pos = lane >> 1; c.half = lane & 1; c.lane = lane;
 
-  SoftmaxState st;
+  SoftmaxState<G> st;
 #pragma unroll
-  for (int g = 0; g < MAXG; ++g) {
+  for (int g = 0; g < G; ++g) {
     st.m[g] = -1e30f; st.lsum[g] = 0.f; st.acc0[g] = 0.f; st.acc1[g] = 0.f;
   }
 
@@ -171,7 +175,7 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     return (((long)bt[pg] * Hk + kh) * BS) * D;
   };
 
-  u16x8 kA[D / 16], kB[D / 16], vbuf[D / 16];
+  u16x8 kA[D / 16], kB[PF ? D / 16 : 1], vbuf[PF ? D / 16 : 1];
   long base0 = slab(0);
   load_k8<D>(kA, k_cache, base0, c.p_pos, c.half);
 
@@ -179,20 +183,35 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
     const long base_cur = slab(pg);
     const long base_nxt = (pg + 1 < npages) ? slab(pg + 1) : base_cur;
     const bool even = (pg & 1) == 0;
-    // prefetch next K into the other register buffer
-    if (pg + 1 < npages) {
-      if (even) load_k8<D>(kB, k_cache, base_nxt, c.p_pos, c.half);
-      else      load_k8<D>(kA, k_cache, base_nxt, c.p_pos, c.half);
-    }
-    // issue V loads now; stage to LDS after phase A (latency hidden)
-    load_v8<D>(vbuf, v_cache, base_cur, lane);
     c.valid = min(BS, L - pg * BS);
-    if (even) phase_a<D>(sm, st, kA, c);
-    else      phase_a<D>(sm, st, kB, c);
-    stage_v<D>(sm, vbuf, lane);
-    phase_b<D>(sm, st, c);
+    if (PF) {
+      // software-pipelined: prefetch next K into the other register buffer,
+      // issue V loads now and stage after phase A (T14)
+      if (pg + 1 < npages) {
+        if (even) load_k8<D>((u16x8(&)[D / 16])kB, k_cache, base_nxt, c.p_pos, c.half);
+        else      load_k8<D>(kA, k_cache, base_nxt, c.p_pos, c.half);
+      }
+      load_v8<D>((u16x8(&)[D / 16])vbuf, v_cache, base_cur, lane);
+      if (even) phase_a<D, G>(sm, st, kA, c);
+      else      phase_a<D, G>(sm, st, (u16x8(&)[D / 16])kB, c);
+      stage_v<D>(sm, (u16x8(&)[D / 16])vbuf, lane);
+    } else {
+      // minimal-register serial variant: no double buffer, V staged in 2-chunk
+      // granularity to cap live registers (occupancy over pipelining)
+      if (pg > 0) load_k8<D>(kA, k_cache, base_cur, c.p_pos, c.half);
+#pragma unroll
+      for (int j = 0; j < D / 16; j += 2) {
+        u16x8 v0 = *(const u16x8*)(v_cache + base_cur + (j * (int)WAVE + lane) * 8);
+        u16x8 v1 = *(const u16x8*)(v_cache + base_cur + ((j + 1) * (int)WAVE + lane) * 8);
+        *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = v0;
+        *(u16x8*)(sm->vstage + ((j + 1) * (int)WAVE + lane) * 8) = v1;
+      }
+      phase_a<D, G>(sm, st, kA, c);
+    }
+    phase_b<D, G>(sm, st, c);
   }
 
+#pragma unroll
   for (int g = 0; g < G; ++g) {
     const float inv = 1.0f / st.lsum[g];
     if (D / (int)WAVE == 2) {
@@ -206,28 +225,57 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   }
 }
 
+template <int D>
+static void launch_decode(long blocks, int wpb, void* out, const void* q,
+                          const void* k_cache, const void* v_cache,
+                          const int* block_tables, const int* seq_lens,
+                          int bt_stride, int n_dec, int Hq, int Hk,
+                          int seq_offset, float scale, hipStream_t s) {
+  const size_t smem = sizeof(DecodeSmem<D>) * wpb;
+  const int G = Hq / Hk;
+  static const bool nopf = getenv("SUTRO_DECODE_NOPF") != nullptr;
+#define LAUNCH_G(GV)                                                          \
+  do {                                                                        \
+    if (nopf)                                                                 \
+      hipLaunchKernelGGL((attn_decode_kernel<D, GV, false>),                  \
+                         dim3((unsigned)blocks), dim3(wpb * WAVE), smem, s,   \
+                         (u16*)out, (const u16*)q, (const u16*)k_cache,       \
+                         (const u16*)v_cache, block_tables, seq_lens,         \
+                         bt_stride, n_dec, Hq, Hk, seq_offset, scale);        \
+    else                                                                      \
+      hipLaunchKernelGGL((attn_decode_kernel<D, GV, true>),                   \
+                         dim3((unsigned)blocks), dim3(wpb * WAVE), smem, s,   \
+                         (u16*)out, (const u16*)q, (const u16*)k_cache,       \
+                         (const u16*)v_cache, block_tables, seq_lens,         \
+                         bt_stride, n_dec, Hq, Hk, seq_offset, scale);        \
+  } while (0)
+  switch (G) {
+    case 1: LAUNCH_G(1); break;
+    case 2: LAUNCH_G(2); break;
+    case 3: LAUNCH_G(3); break;
+    case 4: LAUNCH_G(4); break;
+    case 5: LAUNCH_G(5); break;
+    case 6: LAUNCH_G(6); break;
+    case 7: LAUNCH_G(7); break;
+    default: LAUNCH_G(8); break;
+  }
+#undef LAUNCH_G
+}
+
 extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   const void* v_cache, const int* block_tables,
                                   const int* seq_lens, int bt_stride, int n_dec,
                                   int Hq, int Hk, int head_dim, int seq_offset,
                                   float scale, hipStream_t s) {
   if (n_dec == 0) return;
-  const int waves_per_block = 4;
+  const int wpb = 4;
   const long items = (long)n_dec * Hk;
-  const long blocks = (items + waves_per_block - 1) / waves_per_block;
+  const long blocks = (items + wpb - 1) / wpb;
   if (head_dim == 128) {
-    const size_t smem = sizeof(DecodeSmem<128>) * waves_per_block;
-    hipLaunchKernelGGL(attn_decode_kernel<128>, dim3((unsigned)blocks),
-                       dim3(waves_per_block * WAVE), smem, s, (u16*)out,
-                       (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
-                       block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
-                       seq_offset, scale);
+    launch_decode<128>(blocks, wpb, out, q, k_cache, v_cache, block_tables,
+                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
   } else {
-    const size_t smem = sizeof(DecodeSmem<64>) * waves_per_block;
-    hipLaunchKernelGGL(attn_decode_kernel<64>, dim3((unsigned)blocks),
-                       dim3(waves_per_block * WAVE), smem, s, (u16*)out,
-                       (const u16*)q, (const u16*)k_cache, (const u16*)v_cache,
-                       block_tables, seq_lens, bt_stride, n_dec, Hq, Hk,
-                       seq_offset, scale);
+    launch_decode<64>(blocks, wpb, out, q, k_cache, v_cache, block_tables,
+                      seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
   }
 }
