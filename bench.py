@@ -32,7 +32,7 @@ def build_args():
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", type=str, default="qwen-3-32b")
-    p.add_argument("--batch", type=int, default=512,
+    p.add_argument("--batch", type=int, default=1024,
                    help="rows in flight per GPU (max_num_seqs)")
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--max-new", type=int, default=128)
