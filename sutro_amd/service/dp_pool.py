@@ -293,6 +293,8 @@ class MultiProcEngineWorker:
                 if el > 0:
                     job.tokens_per_second = (
                         (job.input_tokens + job.output_tokens) / el)
+                if job.completed_rows % 256 == 0:
+                    service.persist_job(job, with_results=True)
                 if job.completed_rows == job.num_rows:
                     job.status = JobStatus.SUCCEEDED
                     job.datetime_completed = time.strftime(

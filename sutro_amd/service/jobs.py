@@ -203,6 +203,10 @@ class EngineWorker:
             el = time.time() - getattr(job, "_t_start", time.time())
             if el > 0:
                 job.tokens_per_second = (job.input_tokens + job.output_tokens) / el
+            if job.completed_rows % 256 == 0:
+                # resumable partial results: a detached client (or a restarted
+                # service) can still read completed shards
+                service.persist_job(job, with_results=True)
             if job.completed_rows == job.num_rows:
                 job.status = JobStatus.SUCCEEDED
                 job.datetime_completed = _now()
