@@ -132,7 +132,7 @@ __device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState<G>& st,
 }
 
 template <int D, int G, bool PF>
-__global__ __launch_bounds__(256) void attn_decode_kernel(
+__device__ __forceinline__ void attn_decode_body(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
     const u16* __restrict__ k_cache,  // [nb, Hk, BS, D]
@@ -225,6 +225,30 @@ __global__ __launch_bounds__(256) void attn_decode_kernel(
   }
 }
 
+template <int D, int G, bool PF>
+__global__ __launch_bounds__(256) void attn_decode_kernel(
+    u16* __restrict__ out, const u16* __restrict__ q,
+    const u16* __restrict__ k_cache, const u16* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ seq_lens,
+    int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
+  attn_decode_body<D, G, PF>(out, q, k_cache, v_cache, block_tables, seq_lens,
+                             bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+}
+
+// forced 3-waves/SIMD variant (<=168 VGPR; the allocator spills ~164 B/lane
+// of softmax state to scratch in exchange for 50% more resident waves) —
+// pick at runtime with SUTRO_DECODE_W3=1 for A/B
+template <int D, int G>
+__global__ __launch_bounds__(256, 3) void attn_decode_kernel_w3(
+    u16* __restrict__ out, const u16* __restrict__ q,
+    const u16* __restrict__ k_cache, const u16* __restrict__ v_cache,
+    const int* __restrict__ block_tables, const int* __restrict__ seq_lens,
+    int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
+  attn_decode_body<D, G, false>(out, q, k_cache, v_cache, block_tables,
+                                seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset,
+                                scale);
+}
+
 template <int D>
 static void launch_decode(long blocks, int wpb, void* out, const void* q,
                           const void* k_cache, const void* v_cache,
@@ -236,21 +260,18 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
   // minimal-register serial variant is the measured default (8451 vs 7665
   // tok/s on Qwen3-32B batch-512 decode); SUTRO_DECODE_PF=1 re-enables the
   // software-pipelined variant for A/B runs.
-  static const bool nopf = getenv("SUTRO_DECODE_PF") == nullptr;
+  static const bool pf = getenv("SUTRO_DECODE_PF") != nullptr;
+  static const bool w3 = getenv("SUTRO_DECODE_W3") != nullptr;
+#define LAUNCH_K(KERNEL)                                                      \
+  hipLaunchKernelGGL((KERNEL), dim3((unsigned)blocks), dim3(wpb * WAVE),      \
+                     smem, s, (u16*)out, (const u16*)q, (const u16*)k_cache,  \
+                     (const u16*)v_cache, block_tables, seq_lens, bt_stride,  \
+                     n_dec, Hq, Hk, seq_offset, scale)
 #define LAUNCH_G(GV)                                                          \
   do {                                                                        \
-    if (nopf)                                                                 \
-      hipLaunchKernelGGL((attn_decode_kernel<D, GV, false>),                  \
-                         dim3((unsigned)blocks), dim3(wpb * WAVE), smem, s,   \
-                         (u16*)out, (const u16*)q, (const u16*)k_cache,       \
-                         (const u16*)v_cache, block_tables, seq_lens,         \
-                         bt_stride, n_dec, Hq, Hk, seq_offset, scale);        \
-    else                                                                      \
-      hipLaunchKernelGGL((attn_decode_kernel<D, GV, true>),                   \
-                         dim3((unsigned)blocks), dim3(wpb * WAVE), smem, s,   \
-                         (u16*)out, (const u16*)q, (const u16*)k_cache,       \
-                         (const u16*)v_cache, block_tables, seq_lens,         \
-                         bt_stride, n_dec, Hq, Hk, seq_offset, scale);        \
+    if (w3)      LAUNCH_K((attn_decode_kernel_w3<D, GV>));                    \
+    else if (pf) LAUNCH_K((attn_decode_kernel<D, GV, true>));                 \
+    else         LAUNCH_K((attn_decode_kernel<D, GV, false>));                \
   } while (0)
   switch (G) {
     case 1: LAUNCH_G(1); break;
@@ -263,6 +284,7 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
     default: LAUNCH_G(8); break;
   }
 #undef LAUNCH_G
+#undef LAUNCH_K
 }
 
 extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
