@@ -46,6 +46,9 @@ def build_args():
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree (ranks per engine replica; "
                         "dp = world_size // tp)")
+    p.add_argument("--schema", action="store_true",
+                   help="FSM-guided structured JSON extraction workload "
+                        "(BASELINE 1M-row p1 config shape)")
     return p.parse_args()
 
 
@@ -112,6 +115,18 @@ def main():
         return [1] + body  # BOS + random bytes
 
     sp_kwargs = dict(max_tokens=args.max_new, temperature=0.8, top_p=0.95)
+    fsm_id = None
+    if args.schema:
+        schema = {"type": "object", "properties": {
+            "name": {"type": "string", "maxLength": 48},
+            "category": {"enum": ["news", "review", "spam", "other"]},
+            "sentiment": {"enum": ["positive", "negative", "neutral"]},
+            "score": {"type": "integer", "minimum": 0, "maximum": 100},
+            "tags": {"type": "array", "items": {"type": "string",
+                                                "maxLength": 16},
+                     "minItems": 1, "maxItems": 4}}}
+        fsm_id = eng.register_fsm(schema)
+        sp_kwargs = dict(max_tokens=max(args.max_new, 512), temperature=0.9)
     arrival = [0]
 
     def refill():
@@ -119,7 +134,7 @@ def main():
             return
         while len(eng.scheduler.running) + len(eng.scheduler.waiting_p0) < args.batch:
             eng.add_request(make_prompt(), SamplingParams(**sp_kwargs),
-                            arrival_idx=arrival[0])
+                            fsm_id=fsm_id, arrival_idx=arrival[0])
             arrival[0] += 1
 
     refill()
