@@ -29,6 +29,8 @@ class EngineConfig:
     num_kv_blocks: Optional[int] = None  # None = derive from free memory
     default_max_new_tokens: int = 256
     seed: int = 0
+    # optional safetensors checkpoint dir; None = deterministic random init
+    weights_path: Optional[str] = None
     enforce_eager: bool = False      # disable hipGraph capture of the decode step
     # tensor parallelism (process group set up by the caller)
     tp_size: int = 1

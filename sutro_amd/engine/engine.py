@@ -75,7 +75,12 @@ class LLMEngine:
             with torch.device(cfg.device):
                 model = Qwen3Model(self.spec, cfg.dtype, cfg.max_model_len,
                                    self.tp, moe_ep=cfg.moe_ep)
-            model.init_random_weights(cfg.seed)
+            if cfg.weights_path:
+                from ..models.loader import load_weights
+
+                load_weights(model, cfg.weights_path)
+            else:
+                model.init_random_weights(cfg.seed)
         self.model = model.to(cfg.device).eval()
 
         num_blocks = cfg.num_kv_blocks
