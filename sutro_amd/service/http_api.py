@@ -186,6 +186,36 @@ def create_app(home: Optional[str] = None, device: str = "auto",
         payload = await request.json()
         return functions.create(payload)
 
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus text exposition of service/job counters."""
+        from prometheus_client import (CONTENT_TYPE_LATEST, CollectorRegistry,
+                                       Gauge, generate_latest)
+
+        reg = CollectorRegistry()
+        by_status: Dict[str, int] = {}
+        in_tok = out_tok = rows_done = rows_total = 0
+        for job in service.jobs.values():
+            by_status[job.status.value] = by_status.get(job.status.value, 0) + 1
+            in_tok += job.input_tokens
+            out_tok += job.output_tokens
+            rows_done += job.completed_rows
+            rows_total += job.num_rows
+        g = Gauge("sutro_jobs", "jobs by status", ["status"], registry=reg)
+        for st, n in by_status.items():
+            g.labels(status=st).set(n)
+        Gauge("sutro_input_tokens_total", "input tokens across jobs",
+              registry=reg).set(in_tok)
+        Gauge("sutro_output_tokens_total", "output tokens across jobs",
+              registry=reg).set(out_tok)
+        Gauge("sutro_rows_completed_total", "completed rows",
+              registry=reg).set(rows_done)
+        Gauge("sutro_rows_total", "submitted rows", registry=reg).set(rows_total)
+        Gauge("sutro_engine_workers", "live engine workers",
+              registry=reg).set(len(service.workers))
+        return Response(content=generate_latest(reg),
+                        media_type=CONTENT_TYPE_LATEST)
+
     return app
 
 

@@ -166,3 +166,15 @@ def test_sdk_against_http_server(client, sutro_home, monkeypatch):
                       sampling_params={"max_tokens": 4})
     df = so.await_job_completion(job_id)
     assert len(df) == 2
+
+
+def test_metrics_endpoint(client):
+    job_id = _submit(client, {"model": "qwen-3.5-2b", "inputs": ["m1", "m2"],
+                              "sampling_params": {"max_tokens": 4}})
+    _await(client, job_id)
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    body = r.text
+    assert "sutro_jobs" in body
+    assert "sutro_output_tokens_total" in body
+    assert 'status="SUCCEEDED"' in body
