@@ -10,7 +10,7 @@ from __future__ import annotations
 
 __version__ = "0.1.0"
 
-from .interfaces import JobStatus  # noqa: F401
+from .interfaces import JobStatus  # noqa: F401 (public API)
 from .sdk import Sutro  # noqa: F401
 
 _instance = Sutro()

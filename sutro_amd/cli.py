@@ -8,7 +8,6 @@ set-base-url, serve, bench-info.
 from __future__ import annotations
 
 import json
-import sys
 from typing import Optional
 
 import click

@@ -8,7 +8,6 @@ from a worker thread/process; `bench.py` drives it directly.
 
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
 

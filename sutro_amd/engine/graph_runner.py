@@ -14,7 +14,6 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional
 
-import numpy as np
 import torch
 
 from .batch import ForwardBatch, ScheduledBatch

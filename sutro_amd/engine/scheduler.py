@@ -16,7 +16,7 @@ Invariants:
 from __future__ import annotations
 
 from collections import deque
-from typing import Deque, Dict, List, Optional
+from typing import Deque, Dict, List
 
 from .batch import ScheduledBatch
 from .config import EngineConfig

@@ -14,9 +14,8 @@ accept split q/k/v and gate/up tensors.
 
 from __future__ import annotations
 
-import json
 import os
-from typing import Dict, Iterator, Optional, Tuple
+from typing import Dict, Iterator, Tuple
 
 import torch
 

@@ -9,7 +9,7 @@ hand-written HIP kernel on GPU (see sutro_amd/ops).
 from __future__ import annotations
 
 import math
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

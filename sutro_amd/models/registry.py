@@ -13,8 +13,8 @@ HIP kernels covers the whole registry.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field, replace
-from typing import Dict, Optional
+from dataclasses import dataclass, replace
+from typing import Dict
 
 
 @dataclass(frozen=True)

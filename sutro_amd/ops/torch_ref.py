@@ -7,8 +7,6 @@ All reductions accumulate in fp32.
 
 from __future__ import annotations
 
-from typing import List
-
 import torch
 
 

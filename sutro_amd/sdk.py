@@ -27,7 +27,6 @@ from .common import (
     DEFAULT_MODEL,
     ModelOptions,
     fancy_tqdm,
-    make_clickable_link,
     normalize_output_schema,
     prepare_input_data,
     to_colored_text,

@@ -9,11 +9,10 @@ cancel, auth, quotas, datasets, Functions.
 
 from __future__ import annotations
 
-import io
 import json
-from typing import Any, Dict, Optional
+from typing import Dict, Optional
 
-from fastapi import FastAPI, Header, HTTPException, Request, Response, UploadFile
+from fastapi import FastAPI, Header, HTTPException, Request, Response
 from fastapi.responses import StreamingResponse
 
 

@@ -8,7 +8,6 @@ completion, and optional scratchpad stripping.
 
 from __future__ import annotations
 
-import json
 from typing import Dict, List, Optional, Union
 
 import pandas as pd

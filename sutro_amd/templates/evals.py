@@ -12,7 +12,7 @@ from __future__ import annotations
 
 import json
 from itertools import combinations
-from typing import Dict, List, Optional, Tuple, Union
+from typing import List, Optional, Tuple, Union
 
 import numpy as np
 import pandas as pd
