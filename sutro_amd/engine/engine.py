@@ -333,6 +333,15 @@ class LLMEngine:
         req.output_token_ids.append(tok)
         req.cumulative_logprob += lp
         stats.output_tokens += 1
+        if sp.stop:
+            # byte tokenizer: token == byte, so stop strings are byte suffixes
+            out = self.tokenizer.decode(req.output_token_ids[-64:])
+            for s in sp.stop:
+                if s and out.endswith(s):
+                    del req.output_token_ids[len(req.output_token_ids) - len(s.encode()):]
+                    self.scheduler.finish(req, FinishReason.STOP)
+                    stats.finished.append(req)
+                    return
         if req.fsm_id is not None:
             fsm = self._fsms[req.fsm_id]
             req.fsm_state = fsm.advance(req.fsm_state, tok)

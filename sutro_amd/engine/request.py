@@ -15,6 +15,7 @@ class SamplingParams:
     max_tokens: int = 256
     seed: Optional[int] = None
     stop_token_ids: Optional[List[int]] = None
+    stop: Optional[List[str]] = None  # stop strings (trimmed from the output)
     logprobs: bool = True     # accumulate cumulative logprob of sampled tokens
 
     @classmethod
@@ -23,6 +24,8 @@ class SamplingParams:
         # accept common aliases the reference's payload passes through verbatim
         if "max_new_tokens" in d and "max_tokens" not in d:
             d["max_tokens"] = d.pop("max_new_tokens")
+        if isinstance(d.get("stop"), str):
+            d["stop"] = [d["stop"]]
         known = {f for f in cls.__dataclass_fields__}  # type: ignore[attr-defined]
         kwargs = {k: v for k, v in d.items() if k in known}
         sp = cls(**kwargs)

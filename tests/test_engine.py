@@ -198,3 +198,25 @@ def test_moe_model_generates():
     outs = eng.generate(["moe row one", "moe row two"],
                         sampling=SamplingParams(max_tokens=8, temperature=0.5))
     assert len(outs) == 2 and eng.total_output_tokens > 0
+
+
+def test_stop_strings():
+    """Generation halts on a stop string; the stop text is trimmed."""
+    eng = _engine()
+    # force a deterministic output, then use its own prefix as the stop string
+    probe = eng.add_request(eng.tokenizer.encode("stop test"),
+                            SamplingParams(max_tokens=12, temperature=0))
+    while eng.has_work():
+        eng.step()
+    full = eng.tokenizer.decode(probe.output_token_ids)
+    assert len(full) >= 4
+    stop = full[2:5]
+    eng2 = _engine()
+    req = eng2.add_request(eng2.tokenizer.encode("stop test"),
+                           SamplingParams(max_tokens=12, temperature=0,
+                                          stop=[stop]))
+    while eng2.has_work():
+        eng2.step()
+    out = eng2.tokenizer.decode(req.output_token_ids)
+    assert stop not in out
+    assert len(req.output_token_ids) < 12
