@@ -36,7 +36,7 @@ def build_args():
                    help="rows in flight per GPU (max_num_seqs)")
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--max-new", type=int, default=128)
-    p.add_argument("--tokens-per-step", type=int, default=16384)
+    p.add_argument("--tokens-per-step", type=int, default=32768)
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
     p.add_argument("--no-refill", action="store_true",

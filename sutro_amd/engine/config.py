@@ -20,7 +20,7 @@ class EngineConfig:
     kv_block_size: int = KV_BLOCK_SIZE
     max_model_len: int = 8192        # scheduler cap on prompt+output length
     max_num_seqs: int = 1024         # max concurrently running sequences
-    max_tokens_per_step: int = 16384  # token budget per scheduler step (prefill chunking)
+    max_tokens_per_step: int = 32768  # token budget per scheduler step (prefill chunking)
     # throughput policy: while decodes are running, hold back new prefills
     # until this many prompt tokens have accumulated (amortizes the eager
     # prefill pass; hipGraph decode steps stay pure). 0 = admit eagerly.
