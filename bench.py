@@ -46,6 +46,9 @@ def build_args():
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree (ranks per engine replica; "
                         "dp = world_size // tp)")
+    p.add_argument("--kv-fp8", action="store_true",
+                   help="OCP e4m3 KV cache (halves KV bytes; labeled variant, "
+                        "not the default bf16 measurement)")
     p.add_argument("--schema", action="store_true",
                    help="FSM-guided structured JSON extraction workload "
                         "(BASELINE 1M-row p1 config shape)")
@@ -101,6 +104,7 @@ def main():
         num_kv_blocks=args.kv_blocks if device != "cpu" else 512,
         seed=dp_idx,  # identical within a TP group (lockstep), unique per replica
         tp_size=tp,
+        kv_dtype="fp8_e4m3" if args.kv_fp8 else "bf16",
     )
     t_init0 = time.time()
     eng = LLMEngine(cfg)
@@ -207,6 +211,8 @@ def main():
             "scaling": "weak",
             "vs_baseline": None,
             "dtype": "bf16" if device != "cpu" else "fp32",
+            "kv_dtype": "fp8_e4m3" if args.kv_fp8 else
+                        ("bf16" if device != "cpu" else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": spec.name,

@@ -18,46 +18,54 @@
 #define BS 32       // KV page size (tokens) == EngineConfig.kv_block_size
 #define MAXG 8      // max grouped q-heads per kv head handled per wave
 
-template <int D>
+template <int D, bool KV8>
 struct DecodeSmem {
-  float qs[MAXG * D];   // Q rows, pre-scaled, f32
-  u16 vstage[BS * D];   // V page, row-major bf16
-  float p[MAXG * BS];   // softmax weights for current page
+  float qs[MAXG * D];                  // Q rows, pre-scaled, f32
+  u8 vstage[BS * D * (KV8 ? 1 : 2)];   // V page, row-major bf16 or e4m3
+  float p[MAXG * BS];                  // softmax weights for current page
+};
+
+// per-lane K/V register image: 16-byte chunks of the lane's share
+template <int D, bool KV8>
+struct KVBuf {
+  static constexpr int N = KV8 ? D / 32 : D / 16;  // chunks per half-row
+  u32x4 v[N];
 };
 
 struct DecodeCtx {
-  const u16* k_cache;
-  const u16* v_cache;
   int p_pos;        // position this lane scores (phase A): lane>>1
   int half;         // which 64-dim half: lane&1
   int lane;
   int valid;        // valid positions in current page
 };
 
-template <int D>
-__device__ __forceinline__ void load_k8(u16x8 (&kreg)[D / 16], const u16* k_cache,
+template <int D, bool KV8>
+__device__ __forceinline__ void load_k8(KVBuf<D, KV8>& kreg, const u8* k_cache,
                                         long kv_base, int p_pos, int half) {
-  const u16* krow = k_cache + kv_base + (long)p_pos * D + half * (D / 2);
+  // kv_base/offsets in ELEMENTS; element size is 2 (bf16) or 1 (e4m3) bytes
+  constexpr int ES = KV8 ? 1 : 2;
+  const u8* krow = k_cache + ((long)kv_base + (long)p_pos * D + half * (D / 2)) * ES;
 #pragma unroll
-  for (int j = 0; j < D / 16; ++j) kreg[j] = *(const u16x8*)(krow + j * 8);
+  for (int j = 0; j < KVBuf<D, KV8>::N; ++j)
+    kreg.v[j] = *(const u32x4*)(krow + j * 16);
 }
 
-template <int D>
-__device__ __forceinline__ void load_v8(u16x8 (&vreg)[D / 16], const u16* v_cache,
+template <int D, bool KV8>
+__device__ __forceinline__ void load_v8(KVBuf<D, KV8>& vreg, const u8* v_cache,
                                         long kv_base, int lane) {
+  constexpr int ES = KV8 ? 1 : 2;
+  const u8* base = v_cache + kv_base * ES;
 #pragma unroll
-  for (int j = 0; j < D / 16; ++j) {
-    const int off = (j * (int)WAVE + lane) * 8;
-    vreg[j] = *(const u16x8*)(v_cache + kv_base + off);
-  }
+  for (int j = 0; j < KVBuf<D, KV8>::N; ++j)
+    vreg.v[j] = *(const u32x4*)(base + (j * (int)WAVE + lane) * 16);
 }
 
-template <int D>
-__device__ __forceinline__ void stage_v(DecodeSmem<D>* sm,
-                                        const u16x8 (&vreg)[D / 16], int lane) {
+template <int D, bool KV8>
+__device__ __forceinline__ void stage_v(DecodeSmem<D, KV8>* sm,
+                                        const KVBuf<D, KV8>& vreg, int lane) {
 #pragma unroll
-  for (int j = 0; j < D / 16; ++j)
-    *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = vreg[j];
+  for (int j = 0; j < KVBuf<D, KV8>::N; ++j)
+    *(u32x4*)(sm->vstage + (j * (int)WAVE + lane) * 16) = vreg.v[j];
 }
 
 // G is a compile-time parameter everywhere the per-head state is indexed:
@@ -69,27 +77,44 @@ struct SoftmaxState {
 };
 
 // phase A: scores for page from K regs + online-softmax update + P -> LDS
-template <int D, int G>
-__device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState<G>& st,
-                                        const u16x8 (&kreg)[D / 16],
+template <int D, int G, bool KV8>
+__device__ __forceinline__ void phase_a(DecodeSmem<D, KV8>* sm,
+                                        SoftmaxState<G>& st,
+                                        const KVBuf<D, KV8>& kreg,
                                         const DecodeCtx& c) {
-  // convert K bf16->f32 inline per use: a staged float kf[D/2] array costs 64
-  // VGPRs and pushed the kernel to 215 VGPR + scratch spill (occupancy 4.5
-  // waves/CU measured); the extra v_cvt sits in idle VALU headroom.
+  // g-outer with inline K decode per use: a staged float kf[D/2] image (or a
+  // j-outer shared-decode order) pushes the kernel past 215 VGPR into scratch
+  // spill (occupancy 4.5 waves/CU measured; §5.4 rule 20). Decoding K
+  // redundantly per head sits in idle VALU headroom (VALUBusy 15% measured).
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     const float* qv = sm->qs + g * D + c.half * (D / 2);
     float s = 0.f;
 #pragma unroll
-    for (int j = 0; j < D / 16; ++j) {
-      const u16x8 k8 = kreg[j];
+    for (int e = 0; e < KVBuf<D, KV8>::N; ++e) {
+      const u32x4 entry = kreg.v[e];
+      if constexpr (KV8) {
+        float kf[16];  // 16 e4m3 codes per 16B chunk
+        fp8x4_to_f32(entry[0], kf);
+        fp8x4_to_f32(entry[1], kf + 4);
+        fp8x4_to_f32(entry[2], kf + 8);
+        fp8x4_to_f32(entry[3], kf + 12);
 #pragma unroll
-      for (int t = 0; t < 8; t += 4) {
-        f32x4 q4 = *(const f32x4*)(qv + j * 8 + t);
-        s = fmaf(bf2f(k8[t + 0]), q4[0], s);
-        s = fmaf(bf2f(k8[t + 1]), q4[1], s);
-        s = fmaf(bf2f(k8[t + 2]), q4[2], s);
-        s = fmaf(bf2f(k8[t + 3]), q4[3], s);
+        for (int t = 0; t < 16; t += 4) {
+          f32x4 q4 = *(const f32x4*)(qv + e * 16 + t);
+          s = fmaf(kf[t + 0], q4[0], s);
+          s = fmaf(kf[t + 1], q4[1], s);
+          s = fmaf(kf[t + 2], q4[2], s);
+          s = fmaf(kf[t + 3], q4[3], s);
+        }
+      } else {
+#pragma unroll
+        for (int t = 0; t < 4; ++t) {  // 8 bf16 codes per 16B chunk
+          const float k0 = bf2f((u16)(entry[t] & 0xFFFF));
+          const float k1 = bf2f((u16)(entry[t] >> 16));
+          f32x2 q2 = *(const f32x2*)(qv + e * 8 + t * 2);
+          s = fmaf(k0, q2[0], fmaf(k1, q2[1], s));
+        }
       }
     }
     s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
@@ -108,18 +133,30 @@ __device__ __forceinline__ void phase_a(DecodeSmem<D>* sm, SoftmaxState<G>& st,
 }
 
 // phase B: PV accumulate; lane owns D/64 dims (2 at D=128, 1 at D=64)
-template <int D, int G>
-__device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState<G>& st,
+template <int D, int G, bool KV8>
+__device__ __forceinline__ void phase_b(DecodeSmem<D, KV8>* sm,
+                                        SoftmaxState<G>& st,
                                         const DecodeCtx& c) {
-  const int dpl = D / (int)WAVE;  // dims per lane
+  constexpr int dpl = D / (int)WAVE;  // dims per lane
+  constexpr int ES = KV8 ? 1 : 2;
   const int d0 = c.lane * dpl;
   for (int pos = 0; pos < c.valid; ++pos) {
     float v0, v1;
-    if (dpl == 2) {
-      u16x2 v2 = *(const u16x2*)(sm->vstage + pos * D + d0);
-      v0 = bf2f(v2[0]); v1 = bf2f(v2[1]);
+    const u8* vrow = sm->vstage + (pos * D + d0) * ES;
+    if constexpr (KV8) {
+      if (dpl == 2) {
+        f32x2 vv = fp8x2_to_f32(*(const u16*)vrow);
+        v0 = vv[0]; v1 = vv[1];
+      } else {
+        v0 = fp8_to_f32(*vrow); v1 = 0.f;
+      }
     } else {
-      v0 = bf2f(sm->vstage[pos * D + d0]); v1 = 0.f;
+      if (dpl == 2) {
+        u16x2 v2 = *(const u16x2*)vrow;
+        v0 = bf2f(v2[0]); v1 = bf2f(v2[1]);
+      } else {
+        v0 = bf2f(*(const u16*)vrow); v1 = 0.f;
+      }
     }
     const float* prow = sm->p + pos;  // strided by BS per head
 #pragma unroll
@@ -131,12 +168,12 @@ __device__ __forceinline__ void phase_b(DecodeSmem<D>* sm, SoftmaxState<G>& st,
   }
 }
 
-template <int D, int G, bool PF>
+template <int D, int G, bool PF, bool KV8>
 __device__ __forceinline__ void attn_decode_body(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
-    const u16* __restrict__ k_cache,  // [nb, Hk, BS, D]
-    const u16* __restrict__ v_cache,
+    const u8* __restrict__ k_cache,   // [nb, Hk, BS, D] bf16 or e4m3
+    const u8* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, bt_stride]
     const int* __restrict__ seq_lens,      // [S]
     int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
@@ -152,7 +189,7 @@ __device__ __forceinline__ void attn_decode_body(
   const int* bt = block_tables + (long)sg * bt_stride;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  DecodeSmem<D>* sm = ((DecodeSmem<D>*)smem_raw) + wid;
+  DecodeSmem<D, KV8>* sm = ((DecodeSmem<D, KV8>*)smem_raw) + wid;
 
 #pragma unroll
   for (int g = 0; g < G; ++g) {
@@ -162,7 +199,6 @@ __device__ __forceinline__ void attn_decode_body(
   }
 
   DecodeCtx c;
-  c.k_cache = k_cache; c.v_cache = v_cache;
   c.p_pos = lane >> 1; c.half = lane & 1; c.lane = lane;
 
   SoftmaxState<G> st;
@@ -175,9 +211,9 @@ __device__ __forceinline__ void attn_decode_body(
     return (((long)bt[pg] * Hk + kh) * BS) * D;
   };
 
-  u16x8 kA[D / 16], kB[PF ? D / 16 : 1], vbuf[PF ? D / 16 : 1];
+  KVBuf<D, KV8> kA, kB, vbuf;
   long base0 = slab(0);
-  load_k8<D>(kA, k_cache, base0, c.p_pos, c.half);
+  load_k8<D, KV8>(kA, k_cache, base0, c.p_pos, c.half);
 
   for (int pg = 0; pg < npages; ++pg) {
     const long base_cur = slab(pg);
@@ -188,27 +224,32 @@ __device__ __forceinline__ void attn_decode_body(
       // software-pipelined: prefetch next K into the other register buffer,
       // issue V loads now and stage after phase A (T14)
       if (pg + 1 < npages) {
-        if (even) load_k8<D>((u16x8(&)[D / 16])kB, k_cache, base_nxt, c.p_pos, c.half);
-        else      load_k8<D>(kA, k_cache, base_nxt, c.p_pos, c.half);
+        if (even) load_k8<D, KV8>(kB, k_cache, base_nxt, c.p_pos, c.half);
+        else      load_k8<D, KV8>(kA, k_cache, base_nxt, c.p_pos, c.half);
       }
-      load_v8<D>((u16x8(&)[D / 16])vbuf, v_cache, base_cur, lane);
-      if (even) phase_a<D, G>(sm, st, kA, c);
-      else      phase_a<D, G>(sm, st, (u16x8(&)[D / 16])kB, c);
-      stage_v<D>(sm, (u16x8(&)[D / 16])vbuf, lane);
+      load_v8<D, KV8>(vbuf, v_cache, base_cur, lane);
+      if (even) phase_a<D, G, KV8>(sm, st, kA, c);
+      else      phase_a<D, G, KV8>(sm, st, kB, c);
+      stage_v<D, KV8>(sm, vbuf, lane);
     } else {
       // minimal-register serial variant: no double buffer, V staged in 2-chunk
       // granularity to cap live registers (occupancy over pipelining)
-      if (pg > 0) load_k8<D>(kA, k_cache, base_cur, c.p_pos, c.half);
+      constexpr int ES = KV8 ? 1 : 2;
+      if (pg > 0) load_k8<D, KV8>(kA, k_cache, base_cur, c.p_pos, c.half);
+      const u8* vsrc = v_cache + base_cur * ES;
 #pragma unroll
-      for (int j = 0; j < D / 16; j += 2) {
-        u16x8 v0 = *(const u16x8*)(v_cache + base_cur + (j * (int)WAVE + lane) * 8);
-        u16x8 v1 = *(const u16x8*)(v_cache + base_cur + ((j + 1) * (int)WAVE + lane) * 8);
-        *(u16x8*)(sm->vstage + (j * (int)WAVE + lane) * 8) = v0;
-        *(u16x8*)(sm->vstage + ((j + 1) * (int)WAVE + lane) * 8) = v1;
+      for (int j = 0; j < KVBuf<D, KV8>::N; j += 2) {
+        u32x4 v0 = *(const u32x4*)(vsrc + (j * (int)WAVE + lane) * 16);
+        u32x4 v1 = (j + 1 < KVBuf<D, KV8>::N)
+            ? *(const u32x4*)(vsrc + ((j + 1) * (int)WAVE + lane) * 16)
+            : u32x4{};
+        *(u32x4*)(sm->vstage + (j * (int)WAVE + lane) * 16) = v0;
+        if (j + 1 < KVBuf<D, KV8>::N)
+          *(u32x4*)(sm->vstage + ((j + 1) * (int)WAVE + lane) * 16) = v1;
       }
-      phase_a<D, G>(sm, st, kA, c);
+      phase_a<D, G, KV8>(sm, st, kA, c);
     }
-    phase_b<D, G>(sm, st, c);
+    phase_b<D, G, KV8>(sm, st, c);
   }
 
 #pragma unroll
@@ -225,37 +266,38 @@ __device__ __forceinline__ void attn_decode_body(
   }
 }
 
-template <int D, int G, bool PF>
+template <int D, int G, bool PF, bool KV8>
 __global__ __launch_bounds__(256) void attn_decode_kernel(
     u16* __restrict__ out, const u16* __restrict__ q,
-    const u16* __restrict__ k_cache, const u16* __restrict__ v_cache,
+    const u8* __restrict__ k_cache, const u8* __restrict__ v_cache,
     const int* __restrict__ block_tables, const int* __restrict__ seq_lens,
     int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
-  attn_decode_body<D, G, PF>(out, q, k_cache, v_cache, block_tables, seq_lens,
-                             bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+  attn_decode_body<D, G, PF, KV8>(out, q, k_cache, v_cache, block_tables,
+                                  seq_lens, bt_stride, n_dec, Hq, Hk,
+                                  seq_offset, scale);
 }
 
 // forced 3-waves/SIMD variant (<=168 VGPR; the allocator spills ~164 B/lane
 // of softmax state to scratch in exchange for 50% more resident waves) —
 // pick at runtime with SUTRO_DECODE_W3=1 for A/B
-template <int D, int G>
+template <int D, int G, bool KV8>
 __global__ __launch_bounds__(256, 3) void attn_decode_kernel_w3(
     u16* __restrict__ out, const u16* __restrict__ q,
-    const u16* __restrict__ k_cache, const u16* __restrict__ v_cache,
+    const u8* __restrict__ k_cache, const u8* __restrict__ v_cache,
     const int* __restrict__ block_tables, const int* __restrict__ seq_lens,
     int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
-  attn_decode_body<D, G, false>(out, q, k_cache, v_cache, block_tables,
-                                seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset,
-                                scale);
+  attn_decode_body<D, G, false, KV8>(out, q, k_cache, v_cache, block_tables,
+                                     seq_lens, bt_stride, n_dec, Hq, Hk,
+                                     seq_offset, scale);
 }
 
-template <int D>
+template <int D, bool KV8>
 static void launch_decode(long blocks, int wpb, void* out, const void* q,
                           const void* k_cache, const void* v_cache,
                           const int* block_tables, const int* seq_lens,
                           int bt_stride, int n_dec, int Hq, int Hk,
                           int seq_offset, float scale, hipStream_t s) {
-  const size_t smem = sizeof(DecodeSmem<D>) * wpb;
+  const size_t smem = sizeof(DecodeSmem<D, KV8>) * wpb;
   const int G = Hq / Hk;
   // minimal-register serial variant is the measured default (8451 vs 7665
   // tok/s on Qwen3-32B batch-512 decode); SUTRO_DECODE_PF=1 re-enables the
@@ -264,14 +306,14 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
   static const bool w3 = getenv("SUTRO_DECODE_W3") != nullptr;
 #define LAUNCH_K(KERNEL)                                                      \
   hipLaunchKernelGGL((KERNEL), dim3((unsigned)blocks), dim3(wpb * WAVE),      \
-                     smem, s, (u16*)out, (const u16*)q, (const u16*)k_cache,  \
-                     (const u16*)v_cache, block_tables, seq_lens, bt_stride,  \
+                     smem, s, (u16*)out, (const u16*)q, (const u8*)k_cache,   \
+                     (const u8*)v_cache, block_tables, seq_lens, bt_stride,   \
                      n_dec, Hq, Hk, seq_offset, scale)
 #define LAUNCH_G(GV)                                                          \
   do {                                                                        \
-    if (w3)      LAUNCH_K((attn_decode_kernel_w3<D, GV>));                    \
-    else if (pf) LAUNCH_K((attn_decode_kernel<D, GV, true>));                 \
-    else         LAUNCH_K((attn_decode_kernel<D, GV, false>));                \
+    if (w3)      LAUNCH_K((attn_decode_kernel_w3<D, GV, KV8>));               \
+    else if (pf) LAUNCH_K((attn_decode_kernel<D, GV, true, KV8>));            \
+    else         LAUNCH_K((attn_decode_kernel<D, GV, false, KV8>));           \
   } while (0)
   switch (G) {
     case 1: LAUNCH_G(1); break;
@@ -290,17 +332,29 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
 extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   const void* v_cache, const int* block_tables,
                                   const int* seq_lens, int bt_stride, int n_dec,
-                                  int Hq, int Hk, int head_dim, int seq_offset,
-                                  float scale, hipStream_t s) {
+                                  int Hq, int Hk, int head_dim, int kv_fp8,
+                                  int seq_offset, float scale, hipStream_t s) {
   if (n_dec == 0) return;
   const int wpb = 4;
   const long items = (long)n_dec * Hk;
   const long blocks = (items + wpb - 1) / wpb;
   if (head_dim == 128) {
-    launch_decode<128>(blocks, wpb, out, q, k_cache, v_cache, block_tables,
-                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
+    if (kv_fp8)
+      launch_decode<128, true>(blocks, wpb, out, q, k_cache, v_cache,
+                               block_tables, seq_lens, bt_stride, n_dec, Hq,
+                               Hk, seq_offset, scale, s);
+    else
+      launch_decode<128, false>(blocks, wpb, out, q, k_cache, v_cache,
+                                block_tables, seq_lens, bt_stride, n_dec, Hq,
+                                Hk, seq_offset, scale, s);
   } else {
-    launch_decode<64>(blocks, wpb, out, q, k_cache, v_cache, block_tables,
-                      seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
+    if (kv_fp8)
+      launch_decode<64, true>(blocks, wpb, out, q, k_cache, v_cache,
+                              block_tables, seq_lens, bt_stride, n_dec, Hq,
+                              Hk, seq_offset, scale, s);
+    else
+      launch_decode<64, false>(blocks, wpb, out, q, k_cache, v_cache,
+                               block_tables, seq_lens, bt_stride, n_dec, Hq,
+                               Hk, seq_offset, scale, s);
   }
 }

@@ -48,12 +48,12 @@ struct PrefillSmem {
 
 #define PWAVE_ELEMS (QTILE * VT_PAD)  // P tile per wave (bf16)
 
-template <int D>
+template <int D, bool KV8>
 __global__ void attn_prefill_kernel(
     u16* __restrict__ out,            // [T, Hq, D]
     const u16* __restrict__ q,        // [T, Hq, D]
-    const u16* __restrict__ k_cache,  // [nb, Hk, BS, D]
-    const u16* __restrict__ v_cache,
+    const u8* __restrict__ k_cache,   // [nb, Hk, BS, D] bf16 or e4m3
+    const u8* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, bt_stride]
     const int* __restrict__ seq_lens,      // [S]
     const int* __restrict__ qlocs,         // [S+1]
@@ -106,23 +106,47 @@ __global__ void attn_prefill_kernel(
 
   for (int kt = 0; kt < ntiles_kv; ++kt) {
     const int blk = block_tables[(long)s * bt_stride + kt];
-    const long kv_base = (((long)blk * Hk + kh) * BS) * D;
+    const long kv_base = (((long)blk * Hk + kh) * BS) * D;  // in elements
     const int kv0 = kt * BS;
-    constexpr int SLOTS = D / 8;  // 16B chunks per row
+    constexpr int SLOTS = D / 8;   // 8-element chunks per row
+    constexpr int ES = KV8 ? 1 : 2;
 
-    // ---- cooperative stage: K (swizzled) and V^T into LDS
+    // ---- cooperative stage: K (swizzled) and V^T into LDS, as bf16
+    // (fp8 caches are dequantized here so the MFMA path is unchanged)
     {
       const int tid = threadIdx.x, nthr = blockDim.x;
-      // K: 32 rows x 2D bytes; thread moves 16B: item = row*SLOTS + slot
+      // K: 32 rows; thread moves one 8-element chunk: item = row*SLOTS + slot
       for (int it = tid; it < BS * SLOTS; it += nthr) {
         const int row = it / SLOTS, slot = it % SLOTS;
-        u16x8 kx = *(const u16x8*)(k_cache + kv_base + row * D + slot * 8);
+        const u8* src = k_cache + ((long)kv_base + row * D + slot * 8) * ES;
+        u16x8 kx;
+        if constexpr (KV8) {
+          const u32x2 raw = *(const u32x2*)src;
+          float f[8];
+          fp8x4_to_f32(raw[0], f);
+          fp8x4_to_f32(raw[1], f + 4);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kx[j] = f2bf(f[j]);
+        } else {
+          kx = *(const u16x8*)src;
+        }
         *(u16x8*)((char*)sm->ktile + k_swz<D>(row, slot * 16)) = kx;
       }
       // V^T: read V[kv][d0..d0+8), write 8 u16 at vt[d][kv]
       for (int it = tid; it < BS * SLOTS; it += nthr) {
         const int kv = it / SLOTS, d0 = (it % SLOTS) * 8;
-        u16x8 vx = *(const u16x8*)(v_cache + kv_base + kv * D + d0);
+        const u8* src = v_cache + ((long)kv_base + kv * D + d0) * ES;
+        u16x8 vx;
+        if constexpr (KV8) {
+          const u32x2 raw = *(const u32x2*)src;
+          float f[8];
+          fp8x4_to_f32(raw[0], f);
+          fp8x4_to_f32(raw[1], f + 4);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vx[j] = f2bf(f[j]);
+        } else {
+          vx = *(const u16x8*)src;
+        }
 #pragma unroll
         for (int j = 0; j < 8; ++j) sm->vt[(d0 + j) * VT_PAD + kv] = vx[j];
       }
@@ -228,28 +252,27 @@ extern "C" void sutro_attn_prefill(void* out, const void* q,
                                    const int* seq_lens, const int* qlocs,
                                    const int* tile_seq, const int* tile_q0,
                                    int n_tiles, int bt_stride, int Hq, int Hk,
-                                   int head_dim, float scale, hipStream_t s) {
+                                   int head_dim, int kv_fp8, float scale,
+                                   hipStream_t s) {
   if (n_tiles == 0) return;
   const int G = Hq / Hk;
+#define LAUNCH_PF(DV, KV)                                                     \
+  do {                                                                        \
+    const size_t smem = sizeof(PrefillSmem<DV>) +                             \
+                        (size_t)G * PWAVE_ELEMS * sizeof(u16) +               \
+                        (size_t)G * 2 * QTILE * sizeof(float);                \
+    hipLaunchKernelGGL((attn_prefill_kernel<DV, KV>), dim3(n_tiles, Hk),      \
+                       dim3(G * WAVE), smem, s, (u16*)out, (const u16*)q,     \
+                       (const u8*)k_cache, (const u8*)v_cache, block_tables,  \
+                       seq_lens, qlocs, tile_seq, tile_q0, bt_stride, Hq, Hk, \
+                       scale);                                                \
+  } while (0)
   if (head_dim == 128) {
-    const size_t smem = sizeof(PrefillSmem<128>) +
-                        (size_t)G * PWAVE_ELEMS * sizeof(u16) +
-                        (size_t)G * 2 * QTILE * sizeof(float);
-    hipLaunchKernelGGL(attn_prefill_kernel<128>, dim3(n_tiles, Hk),
-                       dim3(G * WAVE), smem, s, (u16*)out, (const u16*)q,
-                       (const u16*)k_cache, (const u16*)v_cache, block_tables,
-                       seq_lens, qlocs, tile_seq, tile_q0, bt_stride, Hq, Hk,
-                       scale);
+    if (kv_fp8) LAUNCH_PF(128, true); else LAUNCH_PF(128, false);
   } else {
-    const size_t smem = sizeof(PrefillSmem<64>) +
-                        (size_t)G * PWAVE_ELEMS * sizeof(u16) +
-                        (size_t)G * 2 * QTILE * sizeof(float);
-    hipLaunchKernelGGL(attn_prefill_kernel<64>, dim3(n_tiles, Hk),
-                       dim3(G * WAVE), smem, s, (u16*)out, (const u16*)q,
-                       (const u16*)k_cache, (const u16*)v_cache, block_tables,
-                       seq_lens, qlocs, tile_seq, tile_q0, bt_stride, Hq, Hk,
-                       scale);
+    if (kv_fp8) LAUNCH_PF(64, true); else LAUNCH_PF(64, false);
   }
+#undef LAUNCH_PF
 }
 
 // ---------------------------------------------------------------------------

@@ -24,19 +24,22 @@ void sutro_silu_mul(void*, const void*, long, int, hipStream_t);
 void sutro_rope_and_cache(void*, void*, const void*, const long*, const long*,
                           void*, void*, const float*, int, int, int, int, int,
                           hipStream_t);
+static bool is_fp8(const torch::Tensor& t) {
+  return t.scalar_type() == at::kFloat8_e4m3fn;
+}
 void sutro_mean_pool_normalize(float*, const void*, const int*, int, int,
                                hipStream_t);
 void sutro_attn_decode(void*, const void*, const void*, const void*,
                        const int*, const int*, int, int, int, int, int, int,
-                       float, hipStream_t);
+                       int, float, hipStream_t);
 void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         const int*, const int*, const int*, const int*,
-                        const int*, int, int, int, int, int, float,
+                        const int*, int, int, int, int, int, int, float,
                         hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
-                    int, int, int, int, hipStream_t);
+                    int, int, int, int, int, hipStream_t);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -115,8 +118,8 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
                        v_cache.data_ptr(), block_tables.data_ptr<int>(),
                        seq_lens.data_ptr<int>(), qlocs.data_ptr<int>(),
                        tile_seq.data_ptr<int>(), tile_q0.data_ptr<int>(),
-                       n_tiles, bt_stride, Hq, Hk, D, (float)scale,
-                       cur_stream());
+                       n_tiles, bt_stride, Hq, Hk, D,
+                       is_fp8(k_cache) ? 1 : 0, (float)scale, cur_stream());
   }
   if (num_decodes > 0) {
     const long dec_off = prefill_token_count;  // decode rows are the tail
@@ -126,8 +129,9 @@ void paged_attention(torch::Tensor out, torch::Tensor q, torch::Tensor k_cache,
     sutro_attn_decode(op + dec_off * Hq * D, qp + dec_off * Hq * D,
                       k_cache.data_ptr(), v_cache.data_ptr(),
                       block_tables.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                      bt_stride, (int)num_decodes, Hq, Hk, D, seq_offset,
-                      (float)scale, cur_stream());
+                      bt_stride, (int)num_decodes, Hq, Hk, D,
+                      is_fp8(k_cache) ? 1 : 0, seq_offset, (float)scale,
+                      cur_stream());
   }
 }
 
@@ -149,7 +153,8 @@ void qkv_prep(torch::Tensor qkv, torch::Tensor q_out, torch::Tensor k_cache,
   sutro_qkv_prep(qkv.data_ptr(), q_out.data_ptr(), k_cache.data_ptr(),
                  v_cache.data_ptr(), positions.data_ptr<long>(),
                  slot_mapping.data_ptr<long>(), cos_sin.data_ptr<float>(), qw,
-                 kw, (float)eps, T, Hq, Hk, D, bs, row_stride, cur_stream());
+                 kw, (float)eps, T, Hq, Hk, D, bs, row_stride,
+                 is_fp8(k_cache) ? 1 : 0, cur_stream());
 }
 
 torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {

@@ -6,12 +6,16 @@
 
 #define WAVE 64u
 
+using u8 = unsigned char;
 using u16 = unsigned short;
 using u32 = unsigned int;
+using u64 = unsigned long long;
 
 typedef u16 u16x2 __attribute__((ext_vector_type(2)));
 typedef u16 u16x4 __attribute__((ext_vector_type(4)));
 typedef u16 u16x8 __attribute__((ext_vector_type(8)));
+typedef u32 u32x2 __attribute__((ext_vector_type(2)));
+typedef u32 u32x4 __attribute__((ext_vector_type(4)));
 typedef float f32x2 __attribute__((ext_vector_type(2)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
@@ -33,6 +37,32 @@ __device__ __forceinline__ u16 f2bf(float f) {
 __device__ __forceinline__ void bf8_to_f32(const u16x8 v, float* out) {
 #pragma unroll
   for (int j = 0; j < 8; ++j) out[j] = bf2f(v[j]);
+}
+
+// ---- OCP fp8 e4m3 (gfx950 native converts; matches torch.float8_e4m3fn) ----
+
+__device__ __forceinline__ u8 f2fp8(float a) {
+  return (u8)(__builtin_amdgcn_cvt_pk_fp8_f32(a, a, 0, false) & 0xFF);
+}
+
+__device__ __forceinline__ u16 f2fp8x2(float a, float b) {
+  return (u16)(__builtin_amdgcn_cvt_pk_fp8_f32(a, b, 0, false) & 0xFFFF);
+}
+
+__device__ __forceinline__ float fp8_to_f32(u8 v) {
+  return __builtin_amdgcn_cvt_f32_fp8((int)v, 0);
+}
+
+// 2 fp8 codes (low 2 bytes of the int) -> 2 f32
+__device__ __forceinline__ f32x2 fp8x2_to_f32(u32 v) {
+  return __builtin_amdgcn_cvt_pk_f32_fp8((int)v, false);
+}
+
+// 4 fp8 codes in a u32 -> out[0..3]
+__device__ __forceinline__ void fp8x4_to_f32(u32 v, float* out) {
+  f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8((int)v, false);
+  f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8((int)v, true);
+  out[0] = lo[0]; out[1] = lo[1]; out[2] = hi[0]; out[3] = hi[1];
 }
 
 // full-wave f32 reductions (64 lanes)

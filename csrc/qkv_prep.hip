@@ -19,7 +19,8 @@ __global__ void qkv_prep_kernel(
     const float* __restrict__ cos_sin,  // [max_pos, D]
     const u16* __restrict__ qw,      // [D] or null
     const u16* __restrict__ kw,      // [D] or null
-    float eps, int T, int Hq, int Hk, int D, int bs, int row_stride) {
+    float eps, int T, int Hq, int Hk, int D, int bs, int row_stride,
+    int kv_fp8) {                    // caches store OCP e4m3 bytes
   const int lane = threadIdx.x & (WAVE - 1);
   const int H = Hq + 2 * Hk;
   const long item = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -33,20 +34,34 @@ __global__ void qkv_prep_kernel(
   const long blk = slot / bs, off = slot - blk * bs;
 
   if (h >= Hq + Hk) {
-    // V head: plain copy into the paged cache
+    // V head: copy into the paged cache (optionally quantizing to e4m3)
     const int vh = h - Hq - Hk;
     const u16* src = qkv + (long)t * row_stride + Hq * D + Hk * D + vh * D;
-    u16* dst = v_cache + (((long)blk * Hk + vh) * bs + off) * D;
-    for (int j = lane; j < D / 8; j += (int)WAVE)
-      *(u16x8*)(dst + j * 8) = *(const u16x8*)(src + j * 8);
+    if (kv_fp8) {
+      u8* dst8 = (u8*)v_cache + (((long)blk * Hk + vh) * bs + off) * D;
+      for (int j = lane; j < D / 8; j += (int)WAVE) {
+        u16x8 v = *(const u16x8*)(src + j * 8);
+        u16x4 packed;
+#pragma unroll
+        for (int t2 = 0; t2 < 4; ++t2)
+          packed[t2] = f2fp8x2(bf2f(v[2 * t2]), bf2f(v[2 * t2 + 1]));
+        *(u16x4*)(dst8 + j * 8) = packed;
+      }
+    } else {
+      u16* dst = v_cache + (((long)blk * Hk + vh) * bs + off) * D;
+      for (int j = lane; j < D / 8; j += (int)WAVE)
+        *(u16x8*)(dst + j * 8) = *(const u16x8*)(src + j * 8);
+    }
     return;
   }
 
   const bool is_q = h < Hq;
   const u16* src = is_q ? qkv + (long)t * row_stride + h * D
                         : qkv + (long)t * row_stride + Hq * D + (h - Hq) * D;
-  u16* dst = is_q ? q_out + ((long)t * Hq + h) * D
-                  : k_cache + (((long)blk * Hk + (h - Hq)) * bs + off) * D;
+  const long k_slot = (((long)blk * Hk + (h - Hq)) * bs + off) * D;
+  u16* dst = is_q ? q_out + ((long)t * Hq + h) * D : k_cache + k_slot;
+  const bool fp8_out = !is_q && kv_fp8;
+  u8* dst8 = (u8*)k_cache + k_slot;
   const u16* w = is_q ? qw : kw;
 
   // load the row: lane holds pairs (d, d+half) for d = lane + j*WAVE
@@ -80,8 +95,15 @@ __global__ void qkv_prep_kernel(
     const int d = lane + j * (int)WAVE;
     if (d >= half) continue;
     const float c = cs[d], sn = cs[d + half];
-    dst[d] = f2bf(x1[j] * c - x2[j] * sn);
-    dst[d + half] = f2bf(x2[j] * c + x1[j] * sn);
+    const float r1 = x1[j] * c - x2[j] * sn;
+    const float r2 = x2[j] * c + x1[j] * sn;
+    if (fp8_out) {
+      dst8[d] = f2fp8(r1);
+      dst8[d + half] = f2fp8(r2);
+    } else {
+      dst[d] = f2bf(r1);
+      dst[d + half] = f2bf(r2);
+    }
   }
 }
 
@@ -90,7 +112,7 @@ extern "C" void sutro_qkv_prep(const void* qkv, void* q_out, void* k_cache,
                                const long* slots, const float* cos_sin,
                                const void* qw, const void* kw, float eps, int T,
                                int Hq, int Hk, int D, int bs, int row_stride,
-                               hipStream_t s) {
+                               int kv_fp8, hipStream_t s) {
   const long items = (long)T * (Hq + 2 * Hk);
   if (items == 0) return;
   const int wpb = 4;
@@ -98,5 +120,5 @@ extern "C" void sutro_qkv_prep(const void* qkv, void* q_out, void* k_cache,
   hipLaunchKernelGGL(qkv_prep_kernel, dim3((unsigned)blocks), dim3(wpb * WAVE),
                      0, s, (const u16*)qkv, (u16*)q_out, (u16*)k_cache,
                      (u16*)v_cache, pos, slots, cos_sin, (const u16*)qw,
-                     (const u16*)kw, eps, T, Hq, Hk, D, bs, row_stride);
+                     (const u16*)kw, eps, T, Hq, Hk, D, bs, row_stride, kv_fp8);
 }

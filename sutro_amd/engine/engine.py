@@ -104,7 +104,7 @@ class LLMEngine:
             num_kv_heads=kvh,
             block_size=cfg.kv_block_size,
             head_dim=self.spec.head_dim,
-            dtype=cfg.dtype,
+            dtype=cfg.kv_torch_dtype(),
             device=cfg.device,
         )
         # block 0 is reserved scratch: hipGraph padding rows read/write it

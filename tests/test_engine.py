@@ -220,3 +220,30 @@ def test_stop_strings():
     out = eng2.tokenizer.decode(req.output_token_ids)
     assert stop not in out
     assert len(req.output_token_ids) < 12
+
+
+def test_fp8_kv_cache_cpu():
+    """fp8 KV cache (torch.float8_e4m3fn storage) vs bf16: close outputs."""
+    import torch
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import tiny_spec_for_tests
+
+    outs = {}
+    for kvd in ("bf16", "fp8_e4m3"):
+        cfg = EngineConfig(spec=tiny_spec_for_tests(), device="cpu",
+                           max_model_len=256, num_kv_blocks=64,
+                           max_tokens_per_step=128, seed=0, kv_dtype=kvd)
+        eng = LLMEngine(cfg)
+        assert eng.kv.k_cache[0].dtype == (
+            torch.float8_e4m3fn if kvd == "fp8_e4m3" else torch.float32)
+        r = eng.add_request(eng.tokenizer.encode("fp8 kv row"),
+                            SamplingParams(max_tokens=8, temperature=0))
+        while eng.has_work():
+            eng.step()
+        outs[kvd] = list(r.output_token_ids)
+    # greedy argmax is robust to small KV quantization error on 8 tokens
+    same = sum(a == b for a, b in zip(outs["bf16"], outs["fp8_e4m3"]))
+    assert same >= len(outs["bf16"]) - 2

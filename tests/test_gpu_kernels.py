@@ -321,3 +321,98 @@ def test_attn_head_dim_64():
     ref = R.paged_attention(q.float().cpu(), kc.float().cpu(),
                             vc.float().cpu(), bt_c, sl_c, ql_c, scale)
     assert torch.allclose(got.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_fp8_kv_attention():
+    """e4m3 KV cache: kernels vs CPU reference using the SAME quantized cache
+    (both sides round-trip through torch.float8_e4m3fn, so tolerance stays
+    bf16-tight)."""
+    require_gpu()
+    torch.manual_seed(13)
+    bs, Hq, Hk, D = 32, 8, 2, 128
+    seq_lens = [70, 40, 33, 1]
+    new_counts = [70, 40, 1, 1]   # 2 prefills + 2 decodes
+    S = len(seq_lens)
+    nb = sum((L + bs - 1) // bs for L in seq_lens) + 2
+    kc = torch.zeros(nb, Hk, bs, D, dtype=torch.float8_e4m3fn, device=DEV)
+    vc = torch.zeros_like(kc)
+    perm = torch.randperm(nb).tolist()
+    tables, i = [], 0
+    for L in seq_lens:
+        n = (L + bs - 1) // bs
+        tables.append(perm[i:i + n]); i += n
+    max_b = max(len(t) for t in tables)
+    bt = torch.zeros(S, max_b, dtype=torch.int32)
+    for s, t in enumerate(tables):
+        bt[s, :len(t)] = torch.tensor(t, dtype=torch.int32)
+    qs = []
+    for s, L in enumerate(seq_lens):
+        K = bf(torch.randn(L, Hk, D))
+        V = bf(torch.randn(L, Hk, D))
+        slots = torch.tensor([tables[s][p // bs] * bs + p % bs
+                              for p in range(L)], dtype=torch.long)
+        R.write_kv_cache(K, V, kc, vc, slots.to(DEV))
+        qs.append(bf(torch.randn(new_counts[s], Hq, D)))
+    q = torch.cat(qs, 0)
+    qlocs = torch.tensor([0] + list(torch.tensor(new_counts).cumsum(0)),
+                         dtype=torch.int32)
+    sl = torch.tensor(seq_lens, dtype=torch.int32)
+    tiles_s, tiles_q0 = [], []
+    for s in range(2):
+        for j in range(0, new_counts[s], 32):
+            tiles_s.append(s); tiles_q0.append(j)
+    scale = 1.0 / math.sqrt(D)
+    got = ops.paged_attention(
+        q, kc, vc, bt.to(DEV), sl.to(DEV), qlocs.to(DEV), scale,
+        num_decodes_tail=2,
+        tile_seq=torch.tensor(tiles_s, dtype=torch.int32, device=DEV),
+        tile_q0=torch.tensor(tiles_q0, dtype=torch.int32, device=DEV),
+        prefill_token_count=110)
+    ref = R.paged_attention(q.float().cpu(), kc.cpu().float(),
+                            vc.cpu().float(), bt, sl, qlocs, scale)
+    assert torch.allclose(got.float().cpu(), ref, atol=5e-2, rtol=5e-2), (
+        (got.float().cpu() - ref).abs().max())
+
+
+def test_fp8_qkv_prep_write():
+    """qkv_prep writes e4m3 K/V codes identical to torch's RNE cast."""
+    require_gpu()
+    torch.manual_seed(17)
+    T, Hq, Hk, D, bs = 15, 4, 2, 128, 32
+    qkv = bf(torch.randn(T, (Hq + 2 * Hk) * D))
+    pos = torch.randint(0, 100, (T,), dtype=torch.long)
+    slots = torch.randperm(2 * bs)[:T].to(torch.long)
+    cs = R.rope_cos_sin(128, D, 1e6).to(DEV)
+    kc = torch.zeros(2, Hk, bs, D, dtype=torch.float8_e4m3fn, device=DEV)
+    vc = torch.zeros_like(kc)
+    q_out = ops.fused_qkv_prep(qkv, Hq, Hk, D, pos.to(DEV), slots.to(DEV),
+                               kc, vc, cs, None, None, 1e-6)
+    # CPU composition with the same fp8 cache dtype
+    kc_ref = torch.zeros(2, Hk, bs, D, dtype=torch.float8_e4m3fn)
+    vc_ref = torch.zeros_like(kc_ref)
+    q_ref = ops.fused_qkv_prep(qkv.cpu(), Hq, Hk, D, pos, slots, kc_ref,
+                               vc_ref, cs.cpu(), None, None, 1e-6)
+    assert torch.allclose(q_out.float().cpu(), q_ref.float(), atol=3e-2,
+                          rtol=3e-2)
+    # e4m3 codes should agree except for ties at RNE boundaries
+    mism = (kc.cpu().float() - kc_ref.float()).abs().max()
+    assert mism < 0.07, mism
+    mismv = (vc.cpu().float() - vc_ref.float()).abs().max()
+    assert mismv < 0.07, mismv
+
+
+def test_engine_e2e_fp8_gpu():
+    require_gpu()
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import get_model_spec
+
+    cfg = EngineConfig(spec=get_model_spec("qwen-3-0.6b"), device="cuda",
+                       max_model_len=512, num_kv_blocks=512,
+                       max_tokens_per_step=2048, max_num_seqs=64,
+                       kv_dtype="fp8_e4m3")
+    eng = LLMEngine(cfg)
+    outs = eng.generate([f"fp8 row {i}" for i in range(4)],
+                        sampling=SamplingParams(max_tokens=12, temperature=0.8))
+    assert len(outs) == 4 and eng.total_output_tokens >= 4
