@@ -247,3 +247,38 @@ def test_fp8_kv_cache_cpu():
     # greedy argmax is robust to small KV quantization error on 8 tokens
     same = sum(a == b for a, b in zip(outs["bf16"], outs["fp8_e4m3"]))
     assert same >= len(outs["bf16"]) - 2
+
+
+def test_cancel_mid_prefill():
+    eng = _engine(max_tokens_per_step=16)
+    req = eng.add_request(list(range(3, 3 + 60)), SamplingParams(max_tokens=4))
+    eng.step()  # partially prefilled
+    assert req.in_prefill and req in eng.scheduler.running
+    eng.abort_request(req)
+    assert req.finished and req not in eng.scheduler.running
+    # KV blocks released
+    assert req.req_id not in eng.kv.block_tables
+    eng.step()  # no crash with empty schedule
+
+
+def test_priority_preemption_prefers_low_priority_victims():
+    """Preemption evicts the most recent admission; p1 work queued behind p0
+    is readmitted later and still completes."""
+    eng = _engine(num_kv_blocks=10, max_tokens_per_step=128)
+    p0 = [eng.add_request(eng.tokenizer.encode("p0 " + "x" * 50),
+                          SamplingParams(max_tokens=20, temperature=0.5),
+                          priority=0) for _ in range(4)]
+    p1 = [eng.add_request(eng.tokenizer.encode("p1 " + "y" * 50),
+                          SamplingParams(max_tokens=20, temperature=0.5),
+                          priority=1) for _ in range(4)]
+    steps = 0
+    while eng.has_work():
+        eng.step()
+        steps += 1
+        assert steps < 3000
+    assert all(r.finished for r in p0 + p1)
+
+
+def test_engine_empty_step_is_noop(tiny_engine):
+    stats = tiny_engine.step()
+    assert stats.scheduled_tokens == 0 and not stats.finished

@@ -274,3 +274,33 @@ def test_random_seed_per_input_deterministic(local_client):
     # and the two identical rows get DIFFERENT samples within a job
     assert d1["inference_result"].tolist() == d2["inference_result"].tolist()
     assert d1["inference_result"][0] != d1["inference_result"][1]
+
+
+def test_attach_finished_job_short_circuits(local_client, capsys):
+    job_id = local_client.infer(["short"], model="qwen-3.5-2b",
+                                stay_attached=False, sampling_params=SP)
+    _await(local_client, job_id)
+    local_client.attach(job_id)
+    out = capsys.readouterr().out
+    assert "already" in out
+
+
+def test_await_timeout_raises(local_client):
+    job_id = local_client.infer([f"r{i}" for i in range(200)],
+                                model="qwen-3.5-2b", stay_attached=False,
+                                sampling_params={"max_tokens": 400})
+    with pytest.raises(TimeoutError):
+        local_client.await_job_completion(job_id, timeout=0.3)
+    local_client.cancel_job(job_id)
+
+
+def test_stay_attached_prints_progress(local_client, capsys):
+    local_client.infer(["attached row"], model="qwen-3.5-2b",
+                       stay_attached=True, sampling_params=SP)
+    out = capsys.readouterr().out
+    assert "submitted" in out and "completed" in out
+
+
+def test_unknown_model_fails_loudly(local_client):
+    with pytest.raises(RuntimeError):
+        local_client.infer(["x"], model="not-a-model", stay_attached=False)
