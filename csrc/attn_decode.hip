@@ -86,6 +86,7 @@ __device__ __forceinline__ void phase_a(DecodeSmem<D, KV8>* sm,
   // j-outer shared-decode order) pushes the kernel past 215 VGPR into scratch
   // spill (occupancy 4.5 waves/CU measured; §5.4 rule 20). Decoding K
   // redundantly per head sits in idle VALU headroom (VALUBusy 15% measured).
+  float sc[G];
 #pragma unroll
   for (int g = 0; g < G; ++g) {
     const float* qv = sm->qs + g * D + c.half * (D / 2);
@@ -117,18 +118,49 @@ __device__ __forceinline__ void phase_a(DecodeSmem<D, KV8>* sm,
         }
       }
     }
-    s += __shfl_xor(s, 1, 64);  // combine the two half-dim lanes
-    if (c.p_pos >= c.valid) s = -1e30f;
-    const float tile_max = wave_max_f32(s);
-    const float m_new = fmaxf(st.m[g], tile_max);
-    const float alpha = __expf(st.m[g] - m_new);
-    const float p_val = __expf(s - m_new);
-    const float tile_sum = wave_sum_f32(p_val) * 0.5f;  // each pos in 2 lanes
-    st.lsum[g] = st.lsum[g] * alpha + tile_sum;
-    st.acc0[g] *= alpha;
-    st.acc1[g] *= alpha;
+    sc[g] = s;
+  }
+  // Batched cross-lane softmax: offset-outer / head-inner keeps the G
+  // independent shuffle-reduce chains in flight together. The serial
+  // per-head form (12 dependent DS ops x G heads) was ~4 us/page/wave —
+  // the dominant cost of the whole kernel at short context.
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    sc[g] += __shfl_xor(sc[g], 1, 64);  // combine the two half-dim lanes
+    if (c.p_pos >= c.valid) sc[g] = -1e30f;
+  }
+  float mt[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) mt[g] = sc[g];
+  // lanes 2p/2p+1 now duplicate: strides {2..32} close each parity class
+#pragma unroll
+  for (int off = 2; off <= 32; off <<= 1) {
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+      mt[g] = fmaxf(mt[g], __shfl_xor(mt[g], off, 64));
+  }
+  float p_val[G], alpha[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    const float m_new = fmaxf(st.m[g], mt[g]);
+    alpha[g] = __expf(st.m[g] - m_new);
+    p_val[g] = __expf(sc[g] - m_new);
     st.m[g] = m_new;
-    if (c.half == 0) sm->p[g * BS + c.p_pos] = p_val;
+  }
+  float ps[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) ps[g] = p_val[g];
+#pragma unroll
+  for (int off = 2; off <= 32; off <<= 1) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) ps[g] += __shfl_xor(ps[g], off, 64);
+  }
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    st.lsum[g] = st.lsum[g] * alpha[g] + ps[g];
+    st.acc0[g] *= alpha[g];
+    st.acc1[g] *= alpha[g];
+    if (c.half == 0) sm->p[g * BS + c.p_pos] = p_val[g];
   }
 }
 
