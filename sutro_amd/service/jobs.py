@@ -132,7 +132,14 @@ class EngineWorker:
 
     def submit_job(self, job: JobRecord, service: "JobService") -> None:
         with self._lock:
-            self._inbox.append((job, service))
+            self._inbox.append((job, service, None))
+        self._wake.set()
+
+    def submit_partial(self, job: JobRecord, service: "JobService",
+                       rows_idx) -> None:
+        """Admit only the given row indices (engine-restart resume)."""
+        with self._lock:
+            self._inbox.append((job, service, list(rows_idx)))
         self._wake.set()
 
     def cancel_job(self, job_id: str) -> None:
@@ -147,9 +154,10 @@ class EngineWorker:
 
     # ---- worker loop ----
 
-    def _admit(self, job: JobRecord, service: "JobService") -> None:
+    def _admit(self, job: JobRecord, service: "JobService",
+               rows_idx=None) -> None:
         eng = self.engine
-        job.datetime_started = _now()
+        job.datetime_started = job.datetime_started or _now()
         job.status = JobStatus.RUNNING
         service.persist_job(job)
         fsm_id = None
@@ -163,7 +171,9 @@ class EngineWorker:
             fsm_id = eng.register_fsm(schema)
         default_max = 1024 if schema is not None else eng.cfg.default_max_new_tokens
         t_start = time.time()
-        for i, row in enumerate(job.inputs):
+        indices = rows_idx if rows_idx is not None else range(len(job.inputs))
+        for i in indices:
+            row = job.inputs[i]
             text = row if isinstance(row, str) else json.dumps(row)
             ids = self.tokenizer.render_prompt(text, job.system_prompt)
             sp = SamplingParams.from_dict(job.sampling_params, default_max)
@@ -219,9 +229,9 @@ class EngineWorker:
             with self._lock:
                 inbox, self._inbox = self._inbox, []
                 cancelled, self._cancelled_jobs = self._cancelled_jobs, set()
-            for job, service in inbox:
+            for job, service, rows_idx in inbox:
                 try:
-                    self._admit(job, service)
+                    self._admit(job, service, rows_idx)
                 except Exception as e:  # admission failure -> FAILED
                     job.status = JobStatus.FAILED
                     job.failure_reason = {"message": f"{type(e).__name__}: {e}"}
@@ -553,6 +563,9 @@ class JobService:
                     "id_column_values": job.id_column_values,
                     "inputs": [r if isinstance(r, str) else json.dumps(r)
                                for r in job.inputs],
+                    "json_schema": job.json_schema,
+                    "system_prompt": job.system_prompt,
+                    "sampling_params": job.sampling_params,
                 }, f)
 
     def _load_persisted_jobs(self) -> None:
@@ -565,10 +578,6 @@ class JobService:
                     pub = json.load(f)
                 job = JobRecord(job_id=pub["job_id"], model=pub.get("model", ""))
                 job.status = JobStatus(pub.get("status", "UNKNOWN"))
-                if not JobStatus.is_terminal(job.status):
-                    # a job that was live in a dead process cannot resume rows
-                    job.status = JobStatus.FAILED
-                    job.failure_reason = {"message": "service restarted mid-job"}
                 job.name = pub.get("name")
                 job.description = pub.get("description")
                 job.num_rows = pub.get("num_rows", 0)
@@ -591,10 +600,65 @@ class JobService:
                     job.id_column_name = r.get("id_column_name")
                     job.id_column_values = r.get("id_column_values")
                     job.inputs = r.get("inputs", [])
+                    job.json_schema = r.get("json_schema")
+                    job.system_prompt = r.get("system_prompt")
+                    job.sampling_params = r.get("sampling_params")
+                if not JobStatus.is_terminal(job.status):
+                    self._resume_or_fail(job)
                 self.jobs[job.job_id] = job
             except Exception:
                 continue
 
+    def _resume_or_fail(self, job: JobRecord) -> None:
+        """A job that was live when the service died: finish the MISSING rows
+        from the persisted shards (resume), or fail it if inputs are gone."""
+        done = [o is not None for o in job.outputs]
+        if not job.inputs or len(job.inputs) != job.num_rows:
+            job.status = JobStatus.FAILED
+            job.failure_reason = {
+                "message": "service restarted mid-job with no persisted shards"}
+            job.datetime_completed = _now()
+            self.persist_job(job)
+            return
+        job.completed_rows = sum(done)
+        if job.completed_rows == job.num_rows:
+            job.status = JobStatus.SUCCEEDED
+            job.datetime_completed = _now()
+            self.persist_job(job, with_results=True)
+            return
+        # re-run only the incomplete rows through a fresh worker
+        missing = [i for i, d in enumerate(done) if not d]
+        job.status = JobStatus.RUNNING
+        worker = self._get_worker(job.model)
+        self.jobs[job.job_id] = job
+        _submit_rows_into(worker, self, job, missing)
+
     def shutdown(self) -> None:
         for w in self.workers.values():
             w.shutdown()
+
+
+def _submit_rows_into(worker, service: "JobService", job: JobRecord,
+                      rows_idx) -> None:
+    """Feed specific row indices of an existing JobRecord into a worker."""
+    if hasattr(worker, "in_qs"):  # MultiProcEngineWorker
+        opts = {
+            "system_prompt": job.system_prompt,
+            "json_schema": job.json_schema,
+            "sampling_params": job.sampling_params,
+            "random_seed_per_input": job.random_seed_per_input,
+            "truncate_rows": job.truncate_rows,
+            "priority": job.priority,
+        }
+        rows = [(i, job.inputs[i] if isinstance(job.inputs[i], str)
+                 else json.dumps(job.inputs[i])) for i in rows_idx]
+        worker._jobs[job.job_id] = (job, service, time.time())
+        per = (len(rows) + worker.n_replicas - 1) // worker.n_replicas
+        for rep in range(worker.n_replicas):
+            shard = rows[rep * per:(rep + 1) * per]
+            if shard:
+                worker.in_qs[rep * worker.tp].put(
+                    ("run_rows", job.job_id, shard, opts))
+        return
+    # in-process EngineWorker: admit only the missing rows of the SAME record
+    worker.submit_partial(job, service, rows_idx)

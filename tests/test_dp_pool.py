@@ -134,3 +134,41 @@ def test_worker_death_fails_job(sutro_home, monkeypatch):
         assert "crash" in job.failure_reason["message"]
     finally:
         svc.shutdown()
+
+
+def test_engine_restart_resume(sutro_home, monkeypatch):
+    """A job mid-flight when the service dies resumes from persisted shards:
+    the restarted service re-runs only the missing rows."""
+    import json as _json
+    import os
+
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "1")
+    from sutro_amd.service.jobs import JobService
+
+    home = sutro_home + "-resume"
+    os.makedirs(os.path.join(home, "jobs"), exist_ok=True)
+    os.makedirs(os.path.join(home, "job-results"), exist_ok=True)
+    # craft a half-finished persisted job (as the periodic persist writes it)
+    jid = "job-resume-test"
+    with open(os.path.join(home, "jobs", f"{jid}.json"), "w") as f:
+        _json.dump({"job_id": jid, "model": "qwen-3.5-2b", "status": "RUNNING",
+                    "num_rows": 4, "input_tokens": 10, "output_tokens": 5,
+                    "datetime_created": "2026-01-01T00:00:00"}, f)
+    with open(os.path.join(home, "job-results", f"{jid}.json"), "w") as f:
+        _json.dump({"outputs": ["done-a", "done-b", None, None],
+                    "embeddings": [None] * 4,
+                    "cumulative_logprobs": [-1.0, -1.0, None, None],
+                    "confidence_score": [0.5, 0.5, None, None],
+                    "inputs": ["r0", "r1", "r2", "r3"],
+                    "sampling_params": {"max_tokens": 6},
+                    "json_schema": None, "system_prompt": None}, f)
+    svc = JobService(home=home, device="cpu",
+                     engine_kwargs={"num_kv_blocks": 128, "max_model_len": 1024})
+    try:
+        assert _await(svc, jid) == "SUCCEEDED"
+        job = svc.get_job(jid)
+        # previously completed rows untouched; missing rows filled
+        assert job.outputs[0] == "done-a" and job.outputs[1] == "done-b"
+        assert job.outputs[2] is not None and job.outputs[3] is not None
+    finally:
+        svc.shutdown()
