@@ -120,10 +120,12 @@ class Rank(BaseSutroClient):
             output_schema=json_schema, job_priority=job_priority,
             stay_attached=False,
         )
-        res = self.await_job_completion(job_id, output_column=ranking_column_name)
+        res = self.await_job_completion(job_id, output_column=ranking_column_name,
+                                        unpack_json=False)
         if res is None:
             return None
-        rankings = res[ranking_column_name].tolist()
+        rankings = [json.loads(v)[ranking_column_name]
+                    for v in res[ranking_column_name]]
         out = df.assign(**{ranking_column_name: rankings})
         if run_elo:
             elo_df = Rank.elo(out, column=ranking_column_name)

@@ -304,3 +304,30 @@ def test_stay_attached_prints_progress(local_client, capsys):
 def test_unknown_model_fails_loudly(local_client):
     with pytest.raises(RuntimeError):
         local_client.infer(["x"], model="not-a-model", stay_attached=False)
+
+
+def test_score_template(local_client):
+    df = local_client.score(["good thing", "bad thing"], criteria="quality",
+                            model="qwen-3.5-2b", range=(0, 5))
+    assert "score" in df.columns
+    assert all(0 <= int(v) <= 5 for v in df["score"])
+
+
+def test_rank_template_with_elo(local_client, capsys):
+    out = local_client.rank(
+        data=[["answer one", "answer uno"], ["answer two", "answer dos"]],
+        option_labels=["a", "b"], criteria="clarity", model="qwen-3.5-2b")
+    assert "ranking" in out.columns
+    for r in out["ranking"]:
+        assert sorted(r) == ["a", "b"]  # FSM enforces exactly the labels
+    printed = capsys.readouterr().out
+    assert "elo" in printed.lower() or "a" in printed
+
+
+def test_classify_template_end_to_end(local_client):
+    res = local_client.classify(["great stuff", "awful stuff"],
+                                classes={"Positive": "good sentiment",
+                                         "Negative": "bad sentiment"},
+                                model="qwen-3.5-2b")
+    assert set(res["classification"]) <= {"Positive", "Negative"}
+    assert "scratchpad" not in res.columns
