@@ -292,7 +292,26 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
     if t == "null":
         return lit("null")
     if t == "array":
-        item = schema_to_regex(schema.get("items", {}), defs, depth + 1)
+        items_schema = schema.get("items", {})
+        # uniqueItems over a small enum: enumerate permutations (exact
+        # distinctness is beyond a DFA product otherwise) — rank() uses this
+        if (schema.get("uniqueItems") and "enum" in items_schema
+                and len(items_schema["enum"]) <= 6
+                and schema.get("minItems") == schema.get("maxItems")
+                == len(items_schema["enum"])):
+            from itertools import permutations
+
+            opts = []
+            for perm in permutations(items_schema["enum"]):
+                parts = [lit("[")]
+                for i, v in enumerate(perm):
+                    if i:
+                        parts.append(lit(","))
+                    parts.append(schema_to_regex({"enum": [v]}, defs, depth + 1))
+                parts.append(lit("]"))
+                opts.append(cat(*parts))
+            return alt(*opts)
+        item = schema_to_regex(items_schema, defs, depth + 1)
         lo = int(schema.get("minItems", 0))
         hi = min(int(schema.get("maxItems", _MAX_ARR)), _MAX_ARR)
         if hi == 0:

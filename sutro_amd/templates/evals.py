@@ -100,6 +100,7 @@ class Rank(BaseSutroClient):
                     "items": {"enum": list(option_labels)},
                     "minItems": len(option_labels),
                     "maxItems": len(option_labels),
+                    "uniqueItems": True,  # a true permutation of the labels
                 },
             },
             "required": [ranking_column_name],

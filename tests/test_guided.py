@@ -133,3 +133,13 @@ def test_classify_like_schema_has_enum_tail():
     d = dfa_for(schema)
     assert d.matches(b'{"scratchpad":"ok","classification":"Pos"}')
     assert not d.matches(b'{"scratchpad":"ok","classification":"Other"}')
+
+
+def test_unique_items_permutation():
+    schema = {"type": "array", "items": {"enum": ["x", "y", "z"]},
+              "minItems": 3, "maxItems": 3, "uniqueItems": True}
+    d = dfa_for(schema)
+    assert d.matches(b'["x","y","z"]')
+    assert d.matches(b'["z","x","y"]')
+    assert not d.matches(b'["x","x","y"]')
+    assert not d.matches(b'["x","y"]')
