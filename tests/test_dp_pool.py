@@ -172,3 +172,14 @@ def test_engine_restart_resume(sutro_home, monkeypatch):
         assert job.outputs[2] is not None and job.outputs[3] is not None
     finally:
         svc.shutdown()
+
+
+def test_dp_embedding_job(dp_service):
+    out = dp_service.submit_job({
+        "model": "qwen-3-embedding-0.6b",
+        "inputs": ["vector me", "and me", "me too"]})
+    job_id = out["results"]
+    assert _await(dp_service, job_id) == "SUCCEEDED"
+    results = dp_service.job_results(job_id)["results"]
+    assert len(results["outputs"]) == 3
+    assert all(isinstance(v, list) and len(v) > 0 for v in results["outputs"])
