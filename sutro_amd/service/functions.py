@@ -59,7 +59,13 @@ class FunctionStore:
         ids = tok.render_prompt(text, fn.get("system_prompt"))
         fsm_id = None
         if fn.get("output_schema"):
-            fsm_id = eng.register_fsm(fn["output_schema"])
+            key = json.dumps(fn["output_schema"], sort_keys=True)
+            cache = getattr(worker, "_fn_fsm_cache", None)
+            if cache is None:
+                cache = worker._fn_fsm_cache = {}
+            if key not in cache:
+                cache[key] = eng.register_fsm(fn["output_schema"])
+            fsm_id = cache[key]
         # add_request only touches python queues (GIL-atomic appends); the
         # worker thread owns all KV/tensor state and picks the request up.
         req = eng.add_request(ids, SamplingParams(max_tokens=512, temperature=0.7),
