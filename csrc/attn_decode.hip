@@ -333,9 +333,9 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
   const int G = Hq / Hk;
   // minimal-register serial variant is the measured default (8451 vs 7665
   // tok/s on Qwen3-32B batch-512 decode); SUTRO_DECODE_PF=1 re-enables the
-  // software-pipelined variant for A/B runs.
-  static const bool pf = getenv("SUTRO_DECODE_PF") != nullptr;
-  static const bool w3 = getenv("SUTRO_DECODE_W3") != nullptr;
+  // software-pipelined variant, SUTRO_DECODE_W3 the forced-3-wave one.
+  const bool pf = getenv("SUTRO_DECODE_PF") != nullptr;
+  const bool w3 = getenv("SUTRO_DECODE_W3") != nullptr;
 #define LAUNCH_K(KERNEL)                                                      \
   hipLaunchKernelGGL((KERNEL), dim3((unsigned)blocks), dim3(wpb * WAVE),      \
                      smem, s, (u16*)out, (const u16*)q, (const u8*)k_cache,   \
@@ -361,12 +361,23 @@ static void launch_decode(long blocks, int wpb, void* out, const void* q,
 #undef LAUNCH_K
 }
 
+extern "C" void sutro_attn_decode_mfma(void*, const void*, const void*,
+                                       const void*, const int*, const int*,
+                                       int, int, int, int, int, float,
+                                       hipStream_t);
+
 extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   const void* v_cache, const int* block_tables,
                                   const int* seq_lens, int bt_stride, int n_dec,
                                   int Hq, int Hk, int head_dim, int kv_fp8,
                                   int seq_offset, float scale, hipStream_t s) {
   if (n_dec == 0) return;
+  if (getenv("SUTRO_DECODE_MFMA") != nullptr && head_dim == 128 && !kv_fp8 &&
+      Hq / Hk <= 8) {
+    sutro_attn_decode_mfma(out, q, k_cache, v_cache, block_tables, seq_lens,
+                           bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
+    return;
+  }
   const int wpb = 4;
   const long items = (long)n_dec * Hk;
   const long blocks = (items + wpb - 1) / wpb;

@@ -132,9 +132,12 @@ __global__ void attn_prefill_kernel(
         }
         *(u16x8*)((char*)sm->ktile + k_swz<D>(row, slot * 16)) = kx;
       }
-      // V^T: read V[kv][d0..d0+8), write 8 u16 at vt[d][kv]
+      // V^T: read V[kv][d0..d0+8), write 8 u16 at vt[d][kv].
+      // kv-major across consecutive threads: the 2B scatter writes of a
+      // 16-lane group then span 16 banks instead of hitting one (d-stride
+      // 8*PV pad rows is 0 mod 32 banks).
       for (int it = tid; it < BS * SLOTS; it += nthr) {
-        const int kv = it / SLOTS, d0 = (it % SLOTS) * 8;
+        const int kv = it % BS, d0 = (it / BS) * 8;
         const u8* src = v_cache + ((long)kv_base + kv * D + d0) * ES;
         u16x8 vx;
         if constexpr (KV8) {

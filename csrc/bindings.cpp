@@ -37,6 +37,7 @@ void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         const int*, int, int, int, int, int, int, float,
                         hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
+void sutro_mfma16_probe(float*, const void*, const void*, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
                     int, int, int, int, int, hipStream_t);
@@ -165,6 +166,14 @@ torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
+  CHECK_CUDA(a); CHECK_BF16(a);
+  auto c = torch::zeros({16, 16}, a.options().dtype(at::kFloat));
+  sutro_mfma16_probe(c.data_ptr<float>(), a.data_ptr(), b.data_ptr(),
+                     cur_stream());
+  return c;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual+=x; x=rmsnorm");
@@ -174,4 +183,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mean_pool_normalize", &mean_pool_normalize, "varlen mean pool + L2");
   m.def("paged_attention", &paged_attention, "paged prefill+decode attention");
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
+  m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
 }

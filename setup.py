@@ -18,6 +18,7 @@ SOURCES = [
     "csrc/qkv_prep.hip",
     "csrc/pool.hip",
     "csrc/attn_decode.hip",
+    "csrc/attn_decode_mfma.hip",
     "csrc/attn_prefill.hip",
 ]
 

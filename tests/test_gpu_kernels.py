@@ -418,3 +418,32 @@ def test_engine_e2e_fp8_gpu():
     outs = eng.generate([f"fp8 row {i}" for i in range(4)],
                         sampling=SamplingParams(max_tokens=12, temperature=0.8))
     assert len(outs) == 4 and eng.total_output_tokens >= 4
+
+
+def test_mfma16_layout_probe():
+    require_gpu()
+    torch.manual_seed(21)
+    a = torch.randn(16, 32)
+    b = torch.randn(32, 16) * torch.arange(1, 17).float() / 4.0
+    c = _C.mfma16_probe(bf(a), bf(b))
+    ref = (bf(a).float() @ bf(b).float())
+    assert torch.allclose(c.cpu(), ref.cpu(), atol=2e-2, rtol=2e-2), (
+        (c.cpu() - ref.cpu()).abs().max())
+
+
+def test_attn_decode_mfma_matches_ref(monkeypatch):
+    require_gpu()
+    monkeypatch.setenv("SUTRO_DECODE_MFMA", "1")
+    for hq, hk in ((8, 1), (8, 2), (16, 2)):
+        seq_lens = [1, 31, 32, 33, 100, 257]
+        q, kc, vc, bt, sl, ql, sl_c, ql_c, bt_c = _attn_setup(
+            seq_lens, [1] * len(seq_lens), Hq=hq, Hk=hk, seed=23)
+        scale = 1.0 / math.sqrt(128)
+        got = ops.paged_attention(q, kc, vc, bt, sl, ql, scale,
+                                  num_decodes_tail=len(seq_lens),
+                                  prefill_token_count=0)
+        ref = R.paged_attention(q.float().cpu(), kc.float().cpu(),
+                                vc.float().cpu(), bt_c, sl_c, ql_c, scale)
+        err = (got.float().cpu() - ref).abs().max()
+        assert torch.allclose(got.float().cpu(), ref, atol=3e-2,
+                              rtol=3e-2), (hq, hk, err)
