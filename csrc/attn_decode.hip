@@ -372,8 +372,11 @@ extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
                                   int Hq, int Hk, int head_dim, int kv_fp8,
                                   int seq_offset, float scale, hipStream_t s) {
   if (n_dec == 0) return;
-  if (getenv("SUTRO_DECODE_MFMA") != nullptr && head_dim == 128 && !kv_fp8 &&
-      Hq / Hk <= 8) {
+  // MFMA variant is the measured default for bf16 KV at head_dim 128
+  // (10654 vs 10134 tok/s at batch 1024 / ctx 128; attention slope halves at
+  // ctx 512). SUTRO_DECODE_VALU=1 falls back to the VALU kernel for A/B.
+  if (head_dim == 128 && !kv_fp8 && Hq / Hk <= 8 &&
+      getenv("SUTRO_DECODE_VALU") == nullptr) {
     sutro_attn_decode_mfma(out, q, k_cache, v_cache, block_tables, seq_lens,
                            bt_stride, n_dec, Hq, Hk, seq_offset, scale, s);
     return;
