@@ -19,6 +19,7 @@
 //   B[32,16]: lane l holds B[(l/16)*8 + j][l%16]
 //   C/D     : lane l holds D[(l/16)*4 + r][l%16], r=0..3 (guide §3)
 #include "common.h"
+#include <cstdlib>
 
 #define BS 32
 #define DM 128       // head_dim
@@ -212,7 +213,9 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
                                        int n_dec, int Hq, int Hk,
                                        int seq_offset, float scale,
                                        hipStream_t s) {
-  const int wpb = 4;
+  // 2 waves/block: 31.4 KB LDS -> 5 blocks/CU (10 waves) vs 2 at wpb=4
+  static const char* wpb_env = getenv("SUTRO_DECODE_WPB");
+  const int wpb = wpb_env ? atoi(wpb_env) : 2;
   const long items = (long)n_dec * Hk;
   const long blocks = (items + wpb - 1) / wpb;
   const size_t smem = sizeof(MfmaSmem) * wpb;
