@@ -325,8 +325,7 @@ class LLMEngine:
 
     def _apply_sampled(self, req: Request, tok: int, lp: float, stats: StepStats) -> None:
         sp = req.sampling
-        stop_ids = set(sp.stop_token_ids or ()) | {EOS_ID}
-        if tok in stop_ids:
+        if tok in req.stop_ids(EOS_ID):
             self.scheduler.finish(req, FinishReason.STOP)
             stats.finished.append(req)
             return

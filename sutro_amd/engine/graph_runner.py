@@ -14,6 +14,7 @@ from __future__ import annotations
 
 from typing import Dict, List, Optional
 
+import numpy as np
 import torch
 
 from .batch import ForwardBatch, ScheduledBatch
@@ -49,6 +50,14 @@ class DecodeGraphRunner:
         self.h_slots = torch.zeros(B, dtype=torch.long, pin_memory=True)
         self.h_bt = torch.zeros(B, W, dtype=torch.int32, pin_memory=True)
         self.h_sl = torch.ones(B, dtype=torch.int32, pin_memory=True)
+        # incremental block-table state: which request's table row i holds, how
+        # many blocks of it are written, and its alloc_gen (preemption counter)
+        # — at steady decode a row's table only gains one block every
+        # kv_block_size steps, so per-step work is a compare, not a copy
+        self._row_req = np.full(B, -1, dtype=np.int64)
+        self._row_nb = np.zeros(B, dtype=np.int32)
+        self._row_gen = np.zeros(B, dtype=np.int64)
+        self._arange = np.arange(B, dtype=np.int64)
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self.logits_out: Dict[int, torch.Tensor] = {}
         self._pool = None
@@ -100,9 +109,21 @@ class DecodeGraphRunner:
     @torch.no_grad()
     def run(self, sb: ScheduledBatch) -> torch.Tensor:
         """Fill static buffers, replay, return logits[:n]."""
-        eng = self.engine
         n = len(sb.reqs)
         bucket = next(b for b in self.buckets if b >= n)
+        self._fill_host(sb, bucket)
+        self.d_ids[:bucket].copy_(self.h_ids[:bucket], non_blocking=True)
+        self.d_pos[:bucket].copy_(self.h_pos[:bucket], non_blocking=True)
+        self.d_slots[:bucket].copy_(self.h_slots[:bucket], non_blocking=True)
+        self.d_bt[:bucket].copy_(self.h_bt[:bucket], non_blocking=True)
+        self.d_sl[:bucket].copy_(self.h_sl[:bucket], non_blocking=True)
+        self.graphs[bucket].replay()
+        return self.logits_out[bucket][:n]
+
+    def _fill_host(self, sb: ScheduledBatch, bucket: int) -> None:
+        """Fill the pinned host staging buffers for a decode batch."""
+        eng = self.engine
+        n = len(sb.reqs)
         kv = eng.kv
         bs = kv.block_size
         scratch = eng.scratch_block
@@ -112,24 +133,42 @@ class DecodeGraphRunner:
         slots = self.h_slots.numpy()
         bt = self.h_bt.numpy()
         sl = self.h_sl.numpy()
-        ids[:bucket] = 0
-        pos[:bucket] = 0
-        sl[:bucket] = 1
-        bt[:bucket, 0] = scratch
-        slots[:bucket] = scratch * bs
-        for i, req in enumerate(sb.reqs):
-            p = req.num_computed_tokens
-            table = kv.block_tables[req.req_id]
-            ids[i] = req.token_at(p)
-            pos[i] = p
-            slots[i] = table[p // bs] * bs + p % bs
-            sl[i] = p + 1
-            bt[i, :len(table)] = table
 
-        self.d_ids[:bucket].copy_(self.h_ids[:bucket], non_blocking=True)
-        self.d_pos[:bucket].copy_(self.h_pos[:bucket], non_blocking=True)
-        self.d_slots[:bucket].copy_(self.h_slots[:bucket], non_blocking=True)
-        self.d_bt[:bucket].copy_(self.h_bt[:bucket], non_blocking=True)
-        self.d_sl[:bucket].copy_(self.h_sl[:bucket], non_blocking=True)
-        self.graphs[bucket].replay()
-        return self.logits_out[bucket][:n]
+        p = np.fromiter((r.num_computed_tokens for r in sb.reqs), np.int64, n)
+        pos[:n] = p
+        sl[:n] = p + 1
+        # decode input = last sampled token (or prompt tail right after prefill)
+        ids[:n] = np.fromiter(
+            ((r.output_token_ids[-1] if r.output_token_ids
+              else r.prompt_token_ids[-1]) for r in sb.reqs), np.int64, n)
+
+        row_req, row_nb, row_gen = self._row_req, self._row_nb, self._row_gen
+        tables = kv.block_tables
+        rid = np.fromiter((r.req_id for r in sb.reqs), np.int64, n)
+        gen = np.fromiter((r.alloc_gen for r in sb.reqs), np.int64, n)
+        # a row needs no table work when the same (req, alloc_gen) occupies it
+        # and the already-written blocks cover the token being written
+        stale = ((row_req[:n] != rid) | (row_gen[:n] != gen)
+                 | (row_nb[:n] < p // bs + 1))
+        for i in np.nonzero(stale)[0]:
+            req = sb.reqs[i]
+            table = tables[req.req_id]
+            tl = len(table)
+            if row_req[i] != req.req_id or row_gen[i] != req.alloc_gen:
+                bt[i, :tl] = table
+                row_req[i] = req.req_id
+                row_nb[i] = tl
+                row_gen[i] = req.alloc_gen
+            else:
+                bt[i, row_nb[i]:tl] = table[row_nb[i]:tl]
+                row_nb[i] = tl
+        rows = self._arange[:n]
+        slots[:n] = bt[rows, p // bs].astype(np.int64) * bs + p % bs
+
+        if n < bucket:  # padding rows target the reserved scratch block
+            ids[n:bucket] = 0
+            pos[n:bucket] = 0
+            sl[n:bucket] = 1
+            bt[n:bucket, 0] = scratch
+            slots[n:bucket] = scratch * bs
+            row_req[n:bucket] = -1

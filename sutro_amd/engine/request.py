@@ -61,6 +61,17 @@ class Request:
     num_computed_tokens: int = 0   # prompt tokens whose KV is already cached
     fsm_state: int = 0
     finish_reason: Optional[FinishReason] = None
+    # bumped when the request's KV pages are released while it stays schedulable
+    # (preemption) so cached per-row block-table state can be invalidated
+    alloc_gen: int = 0
+    _stop_ids: Optional[frozenset] = field(default=None, repr=False)
+
+    def stop_ids(self, eos_id: int) -> frozenset:
+        """Cached set of stop token ids (incl. EOS) — built once, checked every
+        decode step."""
+        if self._stop_ids is None:
+            self._stop_ids = frozenset(self.sampling.stop_token_ids or ()) | {eos_id}
+        return self._stop_ids
 
     @property
     def num_prompt_tokens(self) -> int:

@@ -71,6 +71,7 @@ class Scheduler:
             self.running.remove(victim)
             self.kv.release(victim.req_id)
             victim.num_computed_tokens = 0
+            victim.alloc_gen += 1
             q = self.waiting_p0 if self.priorities.get(victim.req_id, 0) == 0 else self.waiting_p1
             q.appendleft(victim)
             return True
@@ -93,14 +94,25 @@ class Scheduler:
         scheduled: set = set()
 
         # 1) decodes for fully-prefilled running requests (cheap, latency-bound)
+        tables = self.kv.block_tables
+        bs = self.kv.block_size
         for req in list(self.running):
             if req.in_prefill or budget <= 0:
+                continue
+            # fast path: the next token's page already exists (true except once
+            # per kv_block_size steps) — skip the allocator entirely
+            table = tables.get(req.req_id)
+            if table is not None and req.total_len <= len(table) * bs:
+                decode_reqs.append(req)
+                scheduled.add(id(req))
+                budget -= 1
                 continue
             if not self._grow_or_preempt(req, req.total_len, scheduled):
                 # could not even hold this one: preempt it too
                 self.running.remove(req)
                 self.kv.release(req.req_id)
                 req.num_computed_tokens = 0
+                req.alloc_gen += 1
                 (self.waiting_p0 if self.priorities.get(req.req_id, 0) == 0
                  else self.waiting_p1).appendleft(req)
                 continue

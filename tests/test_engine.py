@@ -282,3 +282,83 @@ def test_priority_preemption_prefers_low_priority_victims():
 def test_engine_empty_step_is_noop(tiny_engine):
     stats = tiny_engine.step()
     assert stats.scheduled_tokens == 0 and not stats.finished
+
+
+def test_graph_fill_host_incremental():
+    """The hipGraph runner's host-fill path (vectorized + incremental block
+    tables) must reproduce exactly what a naive per-row fill would write —
+    across table growth, row-composition changes, preemption (alloc_gen bump)
+    and padding rows."""
+    import numpy as np
+
+    from sutro_amd.engine.batch import ScheduledBatch
+    from sutro_amd.engine.graph_runner import DecodeGraphRunner
+    from sutro_amd.engine.request import Request
+
+    bs = 4
+    B, W = 8, 16
+
+    class KVStub:
+        block_size = bs
+        block_tables = {}
+
+    class EngStub:
+        kv = KVStub()
+        scratch_block = 0
+
+    r = DecodeGraphRunner.__new__(DecodeGraphRunner)
+    r.engine = EngStub()
+    r.h_ids = torch.zeros(B, dtype=torch.long)
+    r.h_pos = torch.zeros(B, dtype=torch.long)
+    r.h_slots = torch.zeros(B, dtype=torch.long)
+    r.h_bt = torch.zeros(B, W, dtype=torch.int32)
+    r.h_sl = torch.ones(B, dtype=torch.int32)
+    r._row_req = np.full(B, -1, dtype=np.int64)
+    r._row_nb = np.zeros(B, dtype=np.int32)
+    r._row_gen = np.zeros(B, dtype=np.int64)
+    r._arange = np.arange(B, dtype=np.int64)
+
+    def mk(req_id, prompt_len, out):
+        req = Request(req_id=req_id, prompt_token_ids=list(range(10, 10 + prompt_len)),
+                      sampling=SamplingParams(max_tokens=64))
+        req.output_token_ids = list(out)
+        req.num_computed_tokens = prompt_len + len(out) - 1
+        KVStub.block_tables[req_id] = list(
+            range(req_id * 100, req_id * 100 + (req.total_len + bs - 1) // bs))
+        return req
+
+    def check(reqs, bucket):
+        sb = ScheduledBatch(reqs=reqs, num_new_tokens=[1] * len(reqs), num_prefills=0)
+        r._fill_host(sb, bucket)
+        for i, req in enumerate(reqs):
+            p = req.num_computed_tokens
+            table = KVStub.block_tables[req.req_id]
+            assert r.h_ids[i].item() == req.token_at(p)
+            assert r.h_pos[i].item() == p
+            assert r.h_sl[i].item() == p + 1
+            assert r.h_slots[i].item() == table[p // bs] * bs + p % bs
+            assert r.h_bt[i, :len(table)].tolist() == table
+        for i in range(len(reqs), bucket):
+            assert r.h_sl[i].item() == 1
+            assert r.h_bt[i, 0].item() == 0  # scratch
+            assert r.h_slots[i].item() == 0
+
+    a, b, c = mk(1, 5, [60]), mk(2, 6, [61, 62]), mk(3, 3, [])
+    check([a, b, c], 4)
+
+    # steady decode: everyone gains a token; b crosses a block boundary
+    for req in (a, b, c):
+        req.output_token_ids.append(70 + req.req_id)
+        req.num_computed_tokens += 1
+        KVStub.block_tables[req.req_id] = list(
+            range(req.req_id * 100, req.req_id * 100 + (req.total_len + bs - 1) // bs))
+    check([a, b, c], 4)
+
+    # composition change: c finishes, new request d lands in its row
+    d = mk(4, 7, [80])
+    check([a, b, d], 4)
+
+    # preemption: a is re-admitted with a brand-new table of the same length
+    KVStub.block_tables[1] = [v + 1000 for v in KVStub.block_tables[1]]
+    a.alloc_gen += 1
+    check([a, b, d], 8)
