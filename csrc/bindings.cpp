@@ -37,6 +37,8 @@ void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         const int*, int, int, int, int, int, int, float,
                         hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
+void sutro_gemm_tn_launch(void*, const void*, const void*, const void*, int,
+                          int, long, int, int, int, hipStream_t);
 void sutro_mfma16_probe(float*, const void*, const void*, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
@@ -158,6 +160,32 @@ void qkv_prep(torch::Tensor qkv, torch::Tensor q_out, torch::Tensor k_cache,
                  is_fp8(k_cache) ? 1 : 0, cur_stream());
 }
 
+// out[M,N] = x[M,K] @ w[N,K]^T (+ res), bf16 in/out, fp32 accumulate.
+// bm/bn pick the macro-tile; M%bm==0 and K%64==0 required (caller falls back
+// to torch.mm otherwise). res, when given, is added to the output (fused
+// residual epilogue).
+torch::Tensor gemm_tn(torch::Tensor x, torch::Tensor w,
+                      c10::optional<torch::Tensor> res, long bm, long bn,
+                      long xcd_swz) {
+  CHECK_CUDA(x); CHECK_BF16(x); CHECK_CUDA(w); CHECK_BF16(w);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous(), "contiguous required");
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "K mismatch");
+  TORCH_CHECK(M % bm == 0 && K % 64 == 0, "shape not tile-divisible");
+  auto out = torch::empty({M, N}, x.options());
+  const void* rp = nullptr;
+  if (res.has_value()) {
+    CHECK_CUDA(*res); CHECK_BF16(*res);
+    TORCH_CHECK(res->is_contiguous() && res->size(0) == M && res->size(1) == N,
+                "res shape");
+    rp = res->data_ptr();
+  }
+  sutro_gemm_tn_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), rp, (int)M,
+                       (int)N, K, (int)bm, (int)bn, (int)xcd_swz,
+                       cur_stream());
+  return out;
+}
+
 torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
   CHECK_CUDA(a); CHECK_BF16(a);
   auto c = torch::zeros({32, 32}, a.options().dtype(at::kFloat));
@@ -184,4 +212,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention", &paged_attention, "paged prefill+decode attention");
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
+  m.def("gemm_tn", &gemm_tn, "bf16 TN GEMM (MFMA, glds dbuf)",
+        py::arg("x"), py::arg("w"), py::arg("res") = c10::nullopt,
+        py::arg("bm") = 256, py::arg("bn") = 256, py::arg("xcd_swz") = 0);
 }

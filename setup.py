@@ -20,6 +20,7 @@ SOURCES = [
     "csrc/attn_decode.hip",
     "csrc/attn_decode_mfma.hip",
     "csrc/attn_prefill.hip",
+    "csrc/gemm_tn.hip",
 ]
 
 setup(

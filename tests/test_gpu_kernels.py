@@ -447,3 +447,37 @@ def test_attn_decode_mfma_matches_ref(monkeypatch):
         err = (got.float().cpu() - ref).abs().max()
         assert torch.allclose(got.float().cpu(), ref, atol=3e-2,
                               rtol=3e-2), (hq, hk, err)
+
+
+@pytest.mark.gpu
+def test_gemm_tn_matches_torch():
+    """gemm_tn (MFMA glds double-buffer) vs fp32 torch reference across tile
+    configs, XCD swizzle, N-edge and the fused residual epilogue."""
+    from sutro_amd import _C
+
+    torch.manual_seed(7)
+    for (M, N, K), cfgs in [
+        ((256, 512, 256), [(256, 256), (128, 256), (128, 128), (256, 128)]),
+        ((1024, 1000, 640), [(256, 256), (128, 128)]),  # N edge, K%128 != 0
+        ((128, 256, 128), [(128, 128)]),
+    ]:
+        x = (torch.randn(M, K, device="cuda") * 0.5).bfloat16()
+        w = (torch.randn(N, K, device="cuda") * 0.1).bfloat16()
+        ref = x.float() @ w.t().float()
+        scale = ref.abs().max().item()
+        for bm, bn in cfgs:
+            if M % bm:
+                continue
+            for swz in (0, 1):
+                out = _C.gemm_tn(x, w, None, bm, bn, swz)
+                rel = (out.float() - ref).abs().max().item() / scale
+                assert rel < 2e-2, (M, N, K, bm, bn, swz, rel)
+
+    # fused residual
+    x = (torch.randn(256, 384, device="cuda") * 0.5).bfloat16()
+    w = (torch.randn(512, 384, device="cuda") * 0.1).bfloat16()
+    res = torch.randn(256, 512, device="cuda").bfloat16()
+    out = _C.gemm_tn(x, w, res, 128, 128, 0)
+    ref = x.float() @ w.t().float() + res.float()
+    rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert rel < 2e-2
