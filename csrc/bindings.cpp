@@ -38,7 +38,7 @@ void sutro_attn_prefill(void*, const void*, const void*, const void*,
                         hipStream_t);
 void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
 void sutro_gemm_tn_launch(void*, const void*, const void*, const void*, int,
-                          int, long, int, int, int, hipStream_t);
+                          int, long, int, int, int, int, hipStream_t);
 void sutro_mfma16_probe(float*, const void*, const void*, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
@@ -166,7 +166,7 @@ void qkv_prep(torch::Tensor qkv, torch::Tensor q_out, torch::Tensor k_cache,
 // residual epilogue).
 torch::Tensor gemm_tn(torch::Tensor x, torch::Tensor w,
                       c10::optional<torch::Tensor> res, long bm, long bn,
-                      long xcd_swz) {
+                      long swz, long xcd_swz) {
   CHECK_CUDA(x); CHECK_BF16(x); CHECK_CUDA(w); CHECK_BF16(w);
   TORCH_CHECK(x.is_contiguous() && w.is_contiguous(), "contiguous required");
   const long M = x.size(0), K = x.size(1), N = w.size(0);
@@ -181,7 +181,7 @@ torch::Tensor gemm_tn(torch::Tensor x, torch::Tensor w,
     rp = res->data_ptr();
   }
   sutro_gemm_tn_launch(out.data_ptr(), x.data_ptr(), w.data_ptr(), rp, (int)M,
-                       (int)N, K, (int)bm, (int)bn, (int)xcd_swz,
+                       (int)N, K, (int)bm, (int)bn, (int)swz, (int)xcd_swz,
                        cur_stream());
   return out;
 }
@@ -214,5 +214,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
   m.def("gemm_tn", &gemm_tn, "bf16 TN GEMM (MFMA, glds dbuf)",
         py::arg("x"), py::arg("w"), py::arg("res") = c10::nullopt,
-        py::arg("bm") = 256, py::arg("bn") = 256, py::arg("xcd_swz") = 0);
+        py::arg("bm") = 256, py::arg("bn") = 256, py::arg("swz") = 2,
+        py::arg("xcd_swz") = 0);
 }

@@ -30,9 +30,16 @@ __device__ __forceinline__ void glds16(const u32* g, u32* lds) {
       (__attribute__((address_space(3))) u32*)lds, 16, 0, 0);
 }
 
-// st_16x32 swizzle of a byte offset within a tile (involution)
+// LDS swizzle of a byte offset within a tile (involutions, bits >=7 fixed):
+// mode 1: the guide's st_16x32 (bit9 -> bit5), 8-way -> 4-way conflicts.
+// mode 2: row bits 1..3 XOR'd into the 16B-chunk index — a ds_read_b128
+//   quarter (16 consecutive rows, fixed k-chunk) hits all 16 bank groups:
+//   group = 8*(row&1) + (chunk ^ ((row>>1)&7)), bijective over 16 rows.
+template <int SWZ>
 __device__ __forceinline__ u32 swz(u32 byte) {
-  return byte ^ (((byte >> 9) & 1u) << 5);
+  if (SWZ == 1) return byte ^ (((byte >> 9) & 1u) << 5);
+  if (SWZ == 2) return byte ^ (((byte >> 8) & 7u) << 4);
+  return byte;
 }
 
 // One operand tile: ROWS x 64 bf16, K-contiguous rows of 128B.
@@ -42,40 +49,43 @@ __device__ __forceinline__ u32 swz(u32 byte) {
 // [c*1024, c*1024+1024), lane l writes byte c*1024 + l*16, whose logical
 // (row, koff) after un-swizzling is row = c*8 + l/8,
 // koff = (l%8)*16 ^ ((l/32)<<5)  (bit9 of the lane byte is l's bit 5).
-template <int ROWS>
+template <int ROWS, int SWZ>
 __device__ __forceinline__ void stage_tile(
     u16* lds, const u16* gsrc, long row_stride_b, int wave, int lane,
     int rows_valid) {
   constexpr int CH = ROWS / 64;  // 1KB chunks per wave (8 waves)
-  const int koff = ((lane & 7) * 16) ^ (((lane >> 5) & 1) << 5);
-  const char* g0 = (const char*)gsrc + koff;
-  u32* ldst = (u32*)((char*)lds + wave * (CH * 1024) + lane * 16);
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
     const int c = wave * CH + j;
-    int row = c * 8 + (lane >> 3);
+    const u32 b = (u32)c * 1024u + (u32)lane * 16u;  // lane-linear LDS byte
+    const u32 lb = swz<SWZ>(b);  // pre-swizzled global source (rule 21)
+    int row = (int)(lb >> 7);
     row = min(row, rows_valid - 1);  // clamp N-edge (dup loads, guarded store)
-    glds16((const u32*)(g0 + (long)row * row_stride_b), ldst + j * 256);
+    glds16((const u32*)((const char*)gsrc + (long)row * row_stride_b +
+                        (lb & 127u)),
+           (u32*)((char*)lds + b));
   }
 }
 
 // ds_read address of an 8-elem K fragment: logical row `row`, k chunk `kk16`
 // (16B units within the 128B row)
+template <int SWZ>
 __device__ __forceinline__ const bf16x8* frag_addr(const u16* lds, int row,
                                                    int kk16) {
   u32 lb = (u32)row * 128u + (u32)kk16 * 16u;
-  return (const bf16x8*)((const char*)lds + swz(lb));
+  return (const bf16x8*)((const char*)lds + swz<SWZ>(lb));
 }
 
 // BM x BN macro-tile, BK=64, 8 waves as 2(M) x 4(N); per-wave BM/2 x BN/4.
 // EPI: 0 = plain store, 1 = residual add (out += res read at same offset)
-template <int BM, int BN, int EPI>
+template <int BM, int BN, int EPI, int SWZ>
 __global__ __launch_bounds__(512) void gemm_tn_kernel(
     u16* __restrict__ out,        // [M, N] bf16
     const u16* __restrict__ x,    // [M, K] bf16
     const u16* __restrict__ w,    // [N, K] bf16
     const u16* __restrict__ res,  // [M, N] bf16 or nullptr
     int M, int N, long K, int n_tiles, int xcd_swz) {
+  (void)n_tiles;
   constexpr int WM = BM / 2;   // wave-tile M (128 for BM=256)
   constexpr int WN = BN / 4;   // wave-tile N (64 for BN=256)
   constexpr int MF = WM / 16;  // m-fragments per wave
@@ -111,8 +121,8 @@ __global__ __launch_bounds__(512) void gemm_tn_kernel(
   const u16* wa = w + (long)n0 * K;
 
   // prologue: stage K-tile 0 into buffer 0, drain
-  stage_tile<BM>(smem[0], xa, Kb, wave, lane, BM);
-  stage_tile<BN>(smem[0] + BM * 64, wa, Kb, wave, lane, N - n0);
+  stage_tile<BM, SWZ>(smem[0], xa, Kb, wave, lane, BM);
+  stage_tile<BN, SWZ>(smem[0] + BM * 64, wa, Kb, wave, lane, N - n0);
   __syncthreads();
 
   const int KT = (int)(K >> 6);
@@ -124,9 +134,10 @@ __global__ __launch_bounds__(512) void gemm_tn_kernel(
     // bottom __syncthreads() (whose fence emits vmcnt(0) with a glds
     // outstanding) both drains it and orders buffer reuse
     if (t + 1 < KT) {
-      stage_tile<BM>(smem[p ^ 1], xa + (long)(t + 1) * 64, Kb, wave, lane, BM);
-      stage_tile<BN>(smem[p ^ 1] + BM * 64, wa + (long)(t + 1) * 64, Kb, wave,
-                     lane, N - n0);
+      stage_tile<BM, SWZ>(smem[p ^ 1], xa + (long)(t + 1) * 64, Kb, wave,
+                          lane, BM);
+      stage_tile<BN, SWZ>(smem[p ^ 1] + BM * 64, wa + (long)(t + 1) * 64, Kb,
+                          wave, lane, N - n0);
     }
 
 #pragma unroll
@@ -135,10 +146,10 @@ __global__ __launch_bounds__(512) void gemm_tn_kernel(
       bf16x8 bf[NF];
 #pragma unroll
       for (int j = 0; j < NF; ++j)
-        bf[j] = *frag_addr(bT, wn * WN + j * 16 + (lane & 15), kk16);
+        bf[j] = *frag_addr<SWZ>(bT, wn * WN + j * 16 + (lane & 15), kk16);
 #pragma unroll
       for (int i = 0; i < MF; ++i) {
-        const bf16x8 af = *frag_addr(aT, wm * WM + i * 16 + (lane & 15), kk16);
+        const bf16x8 af = *frag_addr<SWZ>(aT, wm * WM + i * 16 + (lane & 15), kk16);
 #pragma unroll
         for (int j = 0; j < NF; ++j) acc[i][j] = mfma16(af, bf[j], acc[i][j]);
       }
@@ -166,32 +177,30 @@ __global__ __launch_bounds__(512) void gemm_tn_kernel(
   }
 }
 
-#define INST(BM, BN, EPI)                                             \
-  template __global__ void gemm_tn_kernel<BM, BN, EPI>(               \
-      u16*, const u16*, const u16*, const u16*, int, int, long, int, int);
-
 extern "C" void sutro_gemm_tn_launch(void* out, const void* x, const void* w,
                                      const void* res, int M, int N, long K,
-                                     int bm, int bn, int xcd_swz,
+                                     int bm, int bn, int swz_mode, int xcd_swz,
                                      hipStream_t stream) {
   const int n_tiles = (N + bn - 1) / bn;
   const int blocks = (M / bm) * n_tiles;
   dim3 grid(blocks), block(512);
-#define LAUNCH(BM, BN)                                                       \
-  if (bm == BM && bn == BN) {                                                \
+#define LAUNCH_1(BM, BN, SWZ)                                                \
+  if (bm == BM && bn == BN && swz_mode == SWZ) {                             \
     if (res)                                                                 \
-      gemm_tn_kernel<BM, BN, 1><<<grid, block, 0, stream>>>(                 \
+      gemm_tn_kernel<BM, BN, 1, SWZ><<<grid, block, 0, stream>>>(            \
           (u16*)out, (const u16*)x, (const u16*)w, (const u16*)res, M, N, K, \
           n_tiles, xcd_swz);                                                 \
     else                                                                     \
-      gemm_tn_kernel<BM, BN, 0><<<grid, block, 0, stream>>>(                 \
+      gemm_tn_kernel<BM, BN, 0, SWZ><<<grid, block, 0, stream>>>(            \
           (u16*)out, (const u16*)x, (const u16*)w, nullptr, M, N, K,         \
           n_tiles, xcd_swz);                                                 \
     return;                                                                  \
   }
+#define LAUNCH(BM, BN) LAUNCH_1(BM, BN, 0) LAUNCH_1(BM, BN, 1) LAUNCH_1(BM, BN, 2)
   LAUNCH(256, 256)
   LAUNCH(128, 256)
   LAUNCH(128, 128)
   LAUNCH(256, 128)
 #undef LAUNCH
+#undef LAUNCH_1
 }

@@ -468,16 +468,17 @@ def test_gemm_tn_matches_torch():
         for bm, bn in cfgs:
             if M % bm:
                 continue
-            for swz in (0, 1):
-                out = _C.gemm_tn(x, w, None, bm, bn, swz)
-                rel = (out.float() - ref).abs().max().item() / scale
-                assert rel < 2e-2, (M, N, K, bm, bn, swz, rel)
+            for swz in (0, 1, 2):
+                for xswz in (0, 1):
+                    out = _C.gemm_tn(x, w, None, bm, bn, swz, xswz)
+                    rel = (out.float() - ref).abs().max().item() / scale
+                    assert rel < 2e-2, (M, N, K, bm, bn, swz, xswz, rel)
 
     # fused residual
     x = (torch.randn(256, 384, device="cuda") * 0.5).bfloat16()
     w = (torch.randn(512, 384, device="cuda") * 0.1).bfloat16()
     res = torch.randn(256, 512, device="cuda").bfloat16()
-    out = _C.gemm_tn(x, w, res, 128, 128, 0)
+    out = _C.gemm_tn(x, w, res, 128, 128, 2, 0)
     ref = x.float() @ w.t().float() + res.float()
     rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
     assert rel < 2e-2

@@ -67,8 +67,8 @@ def main():
         for bm, bn in CONFIGS:
             if M % bm:
                 continue
-            for swz in (0, 1):
-                out = _C.gemm_tn(x, w, None, bm, bn, swz)
+            for swz in (1, 2):
+                out = _C.gemm_tn(x, w, None, bm, bn, swz, 0)
                 # bf16 GEMM vs its own bf16 ref; also bound error vs fp32
                 err_ref = (out.float() - ref.float()).abs().max().item()
                 err_f32 = (out.float() - reff).abs().max().item()
@@ -81,9 +81,11 @@ def main():
                 if args.check_only:
                     row.append(f"{tag}: ok rel={rel:.1e}")
                 else:
-                    t = time_fn(lambda: _C.gemm_tn(x, w, None, bm, bn, swz),
+                    t = time_fn(lambda: _C.gemm_tn(x, w, None, bm, bn, swz, 0),
                                 args.iters)
-                    row.append(f"{tag}: {t*1000:8.1f}us {flops/t/1e9:7.1f} TF")
+                    t2 = time_fn(lambda: _C.gemm_tn(x, w, None, bm, bn, swz, 1),
+                                 args.iters)
+                    row.append(f"{tag}: {flops/t/1e9:7.1f} TF | xcd {flops/t2/1e9:7.1f} TF")
         emit("\n    ".join(row))
 
     # fused residual epilogue check
@@ -91,7 +93,7 @@ def main():
     x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
     w = (torch.randn(N, K, device=dev) * 0.02).bfloat16()
     res = torch.randn(M, N, device=dev).bfloat16()
-    out = _C.gemm_tn(x, w, res, 128, 128, 0)
+    out = _C.gemm_tn(x, w, res, 128, 128, 2, 0)
     ref = (x.float() @ w.t().float() + res.float())
     rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
     emit(f"residual epilogue: rel={rel:.3e} {'ok' if rel < 2e-2 else 'FAIL'}")
