@@ -23,6 +23,8 @@ SHAPES = [
     ("down", 1024, 5120, 25600),
     ("lm_head", 1024, 151936, 5120),
 ]
+# M-scaling of the library GEMMs (is a bigger decode batch more efficient?)
+M_SWEEP = [256, 512, 1024, 2048, 4096]
 CONFIGS = [(256, 256), (128, 256), (128, 128), (256, 128)]
 
 
@@ -97,6 +99,16 @@ def main():
     ref = (x.float() @ w.t().float() + res.float())
     rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
     emit(f"residual epilogue: rel={rel:.3e} {'ok' if rel < 2e-2 else 'FAIL'}")
+
+    if not args.check_only:
+        for name, _, N, K in SHAPES:
+            row = [f"M-sweep {name:9s} N={N} K={K} (hipblaslt)"]
+            for m in M_SWEEP:
+                x = (torch.randn(m, K, device=dev) * 0.5).bfloat16()
+                w = (torch.randn(N, K, device=dev) * 0.02).bfloat16()
+                t = time_fn(lambda: x @ w.t(), args.iters)
+                row.append(f"M={m}: {2.0*m*N*K/t/1e9:7.1f} TF")
+            emit("  ".join(row))
 
     outd = os.environ.get("GPURUN_OUT", "gpurun_out")
     os.makedirs(outd, exist_ok=True)
