@@ -37,6 +37,9 @@ def build_args():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--max-new", type=int, default=128)
     p.add_argument("--tokens-per-step", type=int, default=32768)
+    p.add_argument("--min-prefill", type=int, default=None,
+                   help="prefill accumulation threshold (tokens); default = "
+                        "engine default")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--kv-blocks", type=int, default=None)
     p.add_argument("--no-refill", action="store_true",
@@ -101,6 +104,8 @@ def main():
         max_model_len=max(256, args.prompt_len + args.max_new + 32),
         max_num_seqs=args.batch,
         max_tokens_per_step=args.tokens_per_step,
+        **({"min_prefill_batch_tokens": args.min_prefill}
+           if args.min_prefill is not None else {}),
         num_kv_blocks=args.kv_blocks if device != "cpu" else 512,
         seed=dp_idx,  # identical within a TP group (lockstep), unique per replica
         tp_size=tp,
