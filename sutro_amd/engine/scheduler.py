@@ -135,7 +135,11 @@ class Scheduler:
         # 3) admit new requests (p0 ahead of p1). While decodes run, hold
         # back until enough prefill work accumulates to amortize the pass.
         thr = self.cfg.min_prefill_batch_tokens
-        if decode_reqs and not prefill_reqs and thr > 0:
+        if thr is None:
+            thr = self.cfg.max_tokens_per_step
+        # p0 (interactive) rows bypass accumulation — only production (p1)
+        # admission is batched for throughput
+        if decode_reqs and not prefill_reqs and thr > 0 and not self.waiting_p0:
             avail = 0
             for q in (self.waiting_p0, self.waiting_p1):
                 for r in q:

@@ -362,3 +362,42 @@ def test_graph_fill_host_incremental():
     KVStub.block_tables[1] = [v + 1000 for v in KVStub.block_tables[1]]
     a.alloc_gen += 1
     check([a, b, d], 8)
+
+
+def test_prefill_accumulation_p0_bypass():
+    """While decodes run, small p1 prefills are held back until the
+    accumulation threshold, but p0 (interactive) rows are admitted at once."""
+    from sutro_amd.engine.scheduler import Scheduler
+
+    spec = tiny_spec_for_tests()
+    cfg = EngineConfig(spec=spec, device="cpu", max_num_seqs=16,
+                       max_model_len=256, max_tokens_per_step=64,
+                       num_kv_blocks=256)  # threshold = None -> 64 tokens
+    kv = PagedKVCache(num_layers=1, num_blocks=256, num_kv_heads=1,
+                      block_size=cfg.kv_block_size, head_dim=8,
+                      dtype=torch.float32, device="cpu")
+    sch = Scheduler(cfg, kv)
+
+    def mk(rid, plen):
+        return Request(req_id=rid, prompt_token_ids=list(range(3, 3 + plen)),
+                       sampling=SamplingParams(max_tokens=8))
+
+    # a running decode
+    d = mk(0, 4)
+    sch.add_request(d, priority=1)
+    sb = sch.schedule()
+    assert sb.num_prefills == 1
+    d.num_computed_tokens = 4
+    d.output_token_ids.append(7)
+
+    # small p1 prompt: below the 64-token threshold -> held back
+    sch.add_request(mk(1, 8), priority=1)
+    sb = sch.schedule()
+    assert sb.num_prefills == 0 and len(sb.reqs) == 1
+
+    # a p0 prompt bypasses accumulation (and pulls the waiting p1 along)
+    sch.add_request(mk(2, 8), priority=0)
+    sb = sch.schedule()
+    admitted = {r.req_id for r in sb.reqs[:sb.num_prefills]}
+    assert 2 in admitted
+    assert sb.num_prefills >= 1
