@@ -367,6 +367,7 @@ def test_graph_fill_host_incremental():
 def test_prefill_accumulation_p0_bypass():
     """While decodes run, small p1 prefills are held back until the
     accumulation threshold, but p0 (interactive) rows are admitted at once."""
+    from sutro_amd.engine.request import Request
     from sutro_amd.engine.scheduler import Scheduler
 
     spec = tiny_spec_for_tests()
