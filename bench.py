@@ -29,7 +29,7 @@ import torch
 def build_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=32)
+    p.add_argument("--steps", type=int, default=48)
     p.add_argument("--warmup", type=int, default=12)
     p.add_argument("--model", type=str, default="qwen-3-32b")
     p.add_argument("--batch", type=int, default=2048,
