@@ -34,6 +34,21 @@ __device__ __forceinline__ f32x4 mfma16(bf16frag a, bf16frag b, f32x4 c) {
   return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
 }
 
+// 8 consecutive KV elements at byte pointer p -> bf16 fragment.
+// bf16 cache: one 16B load. e4m3 cache: 8B load + native pk converts.
+template <bool KV8>
+__device__ __forceinline__ bf16frag load_kv_frag(const u8* p) {
+  if constexpr (!KV8) return *(const s16x8*)p;
+  const u32x2 raw = *(const u32x2*)p;
+  float f[8];
+  fp8x4_to_f32(raw[0], f);
+  fp8x4_to_f32(raw[1], f + 4);
+  bf16frag r;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) r[j] = (short)f2bf(f[j]);
+  return r;
+}
+
 struct MfmaSmem {
   u16 qt[16 * DM];          // Q^T as [head16][k128], row-swizzled (4 KB)
   u16 vt[DM * PV_PAD];      // V^T [d128][pos32+pad] (10 KB)
@@ -46,11 +61,12 @@ __device__ __forceinline__ int qt_swz(int head, int byte_in_row) {
   return head * QT_PAD_B + (byte_in_row ^ ((head & 15) << 4));
 }
 
+template <bool KV8>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
-    const u16* __restrict__ k_cache,  // [nb, Hk, BS, D] bf16
-    const u16* __restrict__ v_cache,
+    const u8* __restrict__ k_cache,   // [nb, Hk, BS, D] bf16 or e4m3
+    const u8* __restrict__ v_cache,
     const int* __restrict__ block_tables,  // [S, bt_stride]
     const int* __restrict__ seq_lens,      // [S]
     int bt_stride, int n_dec, int Hq, int Hk, int seq_offset, float scale) {
@@ -93,8 +109,9 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
   for (int b = 0; b < 8; ++b) acc[b] = (f32x4)(0.f);
 
+  constexpr int ES = KV8 ? 1 : 2;  // bytes per cache element
   for (int pg = 0; pg < npages; ++pg) {
-    const long kv_base = (((long)bt[pg] * Hk + kh) * BS) * DM;
+    const long kv_base = (((long)bt[pg] * Hk + kh) * BS) * DM * ES;
     const int valid = min(BS, L - pg * BS);
 
     // ---- stage V^T (d-major) while issuing K fragment loads
@@ -102,15 +119,16 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     // elems: item = lane*2+c -> pos = item/16? Simpler: 64 lanes x 8 iters of
     // 8 elems = 4096 elems = the page.
     {
-      const u16* vsrc = v_cache + kv_base;
+      const u8* vsrc = v_cache + kv_base;
 #pragma unroll
       for (int it = 0; it < 8; ++it) {
         const int flat = it * (int)WAVE + lane;   // 8-elem chunk id (0..511)
         const int pos = flat & 31;                // pos-major across lanes:
         const int d0 = (flat >> 5) * 8;           // scatter writes spread banks
-        u16x8 vx = *(const u16x8*)(vsrc + pos * DM + d0);
+        const bf16frag vx = load_kv_frag<KV8>(vsrc + (pos * DM + d0) * ES);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) sm->vt[(d0 + j) * PV_PAD + pos] = vx[j];
+        for (int j = 0; j < 8; ++j)
+          sm->vt[(d0 + j) * PV_PAD + pos] = (u16)vx[j];
       }
     }
 
@@ -122,9 +140,10 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         // A = K frag: row pos = half*16 + l%16, dims kk*32 + (l/16)*8..+8
-        const u16* krow = k_cache + kv_base +
-                          (long)(half * 16 + lo16) * DM + kk * 32 + hi4 * 8;
-        bf16frag ka = *(const s16x8*)krow;
+        const u8* krow = k_cache + kv_base +
+                         ((long)(half * 16 + lo16) * DM + kk * 32 + hi4 * 8) *
+                             ES;
+        bf16frag ka = load_kv_frag<KV8>(krow);
         // B = Qt frag: B[k][head]: lane reads qt[head l%16][kk*32+(l/16)*8]
         bf16frag qb = *(const s16x8*)((char*)sm->qt +
                                       qt_swz(lo16, (kk * 32 + hi4 * 8) * 2));
@@ -211,7 +230,7 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
                                        const int* block_tables,
                                        const int* seq_lens, int bt_stride,
                                        int n_dec, int Hq, int Hk,
-                                       int seq_offset, float scale,
+                                       int seq_offset, float scale, int kv_fp8,
                                        hipStream_t s) {
   // 2 waves/block: 31.4 KB LDS -> 5 blocks/CU (10 waves) vs 2 at wpb=4
   static const char* wpb_env = getenv("SUTRO_DECODE_WPB");
@@ -219,10 +238,16 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
   const long items = (long)n_dec * Hk;
   const long blocks = (items + wpb - 1) / wpb;
   const size_t smem = sizeof(MfmaSmem) * wpb;
-  hipLaunchKernelGGL(attn_decode_mfma_kernel, dim3((unsigned)blocks),
-                     dim3(wpb * WAVE), smem, s, (u16*)out, (const u16*)q,
-                     (const u16*)k_cache, (const u16*)v_cache, block_tables,
-                     seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+  if (kv_fp8)
+    hipLaunchKernelGGL(attn_decode_mfma_kernel<true>, dim3((unsigned)blocks),
+                       dim3(wpb * WAVE), smem, s, (u16*)out, (const u16*)q,
+                       (const u8*)k_cache, (const u8*)v_cache, block_tables,
+                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+  else
+    hipLaunchKernelGGL(attn_decode_mfma_kernel<false>, dim3((unsigned)blocks),
+                       dim3(wpb * WAVE), smem, s, (u16*)out, (const u16*)q,
+                       (const u8*)k_cache, (const u8*)v_cache, block_tables,
+                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
 }
 
 // ---- probe: C[16,16] = A[16,32] @ B[32,16] with the exact frag loaders ----

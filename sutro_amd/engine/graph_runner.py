@@ -26,8 +26,13 @@ class DecodeGraphRunner:
         cfg = engine.cfg
         max_seqs = cfg.max_num_seqs
         if buckets is None:
-            buckets = [b for b in (8, 16, 32, 64, 128, 256, 512, 1024, 2048)
+            # powers of two up to 1024, then 128-step granularity: at the
+            # batch-2048 operating point the resident set oscillates in
+            # [~1792, 2048] between refill waves, and a 128-step bucket caps
+            # decode padding waste at ~7% (vs 2x-bucket's ~50% worst case)
+            buckets = [b for b in (8, 16, 32, 64, 128, 256, 512, 1024)
                        if b <= max_seqs]
+            buckets += list(range(1152, max_seqs + 1, 128))
             if not buckets or buckets[-1] < max_seqs:
                 buckets.append(max_seqs)
         self.buckets = buckets
