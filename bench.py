@@ -20,6 +20,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import numpy as np
@@ -170,9 +171,14 @@ def main():
     sched_tokens = 0
     count_me = (rank % tp) == 0  # one counter per TP replica
     rows_done = 0
+    step_kinds = []  # (prefill_tokens, wall_ms) per step, reported on stderr
     for _ in range(args.steps):
         refill()
+        ts = time.time()
         stats = eng.step()
+        if have_gpu:
+            torch.cuda.synchronize()
+        step_kinds.append((stats.prefill_tokens, (time.time() - ts) * 1e3))
         if count_me:
             out_tokens += stats.output_tokens
             sched_tokens += stats.scheduled_tokens
@@ -180,6 +186,11 @@ def main():
     if have_gpu:
         torch.cuda.synchronize()
     t1 = time.time()
+    if rank == 0:
+        dec = [m for p_, m in step_kinds if p_ == 0]
+        mix = [(p_, m) for p_, m in step_kinds if p_ > 0]
+        print(f"[steps] pure-decode: {len(dec)} x {sum(dec)/max(1,len(dec)):.1f} ms; "
+              f"mixed: {[(p_, round(m,1)) for p_, m in mix]}", file=sys.stderr)
     elapsed = t1 - t0
     sync()
 
