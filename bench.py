@@ -171,14 +171,18 @@ def main():
     sched_tokens = 0
     count_me = (rank % tp) == 0  # one counter per TP replica
     rows_done = 0
-    step_kinds = []  # (prefill_tokens, wall_ms) per step, reported on stderr
+    # SUTRO_BENCH_TRACE=1: per-step wall times on stderr (adds a per-step
+    # sync, so it is NOT on for headline runs)
+    trace = os.environ.get("SUTRO_BENCH_TRACE") == "1"
+    step_kinds = []
     for _ in range(args.steps):
         refill()
         ts = time.time()
         stats = eng.step()
-        if have_gpu:
-            torch.cuda.synchronize()
-        step_kinds.append((stats.prefill_tokens, (time.time() - ts) * 1e3))
+        if trace:
+            if have_gpu:
+                torch.cuda.synchronize()
+            step_kinds.append((stats.prefill_tokens, (time.time() - ts) * 1e3))
         if count_me:
             out_tokens += stats.output_tokens
             sched_tokens += stats.scheduled_tokens
@@ -186,7 +190,7 @@ def main():
     if have_gpu:
         torch.cuda.synchronize()
     t1 = time.time()
-    if rank == 0:
+    if trace and rank == 0:
         dec = [m for p_, m in step_kinds if p_ == 0]
         mix = [(p_, m) for p_, m in step_kinds if p_ > 0]
         print(f"[steps] pure-decode: {len(dec)} x {sum(dec)/max(1,len(dec)):.1f} ms; "
