@@ -145,9 +145,13 @@ def main():
     def refill():
         if args.no_refill and arrival[0] >= args.batch:
             return
-        while len(eng.scheduler.running) + len(eng.scheduler.waiting_p0) < args.batch:
+        # p1 = production batch job: admission is accumulation-batched so
+        # decode steps stay pure hipGraph replays (p0 would bypass that)
+        sch = eng.scheduler
+        while (len(sch.running) + len(sch.waiting_p0) + len(sch.waiting_p1)
+               < args.batch):
             eng.add_request(make_prompt(), SamplingParams(**sp_kwargs),
-                            fsm_id=fsm_id, arrival_idx=arrival[0])
+                            fsm_id=fsm_id, priority=1, arrival_idx=arrival[0])
             arrival[0] += 1
 
     refill()
