@@ -156,7 +156,17 @@ class Scheduler:
         for q in (self.waiting_p0, self.waiting_p1):
             while q and budget > 0 and len(self.running) < self.cfg.max_num_seqs:
                 req = q[0]
-                c = min(req.num_prompt_tokens - req.num_computed_tokens, budget)
+                need = req.num_prompt_tokens - req.num_computed_tokens
+                if (need > budget and decode_reqs
+                        and need <= self.cfg.max_tokens_per_step):
+                    # (prompts larger than a whole step budget must chunk
+                    # regardless — admitting them partial is not a split)
+                    # don't split a prompt across the budget while decodes run:
+                    # the partial row's continuation next step would drag the
+                    # remaining waiting rows into an extra tiny eager pass —
+                    # hold the row for the next accumulated wave instead
+                    break
+                c = min(need, budget)
                 blocks_needed = self.kv.blocks_needed(req.num_computed_tokens + c)
                 if blocks_needed > self.kv.allocator.num_free:
                     break  # don't preempt running work to admit new work

@@ -402,3 +402,53 @@ def test_prefill_accumulation_p0_bypass():
     admitted = {r.req_id for r in sb.reqs[:sb.num_prefills]}
     assert 2 in admitted
     assert sb.num_prefills >= 1
+
+
+def test_no_prompt_split_while_decoding():
+    """While decodes run, admission stops at the largest whole-prompt fit
+    instead of splitting a prompt across the step budget (a split's
+    continuation would drag the held-back queue into an extra eager pass)."""
+    from sutro_amd.engine.request import Request
+    from sutro_amd.engine.scheduler import Scheduler
+
+    spec = tiny_spec_for_tests()
+    cfg = EngineConfig(spec=spec, device="cpu", max_num_seqs=64,
+                       max_model_len=512, max_tokens_per_step=100,
+                       min_prefill_batch_tokens=0, num_kv_blocks=512)
+    kv = PagedKVCache(num_layers=1, num_blocks=512, num_kv_heads=1,
+                      block_size=cfg.kv_block_size, head_dim=8,
+                      dtype=torch.float32, device="cpu")
+    sch = Scheduler(cfg, kv)
+
+    # one running decode
+    d = Request(req_id=0, prompt_token_ids=[3, 4], sampling=SamplingParams(max_tokens=8))
+    sch.add_request(d, priority=1)
+    sch.schedule()
+    d.num_computed_tokens = 2
+    d.output_token_ids.append(5)
+
+    # 2 x 60-token prompts: first fits the 99 remaining budget, second would
+    # have to split -> held back whole
+    for rid in (1, 2):
+        sch.add_request(Request(req_id=rid, prompt_token_ids=list(range(3, 63)),
+                                sampling=SamplingParams(max_tokens=4)), priority=1)
+    sb = sch.schedule()
+    counts = {r.req_id: c for r, c in zip(sb.reqs, sb.num_new_tokens)}
+    assert counts.get(1) == 60
+    assert 2 not in counts  # not admitted partially
+
+    # a prompt longer than a whole step budget must chunk regardless
+    sch2 = Scheduler(cfg, PagedKVCache(num_layers=1, num_blocks=512,
+                                       num_kv_heads=1, block_size=cfg.kv_block_size,
+                                       head_dim=8, dtype=torch.float32, device="cpu"))
+    d2 = Request(req_id=10, prompt_token_ids=[3, 4], sampling=SamplingParams(max_tokens=8))
+    sch2.add_request(d2, priority=1)
+    sch2.schedule()
+    d2.num_computed_tokens = 2
+    d2.output_token_ids.append(5)
+    big = Request(req_id=11, prompt_token_ids=list(range(3, 3 + 150)),
+                  sampling=SamplingParams(max_tokens=4))
+    sch2.add_request(big, priority=1)
+    sb = sch2.schedule()
+    counts = {r.req_id: c for r, c in zip(sb.reqs, sb.num_new_tokens)}
+    assert counts.get(11) == 99  # budget minus the decode token
