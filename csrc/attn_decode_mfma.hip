@@ -22,7 +22,6 @@
 #include <cstdlib>
 
 #define BS 32
-#define DM 128       // head_dim
 #define MAXG 8
 #define QT_PAD_B 256 // Qt row bytes (128 bf16)
 #define PV_PAD 40    // padded row length (elems) for P and Vt tiles
@@ -49,19 +48,23 @@ __device__ __forceinline__ bf16frag load_kv_frag(const u8* p) {
   return r;
 }
 
+template <int D>
 struct MfmaSmem {
-  u16 qt[16 * DM];          // Q^T as [head16][k128], row-swizzled (4 KB)
-  u16 vt[DM * PV_PAD];      // V^T [d128][pos32+pad] (10 KB)
+  u16 qt[16 * D];           // Q^T as [head16][kD], row-swizzled
+  u16 vt[D * PV_PAD];       // V^T [dD][pos32+pad]
   u16 p[16 * PV_PAD];       // P [head16][pos32+pad] (1.25 KB)
   float alpha[16];
   float linv[16];
 };
 
+// row-local XOR swizzle; mask keeps the offset inside the 2*D-byte row
+template <int D>
 __device__ __forceinline__ int qt_swz(int head, int byte_in_row) {
-  return head * QT_PAD_B + (byte_in_row ^ ((head & 15) << 4));
+  constexpr int mask = (D == 128) ? 15 : 7;
+  return head * (2 * D) + (byte_in_row ^ ((head & mask) << 4));
 }
 
-template <bool KV8>
+template <int D, bool KV8>
 __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     u16* __restrict__ out,            // [n_dec, Hq, D]
     const u16* __restrict__ q,        // [n_dec, Hq, D]
@@ -83,7 +86,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   const int* bt = block_tables + (long)sg * bt_stride;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  MfmaSmem* sm = ((MfmaSmem*)smem_raw) + wid;
+  MfmaSmem<D>* sm = ((MfmaSmem<D>*)smem_raw) + wid;
 
   const int lo16 = lane & 15;   // head column / A row
   const int hi4 = lane >> 4;    // 0..3
@@ -92,26 +95,28 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   // lane l writes head l%16, 8 dims at (l/16)*8 + 32*t (4 iterations).
   {
     const int head = lo16;
-    for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int t = 0; t < D / 32; ++t) {
       const int k0 = hi4 * 8 + t * 32;
       u16x8 val = {};
       if (head < G) {
-        const u16* qrow = q + ((long)sd * Hq + kh * G + head) * DM;
+        const u16* qrow = q + ((long)sd * Hq + kh * G + head) * D;
 #pragma unroll
         for (int j = 0; j < 8; ++j) val[j] = f2bf(bf2f(qrow[k0 + j]) * scale);
       }
-      *(u16x8*)((char*)sm->qt + qt_swz(head, k0 * 2)) = val;
+      *(u16x8*)((char*)sm->qt + qt_swz<D>(head, k0 * 2)) = val;
     }
   }
 
   float m_run = -1e30f, l_run = 0.f;   // for head lo16
-  f32x4 acc[8];                        // O[head hi4*4+r][d lo16 + 16*dblk]
+  constexpr int DB = D / 16;           // d-blocks
+  f32x4 acc[DB];                       // O[head hi4*4+r][d lo16 + 16*dblk]
 #pragma unroll
-  for (int b = 0; b < 8; ++b) acc[b] = (f32x4)(0.f);
+  for (int b = 0; b < DB; ++b) acc[b] = (f32x4)(0.f);
 
   constexpr int ES = KV8 ? 1 : 2;  // bytes per cache element
   for (int pg = 0; pg < npages; ++pg) {
-    const long kv_base = (((long)bt[pg] * Hk + kh) * BS) * DM * ES;
+    const long kv_base = (((long)bt[pg] * Hk + kh) * BS) * D * ES;
     const int valid = min(BS, L - pg * BS);
 
     // ---- stage V^T (d-major) while issuing K fragment loads
@@ -121,11 +126,11 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     {
       const u8* vsrc = v_cache + kv_base;
 #pragma unroll
-      for (int it = 0; it < 8; ++it) {
-        const int flat = it * (int)WAVE + lane;   // 8-elem chunk id (0..511)
+      for (int it = 0; it < (BS * D) / ((int)WAVE * 8); ++it) {
+        const int flat = it * (int)WAVE + lane;   // 8-elem chunk id
         const int pos = flat & 31;                // pos-major across lanes:
         const int d0 = (flat >> 5) * 8;           // scatter writes spread banks
-        const bf16frag vx = load_kv_frag<KV8>(vsrc + (pos * DM + d0) * ES);
+        const bf16frag vx = load_kv_frag<KV8>(vsrc + (pos * D + d0) * ES);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           sm->vt[(d0 + j) * PV_PAD + pos] = (u16)vx[j];
@@ -138,15 +143,15 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     for (int half = 0; half < 2; ++half) {
       f32x4 d = (f32x4)(0.f);
 #pragma unroll
-      for (int kk = 0; kk < 4; ++kk) {
+      for (int kk = 0; kk < D / 32; ++kk) {
         // A = K frag: row pos = half*16 + l%16, dims kk*32 + (l/16)*8..+8
         const u8* krow = k_cache + kv_base +
-                         ((long)(half * 16 + lo16) * DM + kk * 32 + hi4 * 8) *
+                         ((long)(half * 16 + lo16) * D + kk * 32 + hi4 * 8) *
                              ES;
         bf16frag ka = load_kv_frag<KV8>(krow);
         // B = Qt frag: B[k][head]: lane reads qt[head l%16][kk*32+(l/16)*8]
         bf16frag qb = *(const s16x8*)((char*)sm->qt +
-                                      qt_swz(lo16, (kk * 32 + hi4 * 8) * 2));
+                                      qt_swz<D>(lo16, (kk * 32 + hi4 * 8) * 2));
         d = mfma16(ka, qb, d);
       }
       s01[half] = d;
@@ -201,7 +206,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
     // ---- PV via MFMA: A = P[head16, pos32], B = Vt-read V[pos32, d16]
     bf16frag pa = *(const s16x8*)(sm->p + lo16 * PV_PAD + hi4 * 8);
 #pragma unroll
-    for (int b = 0; b < 8; ++b) {
+    for (int b = 0; b < DB; ++b) {
       // B frag: V[pos=(l/16)*8+j][d = b*16 + l%16] from vt[d][pos]
       bf16frag vb = *(const s16x8*)(sm->vt + (b * 16 + lo16) * PV_PAD +
                                     hi4 * 8);
@@ -218,9 +223,9 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
   for (int r = 0; r < 4; ++r) {
     const int head = hi4 * 4 + r;
     if (head >= G) continue;
-    u16* orow = out + ((long)sd * Hq + kh * G + head) * DM;
+    u16* orow = out + ((long)sd * Hq + kh * G + head) * D;
 #pragma unroll
-    for (int b = 0; b < 8; ++b) orow[b * 16 + lo16] = f2bf(acc[b][r] * li[r]);
+    for (int b = 0; b < DB; ++b) orow[b * 16 + lo16] = f2bf(acc[b][r] * li[r]);
   }
 }
 
@@ -231,23 +236,27 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
                                        const int* seq_lens, int bt_stride,
                                        int n_dec, int Hq, int Hk,
                                        int seq_offset, float scale, int kv_fp8,
-                                       hipStream_t s) {
-  // 2 waves/block: 31.4 KB LDS -> 5 blocks/CU (10 waves) vs 2 at wpb=4
+                                       int head_dim, hipStream_t s) {
+  // 2 waves/block (bf16 D=128: 31.4 KB LDS -> 5 blocks/CU)
   static const char* wpb_env = getenv("SUTRO_DECODE_WPB");
   const int wpb = wpb_env ? atoi(wpb_env) : 2;
   const long items = (long)n_dec * Hk;
   const long blocks = (items + wpb - 1) / wpb;
-  const size_t smem = sizeof(MfmaSmem) * wpb;
-  if (kv_fp8)
-    hipLaunchKernelGGL(attn_decode_mfma_kernel<true>, dim3((unsigned)blocks),
-                       dim3(wpb * WAVE), smem, s, (u16*)out, (const u16*)q,
-                       (const u8*)k_cache, (const u8*)v_cache, block_tables,
-                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
-  else
-    hipLaunchKernelGGL(attn_decode_mfma_kernel<false>, dim3((unsigned)blocks),
-                       dim3(wpb * WAVE), smem, s, (u16*)out, (const u16*)q,
-                       (const u8*)k_cache, (const u8*)v_cache, block_tables,
-                       seq_lens, bt_stride, n_dec, Hq, Hk, seq_offset, scale);
+#define LAUNCH_MFMA(D, KV8)                                                  \
+  if (head_dim == D && kv_fp8 == (KV8 ? 1 : 0)) {                            \
+    hipLaunchKernelGGL((attn_decode_mfma_kernel<D, KV8>),                    \
+                       dim3((unsigned)blocks), dim3(wpb * WAVE),             \
+                       sizeof(MfmaSmem<D>) * wpb, s, (u16*)out,              \
+                       (const u16*)q, (const u8*)k_cache,                    \
+                       (const u8*)v_cache, block_tables, seq_lens,           \
+                       bt_stride, n_dec, Hq, Hk, seq_offset, scale);         \
+    return;                                                                  \
+  }
+  LAUNCH_MFMA(128, false)
+  LAUNCH_MFMA(128, true)
+  LAUNCH_MFMA(64, false)
+  LAUNCH_MFMA(64, true)
+#undef LAUNCH_MFMA
 }
 
 // ---- probe: C[16,16] = A[16,32] @ B[32,16] with the exact frag loaders ----
