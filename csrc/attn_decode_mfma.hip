@@ -259,6 +259,147 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
 #undef LAUNCH_MFMA
 }
 
+// ---- debug probe: run the D=64 decode stages for one (seq, kv_head) and
+// dump the vt and p LDS images plus the output, to localize a stage that
+// breaks on hardware but not in the CPU simulation (tools/sim_mfma_decode.py)
+__global__ void hd64_stage_probe_kernel(
+    float* __restrict__ vt_dump,     // [64 * PV_PAD]
+    float* __restrict__ p_dump,      // [16 * PV_PAD]
+    u16* __restrict__ out,           // [Hq=1, 64]
+    const u16* __restrict__ q,       // [1, 64]
+    const u16* __restrict__ k_cache, // one page [32, 64]
+    const u16* __restrict__ v_cache,
+    int L, float scale) {
+  constexpr int D = 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  __shared__ MfmaSmem<D> smw;
+  MfmaSmem<D>* sm = &smw;
+  const int lo16 = lane & 15;
+  const int hi4 = lane >> 4;
+  const int G = 1;
+  {
+    const int head = lo16;
+#pragma unroll
+    for (int t = 0; t < D / 32; ++t) {
+      const int k0 = hi4 * 8 + t * 32;
+      u16x8 val = {};
+      if (head < G) {
+        const u16* qrow = q + head * D;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) val[j] = f2bf(bf2f(qrow[k0 + j]) * scale);
+      }
+      *(u16x8*)((char*)sm->qt + qt_swz<D>(head, k0 * 2)) = val;
+    }
+  }
+  float m_run = -1e30f, l_run = 0.f;
+  constexpr int DB = D / 16;
+  f32x4 acc[DB];
+#pragma unroll
+  for (int b = 0; b < DB; ++b) acc[b] = (f32x4)(0.f);
+  const int valid = L;
+  {
+    const u8* vsrc = (const u8*)v_cache;
+#pragma unroll
+    for (int it = 0; it < (BS * D) / ((int)WAVE * 8); ++it) {
+      const int flat = it * (int)WAVE + lane;
+      const int pos = flat & 31;
+      const int d0 = (flat >> 5) * 8;
+      const bf16frag vx = load_kv_frag<false>(vsrc + (pos * D + d0) * 2);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        sm->vt[(d0 + j) * PV_PAD + pos] = (u16)vx[j];
+    }
+  }
+  // dump vt
+  for (int i = lane; i < D * PV_PAD; i += WAVE)
+    vt_dump[i] = bf2f(sm->vt[i]);
+  f32x4 s01[2];
+#pragma unroll
+  for (int half = 0; half < 2; ++half) {
+    f32x4 d = (f32x4)(0.f);
+#pragma unroll
+    for (int kk = 0; kk < D / 32; ++kk) {
+      const u8* krow = (const u8*)k_cache +
+                       ((long)(half * 16 + lo16) * D + kk * 32 + hi4 * 8) * 2;
+      bf16frag ka = load_kv_frag<false>(krow);
+      bf16frag qb = *(const s16x8*)((char*)sm->qt +
+                                    qt_swz<D>(lo16, (kk * 32 + hi4 * 8) * 2));
+      d = mfma16(ka, qb, d);
+    }
+    s01[half] = d;
+  }
+  float sv[8];
+#pragma unroll
+  for (int half = 0; half < 2; ++half)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int pos = half * 16 + hi4 * 4 + r;
+      float x = s01[half][r];
+      if (pos >= valid) x = -1e30f;
+      sv[half * 4 + r] = x;
+    }
+  float tmax = sv[0];
+#pragma unroll
+  for (int i = 1; i < 8; ++i) tmax = fmaxf(tmax, sv[i]);
+  tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+  tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+  const float m_new = fmaxf(m_run, tmax);
+  const float al = __expf(m_run - m_new);
+  m_run = m_new;
+  float psum = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    sv[i] = __expf(sv[i] - m_new);
+    psum += sv[i];
+  }
+  psum += __shfl_xor(psum, 16, 64);
+  psum += __shfl_xor(psum, 32, 64);
+  l_run = l_run * al + psum;
+  if (hi4 == 0) sm->alpha[lo16] = al;
+#pragma unroll
+  for (int half = 0; half < 2; ++half)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      sm->p[lo16 * PV_PAD + half * 16 + hi4 * 4 + r] = f2bf(sv[half * 4 + r]);
+  // dump p
+  for (int i = lane; i < 16 * PV_PAD; i += WAVE)
+    p_dump[i] = bf2f(sm->p[i]);
+  float alr[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) alr[r] = sm->alpha[hi4 * 4 + r];
+#pragma unroll
+  for (int b = 0; b < DB; ++b)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) acc[b][r] *= alr[r];
+  bf16frag pa = *(const s16x8*)(sm->p + lo16 * PV_PAD + hi4 * 8);
+#pragma unroll
+  for (int b = 0; b < DB; ++b) {
+    bf16frag vb = *(const s16x8*)(sm->vt + (b * 16 + lo16) * PV_PAD + hi4 * 8);
+    acc[b] = mfma16(pa, vb, acc[b]);
+  }
+  if (hi4 == 0) sm->linv[lo16] = 1.0f / l_run;
+  float li[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) li[r] = sm->linv[hi4 * 4 + r];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int head = hi4 * 4 + r;
+    if (head >= G) continue;
+    u16* orow = out + head * D;
+#pragma unroll
+    for (int b = 0; b < DB; ++b) orow[b * 16 + lo16] = f2bf(acc[b][r] * li[r]);
+  }
+}
+
+extern "C" void sutro_hd64_stage_probe(float* vt_dump, float* p_dump, void* out,
+                                       const void* q, const void* k,
+                                       const void* v, int L, float scale,
+                                       hipStream_t s) {
+  hipLaunchKernelGGL(hd64_stage_probe_kernel, dim3(1), dim3(64), 0, s, vt_dump,
+                     p_dump, (u16*)out, (const u16*)q, (const u16*)k,
+                     (const u16*)v, L, scale);
+}
+
 // ---- probe: C[16,16] = A[16,32] @ B[32,16] with the exact frag loaders ----
 __global__ void mfma16_probe_kernel(float* __restrict__ c,
                                     const u16* __restrict__ a,   // [16,32]

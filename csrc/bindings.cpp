@@ -40,6 +40,8 @@ void sutro_mfma32_probe(float*, const void*, const void*, hipStream_t);
 void sutro_gemm_tn_launch(void*, const void*, const void*, const void*, int,
                           int, long, int, int, int, int, hipStream_t);
 void sutro_mfma16_probe(float*, const void*, const void*, hipStream_t);
+void sutro_hd64_stage_probe(float*, float*, void*, const void*, const void*,
+                            const void*, int, float, hipStream_t);
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
                     int, int, int, int, int, hipStream_t);
@@ -194,6 +196,20 @@ torch::Tensor mfma32_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+std::vector<torch::Tensor> hd64_stage_probe(torch::Tensor q, torch::Tensor k,
+                                             torch::Tensor v, long L,
+                                             double scale) {
+  CHECK_CUDA(q); CHECK_BF16(q);
+  auto fopt = q.options().dtype(at::kFloat);
+  auto vt = torch::zeros({64 * 40}, fopt);
+  auto p = torch::zeros({16 * 40}, fopt);
+  auto out = torch::zeros({1, 64}, q.options());
+  sutro_hd64_stage_probe(vt.data_ptr<float>(), p.data_ptr<float>(),
+                         out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                         v.data_ptr(), (int)L, (float)scale, cur_stream());
+  return {vt, p, out};
+}
+
 torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
   CHECK_CUDA(a); CHECK_BF16(a);
   auto c = torch::zeros({16, 16}, a.options().dtype(at::kFloat));
@@ -212,6 +228,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attention", &paged_attention, "paged prefill+decode attention");
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
+  m.def("hd64_stage_probe", &hd64_stage_probe, "D=64 decode stage dump probe");
   m.def("gemm_tn", &gemm_tn, "bf16 TN GEMM (MFMA, glds dbuf)",
         py::arg("x"), py::arg("w"), py::arg("res") = c10::nullopt,
         py::arg("bm") = 256, py::arg("bn") = 256, py::arg("swz") = 2,
