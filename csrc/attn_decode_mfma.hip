@@ -262,7 +262,7 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
 // ---- debug probe: run the D=64 decode stages for one (seq, kv_head) and
 // dump the vt and p LDS images plus the output, to localize a stage that
 // breaks on hardware but not in the CPU simulation (tools/sim_mfma_decode.py)
-__global__ void hd64_stage_probe_kernel(
+__global__ __launch_bounds__(256) void hd64_stage_probe_kernel(
     float* __restrict__ vt_dump,     // [64 * PV_PAD]
     float* __restrict__ p_dump,      // [16 * PV_PAD]
     u16* __restrict__ out,           // [Hq=1, 64]
@@ -272,8 +272,10 @@ __global__ void hd64_stage_probe_kernel(
     int L, float scale) {
   constexpr int D = 64;
   const int lane = threadIdx.x & (WAVE - 1);
-  __shared__ MfmaSmem<D> smw;
-  MfmaSmem<D>* sm = &smw;
+  const int wid = threadIdx.x / WAVE;
+  if (wid >= 1) return;  // same early-return shape as the real kernel
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  MfmaSmem<D>* sm = ((MfmaSmem<D>*)smem_raw) + wid;
   const int lo16 = lane & 15;
   const int hi4 = lane >> 4;
   const int G = 1;
@@ -395,9 +397,9 @@ extern "C" void sutro_hd64_stage_probe(float* vt_dump, float* p_dump, void* out,
                                        const void* q, const void* k,
                                        const void* v, int L, float scale,
                                        hipStream_t s) {
-  hipLaunchKernelGGL(hd64_stage_probe_kernel, dim3(1), dim3(64), 0, s, vt_dump,
-                     p_dump, (u16*)out, (const u16*)q, (const u16*)k,
-                     (const u16*)v, L, scale);
+  hipLaunchKernelGGL(hd64_stage_probe_kernel, dim3(1), dim3(128),
+                     sizeof(MfmaSmem<64>) * 2, s, vt_dump, p_dump, (u16*)out,
+                     (const u16*)q, (const u16*)k, (const u16*)v, L, scale);
 }
 
 // ---- probe: C[16,16] = A[16,32] @ B[32,16] with the exact frag loaders ----
