@@ -136,6 +136,10 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
           sm->vt[(d0 + j) * PV_PAD + pos] = (u16)vx[j];
       }
     }
+    // pin LDS order: vt stores above must not sink below the PV ds_reads
+    // (wave-internal; no-op at runtime). Suspected cause of the D=64
+    // mismatch — see tools/diag_hd64*.py and tools/sim_mfma_decode.py.
+    __builtin_amdgcn_wave_barrier();
 
     // ---- QK^T via MFMA: two 16-position halves
     f32x4 s01[2];
@@ -193,6 +197,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       for (int r = 0; r < 4; ++r)
         sm->p[lo16 * PV_PAD + half * 16 + hi4 * 4 + r] =
             f2bf(sv[half * 4 + r]);
+    __builtin_amdgcn_wave_barrier();
 
     // ---- rescale O by alpha of the row's head ((l/16)*4+r)
     float alr[4];
@@ -216,6 +221,7 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 
   // ---- epilogue: divide rows by their head's l and scatter
   if (hi4 == 0) sm->linv[lo16] = 1.0f / l_run;
+  __builtin_amdgcn_wave_barrier();
   float li[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) li[r] = sm->linv[hi4 * 4 + r];
