@@ -31,6 +31,12 @@ class EngineConfig:
     # measured +18%% steady-state tokens/s at batch 2048 vs a 4096 threshold.
     # 0 = admit eagerly. p0 (interactive) rows always bypass the hold-back.
     min_prefill_batch_tokens: Optional[int] = None
+    # EXPERIMENTAL: capture the prefill pass in a hipGraph at a fixed
+    # max_tokens_per_step shape and replay it for accumulated waves (saves the
+    # ~80-180 ms/wave eager launch overhead). Off until GPU-validated
+    # (ROADMAP.md; engine/prefill_graph.py).
+    graph_prefill: bool = False
+    graph_prefill_min_tokens: int = 16384
     gpu_memory_utilization: float = 0.90
     num_kv_blocks: Optional[int] = None  # None = derive from free memory
     default_max_new_tokens: int = 256

@@ -126,6 +126,21 @@ class LLMEngine:
                               f"running decode eagerly")
                 self.graph_runner = None
 
+        self.prefill_graph = None
+        if (cfg.graph_prefill and self.graph_runner is not None
+                and not self.spec.embedding):
+            from .prefill_graph import PrefillGraphRunner
+
+            try:
+                self.prefill_graph = PrefillGraphRunner(self)
+                self.prefill_graph.capture(pool=self.graph_runner._pool)
+            except Exception as e:
+                import warnings
+
+                warnings.warn(f"hipGraph prefill capture failed ({e}); "
+                              f"running prefill eagerly")
+                self.prefill_graph = None
+
         self._next_req_id = 0
         self._fsms: Dict[int, GuidedFSM] = {}
         self._next_fsm_id = 0
@@ -285,6 +300,21 @@ class LLMEngine:
             if (self.graph_runner is not None and self.graph_runner.can_run(sub)):
                 logits = self.graph_runner.run(sub)
                 sample_reqs = list(sub.reqs)
+            elif (self.prefill_graph is not None
+                  and self.prefill_graph.can_run(sub)):
+                hidden = self.prefill_graph.run(sub)
+                sample_reqs = self._sampling_reqs(sub)
+                if sample_reqs:
+                    rows, cum = [], 0
+                    for req, c in zip(sub.reqs, sub.num_new_tokens):
+                        cum += c
+                        if req.num_computed_tokens + c == req.num_prompt_tokens:
+                            rows.append(cum - 1)
+                    idx = torch.tensor(rows, dtype=torch.long,
+                                       device=hidden.device)
+                    logits = self.model.compute_logits(hidden[idx])
+                else:
+                    logits = None
             else:
                 fb = self._build_forward_batch(sub)
                 hidden = self.model(fb, self.kv)

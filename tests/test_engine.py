@@ -452,3 +452,84 @@ def test_no_prompt_split_while_decoding():
     sb = sch2.schedule()
     counts = {r.req_id: c for r, c in zip(sb.reqs, sb.num_new_tokens)}
     assert counts.get(11) == 99  # budget minus the decode token
+
+
+def test_prefill_graph_fill_host():
+    """PrefillGraphRunner's padded host fill must agree with the eager
+    ForwardBatch for the real tokens, and direct every padded token, sequence
+    and tile at inert scratch/dummy targets."""
+    import numpy as np
+
+    from sutro_amd.engine.batch import ScheduledBatch
+    from sutro_amd.engine.prefill_graph import PrefillGraphRunner
+    from sutro_amd.engine.request import Request
+
+    bs = 4
+    T, S, W = 64, 9, 16
+
+    class KVStub:
+        block_size = bs
+        block_tables = {}
+
+    class EngStub:
+        kv = KVStub()
+        scratch_block = 0
+
+    r = PrefillGraphRunner.__new__(PrefillGraphRunner)
+    r.engine = EngStub()
+    r.t_pad, r.s_max, r.bt_width = T, S, W
+    r.tile_max = S + T // 32
+    r.min_tokens = 0
+    for name, shape, dt in [("ids", (T,), torch.long), ("pos", (T,), torch.long),
+                            ("slots", (T,), torch.long), ("bt", (S, W), torch.int32),
+                            ("sl", (S,), torch.int32), ("qlocs", (S + 1,), torch.int32),
+                            ("tile_seq", (r.tile_max,), torch.int32),
+                            ("tile_q0", (r.tile_max,), torch.int32)]:
+        setattr(r, "h_" + name, torch.zeros(*shape, dtype=dt))
+
+    def mk(rid, plen, computed):
+        req = Request(req_id=rid, prompt_token_ids=list(range(10, 10 + plen)),
+                      sampling=SamplingParams(max_tokens=8))
+        req.num_computed_tokens = computed
+        KVStub.block_tables[rid] = list(range(rid * 10, rid * 10 + (plen + bs - 1) // bs))
+        return req
+
+    a = mk(1, 5, 0)        # fresh prompt
+    b = mk(2, 40, 8)       # continuing chunk of 32 (exactly one tile)
+    sb = ScheduledBatch(reqs=[a, b], num_new_tokens=[5, 32], num_prefills=2)
+    real_t = r._fill_host(sb)
+    assert real_t == 37
+
+    ids = r.h_ids.numpy(); pos = r.h_pos.numpy(); slots = r.h_slots.numpy()
+    qlocs = r.h_qlocs.numpy(); sl = r.h_sl.numpy(); bt = r.h_bt.numpy()
+    tseq = r.h_tile_seq.numpy(); tq0 = r.h_tile_q0.numpy()
+
+    # real region matches the eager construction
+    assert list(qlocs[:3]) == [0, 5, 37]
+    assert sl[0] == 5 and sl[1] == 40
+    assert ids[:5].tolist() == a.prompt_token_ids[:5]
+    assert ids[5:37].tolist() == b.prompt_token_ids[8:40]
+    assert pos[:5].tolist() == list(range(5))
+    assert pos[5:37].tolist() == list(range(8, 40))
+    ta, tb = KVStub.block_tables[1], KVStub.block_tables[2]
+    assert slots[0] == ta[0] * bs and slots[4] == ta[1] * bs
+    assert slots[5] == tb[2] * bs  # pos 8 -> block 2
+    assert bt[0, :len(ta)].tolist() == ta
+    assert bt[1, :len(tb)].tolist() == tb
+    # tiles: one for a (q0=0), one for b (q0=0); the rest point at dummy seq 2
+    assert (tseq[0], tq0[0]) == (0, 0)
+    assert (tseq[1], tq0[1]) == (1, 0)
+    assert (tseq[2:] == 2).all() and (tq0[2:] == 0).all()
+    # padding invariants
+    assert (slots[37:] == 0).all()       # scratch block 0
+    assert (qlocs[3:] == 37).all()       # zero-length dummies
+    assert (sl[2:] == 0).all()
+    assert (bt[2:] == 0).all()
+
+    # large chunks split into 32-row tiles
+    c = mk(3, 64, 0)
+    sb2 = ScheduledBatch(reqs=[c], num_new_tokens=[64], num_prefills=1)
+    r._fill_host(sb2)
+    tseq = r.h_tile_seq.numpy(); tq0 = r.h_tile_q0.numpy()
+    assert (tseq[0], tq0[0]) == (0, 0) and (tseq[1], tq0[1]) == (0, 32)
+    assert (tseq[2:] == 1).all()
