@@ -136,10 +136,6 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
           sm->vt[(d0 + j) * PV_PAD + pos] = (u16)vx[j];
       }
     }
-    // pin LDS order: vt stores above must not sink below the PV ds_reads
-    // (wave-internal; no-op at runtime). Suspected cause of the D=64
-    // mismatch — see tools/diag_hd64*.py and tools/sim_mfma_decode.py.
-    __builtin_amdgcn_wave_barrier();
 
     // ---- QK^T via MFMA: two 16-position halves
     f32x4 s01[2];
@@ -197,16 +193,16 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
       for (int r = 0; r < 4; ++r)
         sm->p[lo16 * PV_PAD + half * 16 + hi4 * 4 + r] =
             f2bf(sv[half * 4 + r]);
-    __builtin_amdgcn_wave_barrier();
 
     // ---- rescale O by alpha of the row's head ((l/16)*4+r)
     float alr[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) alr[r] = sm->alpha[hi4 * 4 + r];
 #pragma unroll
-    for (int b = 0; b < 8; ++b)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) acc[b][r] *= alr[r];
+    for (int b = 0; b < DB; ++b)   // NOT 8: at head_dim 64 acc has DB=4
+#pragma unroll                     // entries — the out-of-bounds writes here
+      for (int r = 0; r < 4; ++r)  // poisoned the whole accumulator chain
+        acc[b][r] *= alr[r];       // (ROADMAP.md: D=64 post-mortem)
 
     // ---- PV via MFMA: A = P[head16, pos32], B = Vt-read V[pos32, d16]
     bf16frag pa = *(const s16x8*)(sm->p + lo16 * PV_PAD + hi4 * 8);
@@ -221,7 +217,6 @@ __global__ __launch_bounds__(256) void attn_decode_mfma_kernel(
 
   // ---- epilogue: divide rows by their head's l and scatter
   if (hi4 == 0) sm->linv[lo16] = 1.0f / l_run;
-  __builtin_amdgcn_wave_barrier();
   float li[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) li[r] = sm->linv[hi4 * 4 + r];
@@ -268,7 +263,7 @@ extern "C" void sutro_attn_decode_mfma(void* out, const void* q,
 // ---- debug probe: run the D=64 decode stages for one (seq, kv_head) and
 // dump the vt and p LDS images plus the output, to localize a stage that
 // breaks on hardware but not in the CPU simulation (tools/sim_mfma_decode.py)
-__global__ __launch_bounds__(256) void hd64_stage_probe_kernel(
+__global__ void hd64_stage_probe_kernel(
     float* __restrict__ vt_dump,     // [64 * PV_PAD]
     float* __restrict__ p_dump,      // [16 * PV_PAD]
     u16* __restrict__ out,           // [Hq=1, 64]
@@ -278,10 +273,8 @@ __global__ __launch_bounds__(256) void hd64_stage_probe_kernel(
     int L, float scale) {
   constexpr int D = 64;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x / WAVE;
-  if (wid >= 1) return;  // same early-return shape as the real kernel
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  MfmaSmem<D>* sm = ((MfmaSmem<D>*)smem_raw) + wid;
+  __shared__ MfmaSmem<D> smw;
+  MfmaSmem<D>* sm = &smw;
   const int lo16 = lane & 15;
   const int hi4 = lane >> 4;
   const int G = 1;
@@ -403,9 +396,9 @@ extern "C" void sutro_hd64_stage_probe(float* vt_dump, float* p_dump, void* out,
                                        const void* q, const void* k,
                                        const void* v, int L, float scale,
                                        hipStream_t s) {
-  hipLaunchKernelGGL(hd64_stage_probe_kernel, dim3(1), dim3(128),
-                     sizeof(MfmaSmem<64>) * 2, s, vt_dump, p_dump, (u16*)out,
-                     (const u16*)q, (const u16*)k, (const u16*)v, L, scale);
+  hipLaunchKernelGGL(hd64_stage_probe_kernel, dim3(1), dim3(64), 0, s, vt_dump,
+                     p_dump, (u16*)out, (const u16*)q, (const u16*)k,
+                     (const u16*)v, L, scale);
 }
 
 // ---- probe: C[16,16] = A[16,32] @ B[32,16] with the exact frag loaders ----
