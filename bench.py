@@ -164,7 +164,18 @@ def main():
             if have_gpu:
                 torch.cuda.synchronize()
 
-    # warmup: W untimed steps (covers initial prefill wave)
+    # prime: finish the initial prefill wave before warmup so the measured
+    # window reflects steady state regardless of how small W is (at batch
+    # 2048 the first fill alone is ~8 steps of pure prefill). Generative
+    # only — embedding rows never leave prefill.
+    if not eng.spec.embedding:
+        for _ in range(256):
+            sch = eng.scheduler
+            if not sch.running or all(not r.in_prefill for r in sch.running):
+                break
+            eng.step()
+
+    # warmup: W untimed steps
     for _ in range(args.warmup):
         refill()
         eng.step()
