@@ -127,13 +127,15 @@ class LLMEngine:
                 self.graph_runner = None
 
         self.prefill_graph = None
-        if (cfg.graph_prefill and self.graph_runner is not None
-                and not self.spec.embedding):
+        if (cfg.graph_prefill and cfg.device.startswith("cuda")
+                and not cfg.enforce_eager):
             from .prefill_graph import PrefillGraphRunner
 
             try:
                 self.prefill_graph = PrefillGraphRunner(self)
-                self.prefill_graph.capture(pool=self.graph_runner._pool)
+                self.prefill_graph.capture(
+                    pool=self.graph_runner._pool
+                    if self.graph_runner is not None else None)
             except Exception as e:
                 import warnings
 
@@ -303,6 +305,11 @@ class LLMEngine:
             elif (self.prefill_graph is not None
                   and self.prefill_graph.can_run(sub)):
                 hidden = self.prefill_graph.run(sub)
+                if self.spec.embedding:
+                    self._embedding_update(sub, hidden, stats)
+                    self._advance_computed(sub)
+                    stats.prefill_tokens += sub.total_tokens
+                    continue
                 sample_reqs = self._sampling_reqs(sub)
                 if sample_reqs:
                     rows, cum = [], 0
@@ -319,7 +326,7 @@ class LLMEngine:
                 fb = self._build_forward_batch(sub)
                 hidden = self.model(fb, self.kv)
                 if self.spec.embedding:
-                    self._embedding_update(sub, fb, hidden, stats)
+                    self._embedding_update(sub, hidden, stats)
                     self._advance_computed(sub)
                     stats.prefill_tokens += sub.total_tokens
                     continue
@@ -384,8 +391,8 @@ class LLMEngine:
 
     # ---- embedding mode ----
 
-    def _embedding_update(self, sb: ScheduledBatch, fb: ForwardBatch,
-                          hidden: torch.Tensor, stats: StepStats) -> None:
+    def _embedding_update(self, sb: ScheduledBatch, hidden: torch.Tensor,
+                          stats: StepStats) -> None:
         # batched per-sequence chunk sums (no per-seq host sync)
         S = len(sb.reqs)
         counts = torch.tensor(sb.num_new_tokens, device=hidden.device)

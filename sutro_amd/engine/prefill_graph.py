@@ -16,6 +16,8 @@ Padding is scratch-directed and kernel-verified:
 
 The hidden-states output buffer is returned whole; logits for rows that
 complete their prompt are computed OUTSIDE the graph (dynamic gather).
+Embedding models use the same runner (their whole job is prefill) — the
+mean-pool consumes the returned hidden states directly.
 
 Enable with EngineConfig.graph_prefill=True. Not yet GPU-validated — wired
 behind the flag for round-2 measurement (ROADMAP.md).
