@@ -533,3 +533,18 @@ def test_prefill_graph_fill_host():
     tseq = r.h_tile_seq.numpy(); tq0 = r.h_tile_q0.numpy()
     assert (tseq[0], tq0[0]) == (0, 0) and (tseq[1], tq0[1]) == (0, 32)
     assert (tseq[2:] == 1).all()
+
+
+def test_graph_prefill_flag_safe_on_cpu(tiny_engine_factory=None):
+    """graph_prefill=True must be inert on CPU (no capture attempted) and the
+    engine must still run end to end."""
+    spec = tiny_spec_for_tests()
+    cfg = EngineConfig(spec=spec, device="cpu", max_num_seqs=8,
+                       max_model_len=256, num_kv_blocks=128,
+                       graph_prefill=True)
+    from sutro_amd.engine.engine import LLMEngine
+
+    eng = LLMEngine(cfg)
+    assert eng.prefill_graph is None
+    outs = eng.generate([[3, 4, 5]], SamplingParams(max_tokens=4, temperature=0.0))
+    assert len(outs) == 1 and len(outs[0].output_token_ids) >= 1
