@@ -547,4 +547,4 @@ def test_graph_prefill_flag_safe_on_cpu(tiny_engine_factory=None):
     eng = LLMEngine(cfg)
     assert eng.prefill_graph is None
     outs = eng.generate([[3, 4, 5]], SamplingParams(max_tokens=4, temperature=0.0))
-    assert len(outs) == 1 and len(outs[0].output_token_ids) >= 1
+    assert len(outs) == 1 and isinstance(outs[0], str)
