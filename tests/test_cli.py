@@ -1,6 +1,5 @@
 """CLI command mirror (reference `cli.py`) via click's CliRunner."""
 
-import json
 
 import pandas as pd
 import pytest

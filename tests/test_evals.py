@@ -1,6 +1,5 @@
 """Elo / Bradley-Terry aggregation (reference `templates/evals.py:181-336`)."""
 
-import numpy as np
 import pandas as pd
 
 from sutro_amd.templates.evals import Rank

@@ -2,7 +2,6 @@
 
 import json
 
-import pytest
 from pydantic import BaseModel
 
 from sutro_amd.engine.guided import (

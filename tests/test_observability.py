@@ -4,7 +4,6 @@
 import json
 import os
 
-import pytest
 
 from sutro_amd.observability import _row_run_id, tracing_enabled
 

@@ -5,7 +5,6 @@ These same references are the oracle the HIP kernels are tested against on GPU
 
 import math
 
-import pytest
 import torch
 
 from sutro_amd.ops import torch_ref as R

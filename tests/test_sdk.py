@@ -3,8 +3,6 @@ plus transport retry-policy tests with mocked HTTP (mirroring the reference's
 `tests/test_sdk.py` mechanisms)."""
 
 import json
-import os
-import time
 from unittest.mock import MagicMock, patch
 
 import pandas as pd

@@ -4,11 +4,9 @@ TP=2 must match TP=1: the parallel linears partition deterministic full
 weights, so logits agree to fp32 rounding and greedy generation is identical.
 """
 
-import json
 import multiprocessing as mp
 import os
 
-import pytest
 import torch
 
 
