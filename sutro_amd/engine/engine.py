@@ -190,6 +190,7 @@ class LLMEngine:
         )
         if fsm_id is not None:
             req.fsm_state = self._fsms[fsm_id].start_state()
+            req.fsm_start_state = req.fsm_state
         self._next_req_id += 1
         self.scheduler.add_request(req, priority)
         return req

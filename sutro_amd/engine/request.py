@@ -64,6 +64,8 @@ class Request:
     # bumped when the request's KV pages are released while it stays schedulable
     # (preemption) so cached per-row block-table state can be invalidated
     alloc_gen: int = 0
+    # FSM state to rewind to when the row restarts (set at admission)
+    fsm_start_state: int = 0
     _stop_ids: Optional[frozenset] = field(default=None, repr=False)
 
     def stop_ids(self, eos_id: int) -> frozenset:
@@ -88,6 +90,18 @@ class Request:
     @property
     def in_prefill(self) -> bool:
         return self.num_computed_tokens < self.num_prompt_tokens
+
+    def restart(self) -> None:
+        """Reset generation state for a full re-run (preemption discards
+        sampled tokens: resuming mid-decode after losing the KV would
+        re-sample from a stale position and corrupt the output sequence —
+        caught by tests/test_scheduler_invariants.py). Seeded rows regenerate
+        bit-identically."""
+        self.num_computed_tokens = 0
+        self.output_token_ids = []
+        self.cumulative_logprob = 0.0
+        self.fsm_state = self.fsm_start_state
+        self.alloc_gen += 1
 
     def token_at(self, idx: int) -> int:
         if idx < self.num_prompt_tokens:

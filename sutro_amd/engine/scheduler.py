@@ -70,8 +70,7 @@ class Scheduler:
                 continue
             self.running.remove(victim)
             self.kv.release(victim.req_id)
-            victim.num_computed_tokens = 0
-            victim.alloc_gen += 1
+            victim.restart()
             q = self.waiting_p0 if self.priorities.get(victim.req_id, 0) == 0 else self.waiting_p1
             q.appendleft(victim)
             return True
@@ -111,8 +110,7 @@ class Scheduler:
                 # could not even hold this one: preempt it too
                 self.running.remove(req)
                 self.kv.release(req.req_id)
-                req.num_computed_tokens = 0
-                req.alloc_gen += 1
+                req.restart()
                 (self.waiting_p0 if self.priorities.get(req.req_id, 0) == 0
                  else self.waiting_p1).appendleft(req)
                 continue
