@@ -180,7 +180,7 @@ class DFA:
         return st in self.accepting
 
 
-def compile_dfa(node: _Node, max_states: int = 20000) -> DFA:
+def compile_dfa(node: _Node, max_states: int = 150000) -> DFA:
     nfa = _NFA()
     start, end = _build(nfa, node)
 

@@ -1,0 +1,140 @@
+"""Property-based checks of the guided-decoding FSM: for randomly generated
+JSON schemas, ANY walk through the token masks must terminate in a string
+that json-parses AND validates against the schema."""
+
+import json
+
+from hypothesis import given, settings, strategies as st
+
+from sutro_amd.engine.guided import GuidedFSM
+from sutro_amd.engine.tokenizer import BYTE_OFFSET, EOS_ID
+
+# ---- random schema generator (the subset the engine supports) ----
+
+names = st.sampled_from(["a", "b", "tag", "value", "x1"])
+enum_vals = st.lists(st.sampled_from(["red", "green", "blue", "x", "yy"]),
+                     min_size=1, max_size=3, unique=True)
+
+leaf = st.one_of(
+    st.just({"type": "boolean"}),
+    st.just({"type": "integer"}),
+    st.builds(lambda lo, w: {"type": "integer", "minimum": lo,
+                             "maximum": lo + w},
+              st.integers(-20, 50), st.integers(0, 40)),
+    st.just({"type": "number"}),
+    st.just({"type": "string"}),
+    st.builds(lambda v: {"enum": v}, enum_vals),
+)
+
+
+def arr(inner):
+    return st.builds(
+        lambda it, lo, extra: {"type": "array", "items": it,
+                               "minItems": lo, "maxItems": lo + extra},
+        inner, st.integers(0, 2), st.integers(0, 2))
+
+
+def obj(inner):
+    return st.builds(
+        lambda props: {"type": "object",
+                       "properties": {k: v for k, v in props}},
+        st.lists(st.tuples(names, inner), min_size=1, max_size=3,
+                 unique_by=lambda t: t[0]))
+
+
+schemas = st.recursive(leaf, lambda inner: st.one_of(arr(inner), obj(inner)),
+                       max_leaves=4)
+
+
+def walk(fsm, data, max_steps=400):
+    """Random mask-guided walk; returns the produced bytes."""
+    state = fsm.start_state()
+    out = []
+    for _ in range(max_steps):
+        mask = fsm.mask_for(state)
+        allowed = mask.nonzero().flatten().tolist()
+        assert allowed, "FSM dead-ended with no allowed token"
+        tok = data.draw(st.sampled_from(allowed))
+        if tok == EOS_ID:
+            return bytes(out), True
+        out.append(tok - BYTE_OFFSET)
+        state = fsm.advance(state, tok)
+    # out of steps: walk the shortest path to an accepting state (BFS over
+    # the DFA, computed once) so unbounded repeats always terminate
+    dist = _dist_to_accept(fsm.dfa)
+    for _ in range(4000):
+        if state in fsm.dfa.accepting:
+            return bytes(out), True
+        nxt = min(fsm.dfa.transitions[state].items(),
+                  key=lambda kv: dist.get(kv[1], 1 << 30))
+        out.append(nxt[0])
+        state = nxt[1]
+    return bytes(out), False
+
+
+_dist_cache = {}
+
+
+def _dist_to_accept(dfa):
+    key = id(dfa)
+    if key in _dist_cache:
+        return _dist_cache[key]
+    from collections import deque
+    rev = {}
+    for st_, trans in enumerate(dfa.transitions):
+        for _, t in trans.items():
+            rev.setdefault(t, set()).add(st_)
+    dist = {a: 0 for a in dfa.accepting}
+    q = deque(dfa.accepting)
+    while q:
+        u = q.popleft()
+        for p_ in rev.get(u, ()):
+            if p_ not in dist:
+                dist[p_] = dist[u] + 1
+                q.append(p_)
+    _dist_cache[key] = dist
+    return dist
+
+
+def validate(schema, value):
+    t = schema.get("type")
+    if "enum" in schema:
+        assert value in schema["enum"], (value, schema)
+        return
+    if t == "boolean":
+        assert isinstance(value, bool)
+    elif t == "integer":
+        assert isinstance(value, int) and not isinstance(value, bool)
+        if "minimum" in schema:
+            assert value >= schema["minimum"]
+        if "maximum" in schema:
+            assert value <= schema["maximum"]
+    elif t == "number":
+        assert isinstance(value, (int, float)) and not isinstance(value, bool)
+    elif t == "string":
+        assert isinstance(value, str)
+    elif t == "array":
+        assert isinstance(value, list)
+        if "minItems" in schema:
+            assert len(value) >= schema["minItems"]
+        if "maxItems" in schema:
+            assert len(value) <= schema["maxItems"]
+        for v in value:
+            validate(schema["items"], v)
+    elif t == "object":
+        assert isinstance(value, dict)
+        for k, sub in schema.get("properties", {}).items():
+            assert k in value, (k, value)
+            validate(sub, value[k])
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.data())
+def test_masked_walks_produce_schema_valid_json(data):
+    schema = data.draw(schemas, label="schema")
+    fsm = GuidedFSM.from_schema(schema)
+    out, clean = walk(fsm, data)
+    assert clean, f"walk did not terminate: {out[:120]!r}"
+    assert fsm.dfa.matches(out)
+    value = json.loads(out)
+    validate(schema, value)
