@@ -102,3 +102,21 @@ def test_schema_invalid():
 def test_colored_text_states():
     # non-tty (test env): plain passthrough
     assert "hello" in to_colored_text("hello", "success")
+
+
+def test_tokenizer_roundtrip_property():
+    """Property: the byte tokenizer round-trips ANY unicode text exactly."""
+    from hypothesis import given, settings, strategies as st
+
+    from sutro_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.text(max_size=200))
+    def run(text):
+        ids = tok.encode(text)
+        assert tok.decode(ids) == text
+        assert all(0 <= t < 259 for t in ids)
+
+    run()

@@ -77,3 +77,84 @@ def test_cumulative_logprob_confidence_bounds():
     lp = -0.5
     conf = math.exp(lp)
     assert 0 < conf < 1
+
+
+def test_sampler_support_properties():
+    """Property: sampled tokens always lie inside the top-k/top-p restricted
+    support of the (possibly FSM-masked) distribution."""
+    from hypothesis import given, settings, strategies as st
+
+    from sutro_amd.engine.request import Request, SamplingParams
+    from sutro_amd.engine.sampler import Sampler
+
+    @settings(max_examples=50, deadline=None)
+    @given(st.data())
+    def run(data):
+        n = data.draw(st.integers(1, 8))
+        vl = 32
+        torch.manual_seed(data.draw(st.integers(0, 10_000)))
+        logits = torch.randn(n, vl) * data.draw(st.floats(0.5, 4.0))
+        reqs = []
+        for i in range(n):
+            reqs.append(Request(
+                req_id=i, prompt_token_ids=[3, 4],
+                sampling=SamplingParams(
+                    temperature=data.draw(st.floats(0.0, 2.0)),
+                    top_p=data.draw(st.floats(0.05, 1.0)),
+                    top_k=data.draw(st.sampled_from([0, 1, 3, 8])),
+                    seed=data.draw(st.one_of(st.none(), st.integers(0, 99))),
+                    max_tokens=4)))
+        s = Sampler("cpu", vocab_limit=vl)
+        toks, lps = s.sample(logits, reqs)
+        for i, (tok, lp) in enumerate(zip(toks, lps)):
+            sp = reqs[i].sampling
+            row = logits[i]
+            assert 0 <= tok < vl
+            assert lp <= 1e-5  # log-probability
+            if sp.temperature == 0.0:
+                assert tok == int(row.argmax())
+                continue
+            order = torch.argsort(row, descending=True).tolist()
+            k = sp.top_k if sp.top_k > 0 else vl
+            topk = set(order[:k])
+            assert tok in topk, "outside top-k"
+            # top-p: the kept prefix (smallest set with mass >= p) always
+            # includes rank 0; token must be inside the kept prefix
+            probs = torch.softmax(row / sp.temperature, -1)
+            sp_sorted = probs[order]
+            cdf = torch.cumsum(sp_sorted, 0)
+            keep = int((cdf - sp_sorted < sp.top_p).sum())
+            kept = set(order[:max(1, keep)])
+            assert tok in kept & topk, "outside top-p support"
+
+    run()
+
+
+def test_seeded_sampling_independent_of_batch_composition():
+    """Property: a seeded row draws the same token regardless of which other
+    rows share the batch."""
+    from sutro_amd.engine.request import Request, SamplingParams
+    from sutro_amd.engine.sampler import Sampler
+
+    vl = 24
+    torch.manual_seed(5)
+    base_logits = torch.randn(1, vl)
+
+    def draw(extra_rows):
+        s = Sampler("cpu", vocab_limit=vl)
+        reqs = [Request(req_id=0, prompt_token_ids=[3, 4, 5],
+                        sampling=SamplingParams(temperature=0.9, seed=1234,
+                                                max_tokens=4))]
+        logits = [base_logits]
+        for i in range(extra_rows):
+            reqs.append(Request(req_id=100 + i, prompt_token_ids=[3],
+                                sampling=SamplingParams(temperature=1.1,
+                                                        seed=None,
+                                                        max_tokens=4)))
+            logits.append(torch.randn(1, vl))
+        toks, _ = s.sample(torch.cat(logits), reqs)
+        return toks[0]
+
+    ref = draw(0)
+    for extra in (1, 3, 7):
+        assert draw(extra) == ref
