@@ -156,6 +156,8 @@ class EngineWorker:
 
     def _admit(self, job: JobRecord, service: "JobService",
                rows_idx=None) -> None:
+        if JobStatus.is_terminal(job.status):
+            return  # cancelled while queued — never resurrect to RUNNING
         eng = self.engine
         job.datetime_started = job.datetime_started or _now()
         job.status = JobStatus.RUNNING
@@ -230,6 +232,8 @@ class EngineWorker:
                 inbox, self._inbox = self._inbox, []
                 cancelled, self._cancelled_jobs = self._cancelled_jobs, set()
             for job, service, rows_idx in inbox:
+                if job.job_id in cancelled:
+                    continue  # cancelled before admission
                 try:
                     self._admit(job, service, rows_idx)
                 except Exception as e:  # admission failure -> FAILED
