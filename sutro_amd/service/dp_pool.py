@@ -295,7 +295,8 @@ class MultiProcEngineWorker:
                         (job.input_tokens + job.output_tokens) / el)
                 if job.completed_rows % 256 == 0:
                     service.persist_job(job, with_results=True)
-                if job.completed_rows == job.num_rows:
+                if (job.completed_rows == job.num_rows
+                        and not JobStatus.is_terminal(job.status)):
                     job.status = JobStatus.SUCCEEDED
                     job.datetime_completed = time.strftime(
                         "%Y-%m-%dT%H:%M:%S", time.gmtime())

@@ -219,7 +219,8 @@ class EngineWorker:
                 # resumable partial results: a detached client (or a restarted
                 # service) can still read completed shards
                 service.persist_job(job, with_results=True)
-            if job.completed_rows == job.num_rows:
+            if (job.completed_rows == job.num_rows
+                    and not JobStatus.is_terminal(job.status)):
                 job.status = JobStatus.SUCCEEDED
                 job.datetime_completed = _now()
                 job.job_cost = service.compute_cost(self.spec, job.input_tokens,
