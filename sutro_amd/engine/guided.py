@@ -282,7 +282,7 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return cat(cls('"'), _json_string_body(ml), cls('"'))
     if t == "integer":
         lo, hi = schema.get("minimum"), schema.get("maximum")
-        if lo is not None and hi is not None and 0 < hi - lo <= 4096:
+        if lo is not None and hi is not None and 0 <= hi - lo <= 4096:
             return alt(*[lit(str(v)) for v in range(int(lo), int(hi) + 1)])
         return _INT
     if t == "number":

@@ -17,6 +17,7 @@ enum_vals = st.lists(st.sampled_from(["red", "green", "blue", "x", "yy"]),
 
 leaf = st.one_of(
     st.just({"type": "boolean"}),
+    st.just({"type": "null"}),
     st.just({"type": "integer"}),
     st.builds(lambda lo, w: {"type": "integer", "minimum": lo,
                              "maximum": lo + w},
@@ -24,6 +25,11 @@ leaf = st.one_of(
     st.just({"type": "number"}),
     st.just({"type": "string"}),
     st.builds(lambda v: {"enum": v}, enum_vals),
+    # nullable union + ranked permutation list (rank-template shapes)
+    st.builds(lambda v: {"anyOf": [{"enum": v}, {"type": "null"}]}, enum_vals),
+    st.builds(lambda v: {"type": "array", "items": {"enum": v},
+                         "minItems": len(v), "maxItems": len(v),
+                         "uniqueItems": True}, enum_vals),
 )
 
 
@@ -98,8 +104,20 @@ def _dist_to_accept(dfa):
 
 def validate(schema, value):
     t = schema.get("type")
+    if "anyOf" in schema:
+        errs = []
+        for opt in schema["anyOf"]:
+            try:
+                validate(opt, value)
+                return
+            except AssertionError as e:
+                errs.append(e)
+        raise AssertionError(f"no anyOf branch matched {value!r}: {errs}")
     if "enum" in schema:
         assert value in schema["enum"], (value, schema)
+        return
+    if t == "null":
+        assert value is None
         return
     if t == "boolean":
         assert isinstance(value, bool)
@@ -119,6 +137,8 @@ def validate(schema, value):
             assert len(value) >= schema["minItems"]
         if "maxItems" in schema:
             assert len(value) <= schema["maxItems"]
+        if schema.get("uniqueItems"):
+            assert len(set(map(str, value))) == len(value), value
         for v in value:
             validate(schema["items"], v)
     elif t == "object":
