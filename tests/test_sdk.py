@@ -329,3 +329,49 @@ def test_classify_template_end_to_end(local_client):
                                 model="qwen-3.5-2b")
     assert set(res["classification"]) <= {"Positive", "Negative"}
     assert "scratchpad" not in res.columns
+
+
+def test_unpack_property_roundtrip(local_client):
+    """Property: for random dicts serialized into the outputs column, the
+    auto-unpack explodes them losslessly; non-dict / malformed rows leave the
+    frame untouched."""
+    from hypothesis import given, settings, strategies as st
+
+    import pandas as pd
+
+    vals = st.one_of(st.integers(-5, 5), st.text(max_size=6),
+                     st.booleans(), st.none())
+    dicts = st.dictionaries(
+        st.sampled_from(["a", "b", "c", "score"]), vals,
+        min_size=1, max_size=3)
+
+    @settings(max_examples=40, deadline=None)
+    @given(st.data())
+    def run(data):
+        rows = data.draw(st.lists(dicts, min_size=1, max_size=6))
+        # uniform key structure (the reference unpacks column-wise)
+        keys = sorted({k for d in rows for k in d})
+        rows = [{k: d.get(k) for k in keys} for d in rows]
+        df = pd.DataFrame({"outputs": [json.dumps(d) for d in rows]})
+        out = local_client._maybe_unpack_json(df.copy(), "outputs")
+        for k in keys:
+            assert k in out.columns
+            got = out[k].tolist()
+            want = [d[k] for d in rows]
+            for g, w in zip(got, want):
+                if w is None:
+                    assert g is None or (isinstance(g, float) and g != g)
+                else:
+                    assert g == w
+
+        # malformed rows -> untouched
+        bad = pd.DataFrame({"outputs": ['{"a": 1}', "{not json"]})
+        out2 = local_client._maybe_unpack_json(bad.copy(), "outputs")
+        assert list(out2.columns) == ["outputs"]
+
+        # non-dict JSON -> untouched
+        arr = pd.DataFrame({"outputs": ["[1,2]", "[3]"]})
+        out3 = local_client._maybe_unpack_json(arr.copy(), "outputs")
+        assert list(out3.columns) == ["outputs"]
+
+    run()
