@@ -189,3 +189,23 @@ def test_moe_expert_parallel_matches_single_rank():
     for p in procs:
         p.join(timeout=60)
     assert results[0] == results[1] == ref
+
+
+def test_bench_tp2_torchrun_cpu():
+    """The driver-contract bench entrypoint with --tp 2 runs end to end over
+    gloo on CPU (2 ranks, one TP replica) and prints the JSON line."""
+    import json as _json
+    import subprocess
+    import sys
+
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29753", "bench.py", "--gpus", "2", "--tp", "2",
+           "--steps", "2", "--warmup", "1"]
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    d = _json.loads(line)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp1xtp2"
+    assert d["value"] > 0
