@@ -158,3 +158,52 @@ def test_masked_walks_produce_schema_valid_json(data):
     assert fsm.dfa.matches(out)
     value = json.loads(out)
     validate(schema, value)
+
+
+def sample_value(schema, data):
+    """Draw a schema-VALID python value (mirror of validate())."""
+    if "anyOf" in schema:
+        return sample_value(data.draw(st.sampled_from(schema["anyOf"])), data)
+    if "enum" in schema:
+        return data.draw(st.sampled_from(schema["enum"]))
+    t = schema.get("type")
+    if t == "boolean":
+        return data.draw(st.booleans())
+    if t == "null":
+        return None
+    if t == "integer":
+        lo = schema.get("minimum", -999)
+        hi = schema.get("maximum", 999)
+        return data.draw(st.integers(lo, hi))
+    if t == "number":
+        return data.draw(st.integers(-99, 99))  # ints are valid numbers
+    if t == "string":
+        return data.draw(st.text(
+            alphabet=st.characters(min_codepoint=32, max_codepoint=126,
+                                   blacklist_characters='"\\'),
+            max_size=8))
+    if t == "array":
+        lo = schema.get("minItems", 0)
+        hi = schema.get("maxItems", lo + 2)
+        if schema.get("uniqueItems") and "enum" in schema.get("items", {}):
+            perm = data.draw(st.permutations(schema["items"]["enum"]))
+            return list(perm)[:hi]
+        return [sample_value(schema["items"], data)
+                for _ in range(data.draw(st.integers(lo, hi)))]
+    if t == "object":
+        return {k: sample_value(sub, data)
+                for k, sub in schema.get("properties", {}).items()}
+    raise AssertionError(f"unhandled schema {schema}")
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.data())
+def test_valid_values_are_accepted(data):
+    """Completeness: any schema-valid value, compactly serialized with keys
+    in schema order, must be ACCEPTED by the guided DFA (the FSM must not
+    over-constrain the model)."""
+    schema = data.draw(schemas, label="schema")
+    value = sample_value(schema, data)
+    text = json.dumps(value, separators=(",", ":"))
+    fsm = GuidedFSM.from_schema(schema)
+    assert fsm.dfa.matches(text.encode()), (schema, text)
