@@ -122,6 +122,9 @@ class EngineWorker:
         self.tokenizer = get_tokenizer()
         self._inbox: List[tuple] = []
         self._req_meta: Dict[int, tuple] = {}  # req_id -> (job, row_idx)
+        # compiled guided-decoding FSMs, shared across jobs with the same
+        # schema (DFA compilation for deep schemas can take seconds)
+        self._fsm_by_schema: Dict[str, int] = {}
         self._cancelled_jobs: set = set()
         self._lock = threading.Lock()
         self._wake = threading.Event()
@@ -170,7 +173,11 @@ class EngineWorker:
                 "content": {"type": "string", "maxLength": 512}}}
             job._auto_reasoning_schema = True  # type: ignore[attr-defined]
         if schema is not None and not self.spec.embedding:
-            fsm_id = eng.register_fsm(schema)
+            key = json.dumps(schema, sort_keys=True)
+            fsm_id = self._fsm_by_schema.get(key)
+            if fsm_id is None:
+                fsm_id = eng.register_fsm(schema)
+                self._fsm_by_schema[key] = fsm_id
         default_max = 1024 if schema is not None else eng.cfg.default_max_new_tokens
         t_start = time.time()
         indices = rows_idx if rows_idx is not None else range(len(job.inputs))
