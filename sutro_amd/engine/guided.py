@@ -184,17 +184,33 @@ def compile_dfa(node: _Node, max_states: int = 150000) -> DFA:
     nfa = _NFA()
     start, end = _build(nfa, node)
 
-    def closure(states: FrozenSet[int]) -> FrozenSet[int]:
-        stack, seen = list(states), set(states)
-        while stack:
-            s = stack.pop()
-            for t in nfa.eps[s]:
-                if t not in seen:
-                    seen.add(t)
-                    stack.append(t)
-        return frozenset(seen)
+    # per-NFA-state eps-closure, memoized: closure of a SET is the union of
+    # its members' cached closures (the subset construction calls this for
+    # every (state-set, byte) pair — recomputing the DFS there dominated
+    # compile time ~5x on deep bounded-array schemas)
+    state_closure: Dict[int, FrozenSet[int]] = {}
 
-    start_set = closure(frozenset([start]))
+    def one_closure(s0: int) -> FrozenSet[int]:
+        c = state_closure.get(s0)
+        if c is None:
+            stack, seen = [s0], {s0}
+            while stack:
+                s = stack.pop()
+                for t in nfa.eps[s]:
+                    if t not in seen:
+                        seen.add(t)
+                        stack.append(t)
+            c = frozenset(seen)
+            state_closure[s0] = c
+        return c
+
+    def closure(states) -> FrozenSet[int]:
+        out: Set[int] = set()
+        for s in states:
+            out |= one_closure(s)
+        return frozenset(out)
+
+    start_set = closure((start,))
     index: Dict[FrozenSet[int], int] = {start_set: 0}
     work = [start_set]
     transitions: List[Dict[int, int]] = [{}]
@@ -211,7 +227,7 @@ def compile_dfa(node: _Node, max_states: int = 150000) -> DFA:
                 for b in chars:
                     by_byte.setdefault(b, set()).add(t)
         for b, targets in by_byte.items():
-            tset = closure(frozenset(targets))
+            tset = closure(targets)
             if tset not in index:
                 if len(index) >= max_states:
                     raise ValueError("guided-decoding DFA too large")
