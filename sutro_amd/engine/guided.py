@@ -226,15 +226,23 @@ def compile_dfa(node: _Node, max_states: int = 150000) -> DFA:
             for chars, t in nfa.trans[s]:
                 for b in chars:
                     by_byte.setdefault(b, set()).add(t)
+        # bytes of one character class share a target set — resolve each
+        # distinct set once (a JSON string class spans ~90 bytes)
+        resolved: Dict[FrozenSet[int], int] = {}
         for b, targets in by_byte.items():
-            tset = closure(targets)
-            if tset not in index:
-                if len(index) >= max_states:
-                    raise ValueError("guided-decoding DFA too large")
-                index[tset] = len(index)
-                transitions.append({})
-                work.append(tset)
-            transitions[ci][b] = index[tset]
+            key = frozenset(targets)
+            ti = resolved.get(key)
+            if ti is None:
+                tset = closure(key)
+                if tset not in index:
+                    if len(index) >= max_states:
+                        raise ValueError("guided-decoding DFA too large")
+                    index[tset] = len(index)
+                    transitions.append({})
+                    work.append(tset)
+                ti = index[tset]
+                resolved[key] = ti
+            transitions[ci][b] = ti
     return DFA(transitions, accepting)
 
 
