@@ -92,10 +92,18 @@ def test_scheduler_random_interleaving(data):
 
     n_ops = data.draw(st.integers(5, 30), label="n_ops")
     for _ in range(n_ops):
-        op = data.draw(st.sampled_from(["submit", "step", "step", "step"]),
-                       label="op")
+        op = data.draw(st.sampled_from(["submit", "abort", "step", "step",
+                                        "step"]), label="op")
         if op == "submit" and rid[0] < 24:
             submit()
+        elif op == "abort":
+            live = [r for r in all_reqs if r.finish_reason is None]
+            if live:
+                sch.abort_request(data.draw(st.sampled_from(live),
+                                            label="victim"))
+                check_accounting(sch, kv, all_reqs)
+            else:
+                run_step()
         else:
             run_step()
 
