@@ -37,6 +37,14 @@ class EngineConfig:
     # (ROADMAP.md; engine/prefill_graph.py).
     graph_prefill: bool = False
     graph_prefill_min_tokens: int = 16384
+    # EXPERIMENTAL: one-step-lagged decode. The sampled-token tensor of step
+    # N feeds step N+1's input ids directly (device-side), so the host never
+    # waits on the GPU inside a pure-decode streak: stop/FSM bookkeeping
+    # consumes step N's tokens while step N+1 runs. Rows that finish decode
+    # one extra discarded token. FSM-guided rows force the synchronous path
+    # (their mask needs the sampled token). Off until GPU-validated
+    # (ROADMAP.md item 4 — async decode).
+    async_decode: bool = False
     gpu_memory_utilization: float = 0.90
     num_kv_blocks: Optional[int] = None  # None = derive from free memory
     default_max_new_tokens: int = 256

@@ -144,6 +144,9 @@ class LLMEngine:
                 self.prefill_graph = None
 
         self._next_req_id = 0
+        # async decode: step N's sampled-token tensors, consumed at N+1
+        # (reqs, tokens_t, lps_t, alloc_gens)
+        self._pending_decode: Optional[tuple] = None
         self._fsms: Dict[int, GuidedFSM] = {}
         self._next_fsm_id = 0
         self._emb_sums: Dict[int, torch.Tensor] = {}
@@ -220,7 +223,10 @@ class LLMEngine:
             max_blocks = max(max_blocks, len(table))
             for j in range(c):
                 pos = start + j
-                input_ids[cursor] = req.token_at(pos)
+                # async decode: the row's last token is still in flight
+                # (pending tensor overrides fb.input_ids); placeholder here
+                input_ids[cursor] = (req.token_at(pos)
+                                     if pos < req.total_len else 0)
                 positions[cursor] = pos
                 slots[cursor] = table[pos // bs] * bs + pos % bs
                 cursor += 1
@@ -276,11 +282,52 @@ class LLMEngine:
         out.extend(sb.reqs[sb.num_prefills:])
         return out
 
+    def _consume_pending(self, stats: StepStats) -> None:
+        """Apply the previous async step's tokens (host side of the one-step
+        lag). Skips rows that finished or were preempt-restarted meanwhile."""
+        p = self._pending_decode
+        if p is None:
+            return
+        self._pending_decode = None
+        reqs, toks_t, lps_t, gens = p
+        toks = toks_t.tolist()  # the only device wait on the async path —
+        lps = lps_t.tolist()    # by now the GPU finished this step long ago
+        for req, tok, lp, gen in zip(reqs, toks, lps, gens):
+            if req.finish_reason is not None or req.alloc_gen != gen:
+                continue
+            self._apply_sampled(req, int(tok), float(lp), stats)
+
+    def _async_eligible(self, sub: ScheduledBatch) -> bool:
+        if not (self.cfg.async_decode and not self.spec.embedding
+                and sub.num_prefills == 0 and sub.reqs):
+            return False
+        if any(r.fsm_id is not None for r in sub.reqs):
+            return False  # the FSM mask needs the sampled token NOW
+        p = self._pending_decode
+        if p is None:
+            return True
+        # the lagged token tensor feeds this step's ids verbatim, so the
+        # decode composition (and order) must be unchanged
+        return len(p[0]) == len(sub.reqs) and all(
+            a is b for a, b in zip(p[0], sub.reqs))
+
+    @staticmethod
+    def _drop_finished(sub: ScheduledBatch) -> ScheduledBatch:
+        kept = [(r, c) for r, c in zip(sub.reqs, sub.num_new_tokens)
+                if r.finish_reason is None]
+        return ScheduledBatch(
+            reqs=[r for r, _ in kept],
+            num_new_tokens=[c for _, c in kept],
+            num_prefills=sum(1 for r, _ in kept if r.in_prefill))
+
     @torch.no_grad()
     def step(self) -> StepStats:
         sb = self.scheduler.schedule()
         stats = StepStats()
         if not sb.reqs:
+            # drain the final lagged tokens (finished-row extras only)
+            self._consume_pending(stats)
+            self.total_output_tokens += stats.output_tokens
             return stats
         stats.scheduled_tokens = sb.total_tokens
 
@@ -298,6 +345,41 @@ class LLMEngine:
                 num_prefills=0))
         else:
             sub_batches.append(sb)
+
+        async_mode = (len(sub_batches) == 1
+                      and self._async_eligible(sub_batches[0]))
+        if not async_mode and self._pending_decode is not None:
+            # sync point: consume first; it may finish rows already scheduled
+            # into this batch — drop them before executing
+            self._consume_pending(stats)
+            sub_batches = [self._drop_finished(s) for s in sub_batches]
+            sub_batches = [s for s in sub_batches if s.reqs]
+
+        if async_mode:
+            sub = sub_batches[0]
+            pend = self._pending_decode
+            ids_t = pend[1] if pend is not None else None
+            if self.graph_runner is not None and self.graph_runner.can_run(sub):
+                logits = self.graph_runner.run(sub, ids_override=ids_t)
+            else:
+                fb = self._build_forward_batch(sub)
+                if ids_t is not None:
+                    fb.input_ids = ids_t
+                hidden = self.model(fb, self.kv)
+                logits = self.model.compute_logits(hidden[fb.logits_idx])
+            steps = np.fromiter((r.num_computed_tokens + 1 for r in sub.reqs),
+                                np.uint64, len(sub.reqs))
+            toks_t, lps_t = self.sampler.sample(
+                logits, list(sub.reqs), None, steps_override=steps,
+                return_tensors=True)
+            self._advance_computed(sub)
+            new_pending = (list(sub.reqs), toks_t, lps_t,
+                           [r.alloc_gen for r in sub.reqs])
+            self._consume_pending(stats)  # previous step's tokens (lagged)
+            self._pending_decode = new_pending
+            self.total_prompt_tokens += stats.prefill_tokens
+            self.total_output_tokens += stats.output_tokens
+            return stats
 
         for sub in sub_batches:
             if (self.graph_runner is not None and self.graph_runner.can_run(sub)):

@@ -112,8 +112,11 @@ class DecodeGraphRunner:
         return (sb.num_prefills == 0 and 0 < len(sb.reqs) <= self.max_bucket)
 
     @torch.no_grad()
-    def run(self, sb: ScheduledBatch) -> torch.Tensor:
-        """Fill static buffers, replay, return logits[:n]."""
+    def run(self, sb: ScheduledBatch,
+            ids_override: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Fill static buffers, replay, return logits[:n]. `ids_override`
+        (async decode) is the previous step's sampled-token tensor, copied
+        device-side over the host-staged ids (which lag by one token)."""
         n = len(sb.reqs)
         bucket = next(b for b in self.buckets if b >= n)
         self._fill_host(sb, bucket)
@@ -122,6 +125,8 @@ class DecodeGraphRunner:
         self.d_slots[:bucket].copy_(self.h_slots[:bucket], non_blocking=True)
         self.d_bt[:bucket].copy_(self.h_bt[:bucket], non_blocking=True)
         self.d_sl[:bucket].copy_(self.h_sl[:bucket], non_blocking=True)
+        if ids_override is not None:
+            self.d_ids[:n].copy_(ids_override, non_blocking=True)
         self.graphs[bucket].replay()
         return self.logits_out[bucket][:n]
 

@@ -18,6 +18,10 @@ import torch
 
 from .request import Request
 
+# below this, a temperature is treated as greedy (the reference cloud's
+# sampling_params pass through verbatim, so defend against denormals)
+_GREEDY_EPS = 1e-5
+
 
 def _splitmix64(x: np.ndarray) -> np.ndarray:
     x = (x + np.uint64(0x9E3779B97F4A7C15)) & np.uint64(0xFFFFFFFFFFFFFFFF)
@@ -71,8 +75,12 @@ class Sampler:
         logits: torch.Tensor,            # [n, V] (full model vocab)
         reqs: List[Request],             # the n requests, in logits-row order
         fsm_mask: Optional[torch.Tensor] = None,  # [n, vocab_limit] bool, True=allowed
+        steps_override: Optional[np.ndarray] = None,  # per-row seed step; async
+        return_tensors: bool = False,    # skip tolist (no device sync)
     ):
-        """Returns (token_ids: List[int], logprobs: List[float])."""
+        """Returns (token_ids: List[int], logprobs: List[float]), or the
+        device tensors when return_tensors (the async-decode path defers the
+        host copy by one step)."""
         n = logits.shape[0]
         assert n == len(reqs)
         vl = min(self.vocab_limit, logits.shape[1])
@@ -82,9 +90,12 @@ class Sampler:
 
         temps, top_ps, top_ks, unseeded, seeded, seeds = self._param_tensors(reqs, lg.device)
 
-        # log-softmax over the (possibly masked) support at temperature
-        # max(T, eps) — greedy rows report logprob at T=1 over raw support
-        eff_t = torch.where(temps > 0, temps, torch.ones_like(temps))
+        # log-softmax over the (possibly masked) support at temperature.
+        # Temperatures below _GREEDY_EPS sample greedily (dividing by a
+        # denormal temperature overflows the scaled logits to inf/NaN);
+        # greedy rows report logprob at T=1 over the raw support.
+        is_greedy = temps < _GREEDY_EPS
+        eff_t = torch.where(is_greedy, torch.ones_like(temps), temps)
         scaled = lg / eff_t.unsqueeze(1)
         logprobs_all = scaled - torch.logsumexp(scaled, dim=-1, keepdim=True)
 
@@ -94,8 +105,11 @@ class Sampler:
             u[unseeded] = torch.rand(len(unseeded), generator=self.generator,
                                      dtype=torch.float64)
         if seeded:
-            steps = np.fromiter((reqs[i].total_len for i in seeded), np.uint64,
-                                len(seeded))
+            if steps_override is not None:
+                steps = steps_override[seeded].astype(np.uint64)
+            else:
+                steps = np.fromiter((reqs[i].total_len for i in seeded),
+                                    np.uint64, len(seeded))
             u[seeded] = torch.from_numpy(seeded_uniform(seeds, steps).copy())
         u = u.to(lg.device, non_blocking=True)
 
@@ -115,7 +129,8 @@ class Sampler:
         sampled = sorted_idx.gather(1, choice_rank).squeeze(1)
 
         greedy_choice = lg.argmax(dim=-1)
-        is_greedy = temps <= 0
         tokens = torch.where(is_greedy, greedy_choice, sampled)
         lp = logprobs_all.gather(1, tokens.unsqueeze(1)).squeeze(1)
+        if return_tensors:
+            return tokens, lp
         return tokens.tolist(), lp.tolist()

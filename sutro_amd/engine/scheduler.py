@@ -98,15 +98,19 @@ class Scheduler:
         for req in list(self.running):
             if req.in_prefill or budget <= 0:
                 continue
+            # the decode writes KV at position num_computed; in async-decode
+            # mode total_len lags one token behind, so size by num_computed+1
+            # (identical to total_len on the synchronous path)
+            need_len = req.num_computed_tokens + 1
             # fast path: the next token's page already exists (true except once
             # per kv_block_size steps) — skip the allocator entirely
             table = tables.get(req.req_id)
-            if table is not None and req.total_len <= len(table) * bs:
+            if table is not None and need_len <= len(table) * bs:
                 decode_reqs.append(req)
                 scheduled.add(id(req))
                 budget -= 1
                 continue
-            if not self._grow_or_preempt(req, req.total_len, scheduled):
+            if not self._grow_or_preempt(req, need_len, scheduled):
                 # could not even hold this one: preempt it too
                 self.running.remove(req)
                 self.kv.release(req.req_id)
