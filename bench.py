@@ -41,6 +41,9 @@ def build_args():
     p.add_argument("--prompt-len", type=int, default=128)
     p.add_argument("--max-new", type=int, default=128)
     p.add_argument("--tokens-per-step", type=int, default=32768)
+    p.add_argument("--async-decode", action="store_true",
+                   help="one-step-lagged decode (EngineConfig.async_decode); "
+                        "A/B flag until GPU-validated")
     p.add_argument("--min-prefill", type=int, default=None,
                    help="prefill accumulation threshold (tokens); default = "
                         "engine default")
@@ -108,6 +111,7 @@ def main():
         max_model_len=max(256, args.prompt_len + args.max_new + 32),
         max_num_seqs=args.batch,
         max_tokens_per_step=args.tokens_per_step,
+        async_decode=args.async_decode,
         **({"min_prefill_batch_tokens": args.min_prefill}
            if args.min_prefill is not None else {}),
         num_kv_blocks=args.kv_blocks if device != "cpu" else 512,
