@@ -183,3 +183,29 @@ def test_dp_embedding_job(dp_service):
     results = dp_service.job_results(job_id)["results"]
     assert len(results["outputs"]) == 3
     assert all(isinstance(v, list) and len(v) > 0 for v in results["outputs"])
+
+
+def test_dp_pool_with_async_decode(sutro_home, monkeypatch):
+    """The multi-process DP pool works with async_decode engines (lagged
+    finishes stream back through the collector like any other)."""
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "2")
+    from sutro_amd.service.jobs import JobService
+
+    svc = JobService(home=sutro_home, device="cpu",
+                     engine_kwargs={"num_kv_blocks": 128,
+                                    "max_model_len": 1024,
+                                    "async_decode": True})
+    try:
+        rows = [f"async row {i}" for i in range(6)]
+        out = svc.submit_job({
+            "model": "qwen-3.5-2b", "inputs": rows,
+            "random_seed_per_input": True,
+            "sampling_params": {"max_tokens": 6, "temperature": 0.9},
+        })
+        job_id = out["results"]
+        assert _await(svc, job_id) == "SUCCEEDED"
+        results = svc.job_results(job_id)["results"]
+        assert len(results["outputs"]) == 6
+        assert all(o is not None for o in results["outputs"])
+    finally:
+        svc.shutdown()
