@@ -4,6 +4,13 @@ One engine per GPU process. `step()` runs one scheduler iteration: build the
 flat token batch, forward the model, sample (or pool, for embedding models),
 update request state, and return per-step stats. The service layer drives it
 from a worker thread/process; `bench.py` drives it directly.
+
+With `EngineConfig.async_decode`, pure-decode steps run one-step-lagged: the
+sampled-token tensor of step N feeds step N+1's input ids directly (no host
+round-trip or device sync on the critical path) and the host applies step
+N's tokens while N+1 executes. Rows that finish decode one extra discarded
+token; FSM-guided rows and mixed prefill steps take the synchronous path.
+Sync-equivalence is proven by tests (seeded rows, incl. under preemption).
 """
 
 from __future__ import annotations
