@@ -286,7 +286,6 @@ class JobService:
         self.jobs: Dict[str, JobRecord] = {}
         self.workers: Dict[str, EngineWorker] = {}
         self.quotas = [dict(q) for q in DEFAULT_QUOTAS]
-        self._usage = [dict(rows=0, tokens=0), dict(rows=0, tokens=0)]
         self._lock = threading.Lock()
         self._load_persisted_jobs()
 
@@ -624,7 +623,17 @@ class JobService:
     def _resume_or_fail(self, job: JobRecord) -> None:
         """A job that was live when the service died: finish the MISSING rows
         from the persisted shards (resume), or fail it if inputs are gone."""
-        done = [o is not None for o in job.outputs]
+        try:
+            is_embedding = get_model_spec(job.model).embedding
+        except KeyError:
+            is_embedding = False
+        # embedding rows record their vector in `embeddings` (outputs stay
+        # None by design) — judge completeness by the right column
+        if is_embedding:
+            done = [e is not None for e in (job.embeddings or [])]
+            done += [False] * (job.num_rows - len(done))
+        else:
+            done = [o is not None for o in job.outputs]
         if not job.inputs or len(job.inputs) != job.num_rows:
             job.status = JobStatus.FAILED
             job.failure_reason = {

@@ -209,3 +209,38 @@ def test_dp_pool_with_async_decode(sutro_home, monkeypatch):
         assert all(o is not None for o in results["outputs"])
     finally:
         svc.shutdown()
+
+
+def test_restart_resume_embedding_job(sutro_home, monkeypatch):
+    """Service restart mid-embedding-job resumes by the EMBEDDINGS column
+    (outputs stay None for embedding rows by design)."""
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "1")
+    from sutro_amd.service.jobs import JobService
+
+    svc = JobService(home=sutro_home, device="cpu",
+                     engine_kwargs={"num_kv_blocks": 128,
+                                    "max_model_len": 1024})
+    try:
+        out = svc.submit_job({
+            "model": "qwen-3-embedding-0.6b",
+            "inputs": [f"doc {i}" for i in range(4)],
+        })
+        job_id = out["results"]
+        assert _await(svc, job_id) == "SUCCEEDED"
+        job = svc.get_job(job_id)
+        # persist results (normally every 256 rows / at completion)
+        svc.persist_job(job, with_results=True)
+    finally:
+        svc.shutdown()
+
+    # a fresh service on the same home must see the job complete, not re-run
+    svc2 = JobService(home=sutro_home, device="cpu",
+                      engine_kwargs={"num_kv_blocks": 128,
+                                     "max_model_len": 1024})
+    try:
+        job2 = svc2.get_job(job_id)
+        assert job2.status.value == "SUCCEEDED"
+        res = svc2.job_results(job_id)["results"]
+        assert len(res["outputs"]) == 4
+    finally:
+        svc2.shutdown()
