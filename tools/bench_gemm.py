@@ -46,6 +46,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--check-only", action="store_true")
     ap.add_argument("--iters", type=int, default=20)
+    ap.add_argument("--m", type=int, default=None,
+                    help="override M for every shape (e.g. 2048 for the "
+                         "batch-2048 operating point)")
     args = ap.parse_args()
 
     torch.manual_seed(0)
@@ -57,6 +60,8 @@ def main():
         lines.append(s)
 
     for name, M, N, K in SHAPES:
+        if args.m:
+            M = args.m
         x = (torch.randn(M, K, device=dev) * 0.5).bfloat16()
         w = (torch.randn(N, K, device=dev) * 0.02).bfloat16()
         ref = x @ w.t()
