@@ -452,14 +452,20 @@ class LLMEngine:
 
     def _apply_sampled(self, req: Request, tok: int, lp: float, stats: StepStats) -> None:
         sp = req.sampling
-        if tok in req.stop_ids(EOS_ID):
+        # FSM-guided rows honor ONLY EOS (the mask restricts it to schema-
+        # accepting states): a user stop token/string landing mid-structure
+        # would truncate the JSON and break the schema-validity guarantee
+        # the templates rely on (found by deep property fuzzing)
+        stop_hit = (tok == EOS_ID if req.fsm_id is not None
+                    else tok in req.stop_ids(EOS_ID))
+        if stop_hit:
             self.scheduler.finish(req, FinishReason.STOP)
             stats.finished.append(req)
             return
         req.output_token_ids.append(tok)
         req.cumulative_logprob += lp
         stats.output_tokens += 1
-        if sp.stop:
+        if sp.stop and req.fsm_id is None:
             # byte tokenizer: token == byte, so stop strings are byte suffixes
             out = self.tokenizer.decode(req.output_token_ids[-64:])
             for s in sp.stop:
