@@ -27,3 +27,25 @@ def test_model_generates(name):
                                                     temperature=0.5))
         assert len(outs) == 1
         assert eng.total_output_tokens >= 1
+
+
+def test_registry_specs_are_coherent():
+    """Structural sanity across ALL 40 servable specs: GQA divisibility, TP
+    divisibility for the recommended degree, MoE fields paired."""
+    from sutro_amd.models.registry import MODEL_REGISTRY
+
+    for name, s in MODEL_REGISTRY.items():
+        assert s.num_heads % s.num_kv_heads == 0, name
+        assert s.head_dim in (64, 128), name
+        assert s.hidden_size % s.num_heads != -1  # defined
+        tp = s.recommended_tp
+        assert tp >= 1 and s.num_heads % tp == 0, (name, tp)
+        assert s.num_kv_heads % tp == 0 or s.num_kv_heads < tp, (name, tp)
+        if s.num_experts:
+            assert s.experts_per_token >= 1, name
+            assert s.moe_intermediate_size > 0, name
+        else:
+            assert s.intermediate_size > 0, name
+        assert s.max_context >= 2048, name
+        assert s.param_count() > 0 and s.active_param_count() > 0, name
+        assert s.active_param_count() <= s.param_count(), name
