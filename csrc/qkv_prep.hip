@@ -8,6 +8,7 @@
 //   v heads  -> copied         -> v_cache slot
 // One wave per (token, head) item over Hq+2*Hk heads.
 #include "common.h"
+#include <cstdlib>
 
 __global__ void qkv_prep_kernel(
     const u16* __restrict__ qkv,     // [T, row_stride]
@@ -107,6 +108,105 @@ __global__ void qkv_prep_kernel(
   }
 }
 
+// ---- v2 (EXPERIMENTAL, SUTRO_QKV_PREP_V2=1): fully vectorized layout ----
+//
+// v1 maps one WAVE per (token, head) with lane==dim, which forces scalar
+// 2-byte loads/stores on the q/k path (the RoPE (d, d+half) pairing) — the
+// rocprof capture shows ~1.7 TB/s at prefill shapes (capture 8). v2 maps a
+// 16-lane GROUP per row: lane g holds the row's 8-element chunk g as one
+// b128 load, the RoPE partner chunk (g ^ half_chunks) arrives by shfl_xor,
+// and stores are b128 again. All 64 lanes stay busy for both D=128 (4
+// rows/wave) and D=64 (8 rows/wave). Index flow simulated in
+// tools/sim_qkv_prep_v2.py; GPU numerics A/B is a round-2 item (ROADMAP).
+__global__ void qkv_prep_v2_kernel(
+    const u16* __restrict__ qkv, u16* __restrict__ q_out,
+    u16* __restrict__ k_cache, u16* __restrict__ v_cache,
+    const long* __restrict__ pos, const long* __restrict__ slots,
+    const float* __restrict__ cos_sin, const u16* __restrict__ qw,
+    const u16* __restrict__ kw, float eps, int T, int Hq, int Hk, int D,
+    int bs, int row_stride, int kv_fp8) {
+  const int chunks = D / 8;             // 16-lane group for D=128, 8 for D=64
+  const int g = threadIdx.x % chunks;   // chunk within the row
+  const long item = ((long)blockIdx.x * blockDim.x + threadIdx.x) / chunks;
+  const int H = Hq + 2 * Hk;
+  if (item >= (long)T * H) return;
+  const int t = (int)(item / H);
+  const int h = (int)(item - (long)t * H);
+  const int half = D / 2;
+  const int hc = chunks / 2;            // chunk-index distance of the pair
+
+  const long slot = slots[t];
+  const long blk = slot / bs, off = slot - blk * bs;
+
+  if (h >= Hq + Hk) {                   // V head: vectorized copy / quantize
+    const int vh = h - Hq - Hk;
+    const u16* src = qkv + (long)t * row_stride + (Hq + Hk) * D + vh * D;
+    const u16x8 v = *(const u16x8*)(src + g * 8);
+    if (kv_fp8) {
+      u8* dst8 = (u8*)v_cache + (((long)blk * Hk + vh) * bs + off) * D;
+      u16x4 packed;
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        packed[j] = f2fp8x2(bf2f(v[2 * j]), bf2f(v[2 * j + 1]));
+      *(u16x4*)(dst8 + g * 8) = packed;
+    } else {
+      u16* dst = v_cache + (((long)blk * Hk + vh) * bs + off) * D;
+      *(u16x8*)(dst + g * 8) = v;
+    }
+    return;
+  }
+
+  const bool is_q = h < Hq;
+  const u16* src = is_q ? qkv + (long)t * row_stride + h * D
+                        : qkv + (long)t * row_stride + Hq * D + (h - Hq) * D;
+  const long k_slot = (((long)blk * Hk + (h - Hq)) * bs + off) * D;
+  const u16* w = is_q ? qw : kw;
+
+  const u16x8 raw = *(const u16x8*)(src + g * 8);
+  float x[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) x[j] = bf2f(raw[j]);
+
+  if (w != nullptr) {                   // per-head RMSNorm over the group
+    float ssq = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ssq += x[j] * x[j];
+    for (int m = 1; m < chunks; m <<= 1) ssq += __shfl_xor(ssq, m, 64);
+    const float inv = rsqrtf(ssq / (float)D + eps);
+    const u16x8 wv = *(const u16x8*)(w + g * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) x[j] *= inv * bf2f(wv[j]);
+  }
+
+  // RoPE: element e = 8g + j pairs with e ^ half; the partner's (normed)
+  // value sits at the same j in lane g ^ hc
+  const float* cs = cos_sin + pos[t] * D;
+  const bool lo = g < hc;               // first half of the row
+  float r[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    const int e = g * 8 + j;
+    const float p = __shfl_xor(x[j], hc, 64);
+    const float c = cs[lo ? e : e - half];
+    const float sn = cs[lo ? e + half : e];
+    r[j] = lo ? x[j] * c - p * sn : x[j] * c + p * sn;
+  }
+
+  if (!is_q && kv_fp8) {
+    u8* dst8 = (u8*)k_cache + k_slot;
+    u16x4 packed;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) packed[j] = f2fp8x2(r[2 * j], r[2 * j + 1]);
+    *(u16x4*)(dst8 + g * 8) = packed;
+  } else {
+    u16* dst = is_q ? q_out + ((long)t * Hq + h) * D : k_cache + k_slot;
+    u16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = f2bf(r[j]);
+    *(u16x8*)(dst + g * 8) = out;
+  }
+}
+
 extern "C" void sutro_qkv_prep(const void* qkv, void* q_out, void* k_cache,
                                void* v_cache, const long* pos,
                                const long* slots, const float* cos_sin,
@@ -115,6 +215,18 @@ extern "C" void sutro_qkv_prep(const void* qkv, void* q_out, void* k_cache,
                                int kv_fp8, hipStream_t s) {
   const long items = (long)T * (Hq + 2 * Hk);
   if (items == 0) return;
+  static const char* v2 = getenv("SUTRO_QKV_PREP_V2");
+  if (v2 && v2[0] == '1' && (D == 128 || D == 64)) {
+    const int chunks = D / 8;
+    const int rows_per_block = 256 / chunks;
+    const long blocks = (items + rows_per_block - 1) / rows_per_block;
+    hipLaunchKernelGGL(qkv_prep_v2_kernel, dim3((unsigned)blocks), dim3(256),
+                       0, s, (const u16*)qkv, (u16*)q_out, (u16*)k_cache,
+                       (u16*)v_cache, pos, slots, cos_sin, (const u16*)qw,
+                       (const u16*)kw, eps, T, Hq, Hk, D, bs, row_stride,
+                       kv_fp8);
+    return;
+  }
   const int wpb = 4;
   const long blocks = (items + wpb - 1) / wpb;
   hipLaunchKernelGGL(qkv_prep_kernel, dim3((unsigned)blocks), dim3(wpb * WAVE),

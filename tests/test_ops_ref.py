@@ -151,3 +151,21 @@ def test_mfma_decode_simulation_spec():
     out = buf.getvalue()
     assert "MISMATCH" not in out, out
     assert out.count("OK") == 6, out
+
+
+def test_qkv_prep_v2_simulation_spec():
+    """v2 (vectorized-chunk) qkv_prep index/shuffle flow vs reference
+    semantics — CPU guard for the env-gated kernel (round-2 GPU A/B)."""
+    import importlib.util
+    import io
+    import pathlib
+    from contextlib import redirect_stdout
+
+    spec = importlib.util.spec_from_file_location(
+        "sim_qkv_prep_v2",
+        pathlib.Path(__file__).parent.parent / "tools" / "sim_qkv_prep_v2.py")
+    mod = importlib.util.module_from_spec(spec)
+    buf = io.StringIO()
+    with redirect_stdout(buf):
+        spec.loader.exec_module(mod)  # asserts internally
+    assert buf.getvalue().count("exact") == 2
