@@ -63,6 +63,11 @@ class DecodeGraphRunner:
         self._row_nb = np.zeros(B, dtype=np.int32)
         self._row_gen = np.zeros(B, dtype=np.int64)
         self._arange = np.arange(B, dtype=np.int64)
+        # async decode overwrites the pinned staging before any stream sync;
+        # this event marks the previous step's H2D completion so the refill
+        # never races an in-flight copy (waits ~0 in practice — the copies
+        # finish at the head of the previous step's GPU timeline)
+        self._h2d_done = torch.cuda.Event() if torch.cuda.is_available() else None
         self.graphs: Dict[int, torch.cuda.CUDAGraph] = {}
         self.logits_out: Dict[int, torch.Tensor] = {}
         self._pool = None
@@ -119,6 +124,8 @@ class DecodeGraphRunner:
         device-side over the host-staged ids (which lag by one token)."""
         n = len(sb.reqs)
         bucket = next(b for b in self.buckets if b >= n)
+        if self._h2d_done is not None:
+            self._h2d_done.synchronize()
         self._fill_host(sb, bucket)
         self.d_ids[:bucket].copy_(self.h_ids[:bucket], non_blocking=True)
         self.d_pos[:bucket].copy_(self.h_pos[:bucket], non_blocking=True)
@@ -127,6 +134,8 @@ class DecodeGraphRunner:
         self.d_sl[:bucket].copy_(self.h_sl[:bucket], non_blocking=True)
         if ids_override is not None:
             self.d_ids[:n].copy_(ids_override, non_blocking=True)
+        if self._h2d_done is not None:
+            self._h2d_done.record()
         self.graphs[bucket].replay()
         return self.logits_out[bucket][:n]
 

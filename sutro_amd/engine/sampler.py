@@ -48,6 +48,14 @@ class Sampler:
         # three H2D param uploads + index lists amortize to ~nothing
         self._param_key: Optional[tuple] = None
         self._params: Optional[tuple] = None
+        # pinned ping-pong staging for the per-step uniforms: a pageable
+        # .to(device) is a stream-ordered BLOCKING copy (it would serialize
+        # the async-decode path behind the whole replay). Two buffers: the
+        # async engine's consume() of step N-1 (inside step N) synchronizes
+        # that step's stream work, so buffer (N+1)%2 == (N-1)%2 is free by
+        # the time step N+1 reuses it.
+        self._u_pinned = [None, None]
+        self._u_flip = 0
 
     def _param_tensors(self, reqs: List[Request], device) -> tuple:
         # req_ids are unique for the engine's lifetime and a request's sampling
@@ -111,7 +119,18 @@ class Sampler:
                 steps = np.fromiter((reqs[i].total_len for i in seeded),
                                     np.uint64, len(seeded))
             u[seeded] = torch.from_numpy(seeded_uniform(seeds, steps).copy())
-        u = u.to(lg.device, non_blocking=True)
+        if lg.device.type == "cuda":
+            i = self._u_flip
+            self._u_flip ^= 1
+            buf = self._u_pinned[i]
+            if buf is None or buf.numel() < n:
+                buf = torch.empty(max(n, 256), dtype=torch.float64,
+                                  pin_memory=True)
+                self._u_pinned[i] = buf
+            buf[:n].copy_(u)
+            u = buf[:n].to(lg.device, non_blocking=True)
+        else:
+            u = u.to(lg.device)
 
         sorted_logits, sorted_idx = torch.sort(scaled, dim=-1, descending=True)
         probs = torch.softmax(sorted_logits, dim=-1)
