@@ -442,8 +442,13 @@ class LLMEngine:
             req.num_computed_tokens += c
 
     def _fsm_masks(self, reqs: List[Request], device) -> Optional[torch.Tensor]:
-        if all(r.fsm_id is None for r in reqs):
+        fsm_ids = {r.fsm_id for r in reqs}
+        if fsm_ids == {None}:
             return None
+        if None not in fsm_ids and len(fsm_ids) == 1:
+            # homogeneous guided batch (the structured-job case): one gather
+            fsm = self._fsms[next(iter(fsm_ids))]
+            return fsm.mask_rows([r.fsm_state for r in reqs])
         mask = torch.ones((len(reqs), TOKENIZER_VOCAB), dtype=torch.bool, device=device)
         for i, r in enumerate(reqs):
             if r.fsm_id is not None:

@@ -374,6 +374,9 @@ class GuidedFSM:
         self.dfa = dfa
         self.device = device
         self._masks: Dict[int, torch.Tensor] = {}
+        # dense device table of visited-state masks for batched gathers
+        self._table: Optional[torch.Tensor] = None
+        self._state_row: Dict[int, int] = {}
 
     @classmethod
     def from_schema(cls, schema: dict, device: str = "cpu") -> "GuidedFSM":
@@ -395,6 +398,33 @@ class GuidedFSM:
             m = mask.to(self.device)
             self._masks[state] = m
         return m
+
+    def mask_rows(self, states) -> torch.Tensor:
+        """Batched masks [n, TOKENIZER_VOCAB] via ONE device gather from a
+        growing visited-state table (the per-row `mask[i] = mask_for(s)`
+        pattern costs one device copy kernel per guided row per step)."""
+        rows = []
+        row_of = self._state_row
+        for st in states:
+            r = row_of.get(st)
+            if r is None:
+                m = self.mask_for(st)
+                r = len(row_of)
+                row_of[st] = r
+                if self._table is None:
+                    self._table = torch.zeros(64, TOKENIZER_VOCAB,
+                                              dtype=torch.bool,
+                                              device=m.device)
+                elif r >= self._table.shape[0]:
+                    bigger = torch.zeros(self._table.shape[0] * 2,
+                                         TOKENIZER_VOCAB, dtype=torch.bool,
+                                         device=self._table.device)
+                    bigger[:self._table.shape[0]] = self._table
+                    self._table = bigger
+                self._table[r] = m
+            rows.append(r)
+        idx = torch.tensor(rows, dtype=torch.long, device=self._table.device)
+        return self._table[idx]
 
     def advance(self, state: int, token_id: int) -> int:
         """Next state after a sampled token; EOS leaves the state unchanged."""
