@@ -169,3 +169,22 @@ def test_qkv_prep_v2_simulation_spec():
     with redirect_stdout(buf):
         spec.loader.exec_module(mod)  # asserts internally
     assert buf.getvalue().count("exact") == 2
+
+
+def test_attn_prefill_simulation_spec():
+    """Prefill flash kernel index-flow simulation (k_swz staging, MFMA frags,
+    causal masking, online softmax, P round trip) must match plain attention
+    exactly for D=128 and D=64 — CPU regression guard for kernel edits."""
+    import importlib.util
+    import io
+    import pathlib
+    from contextlib import redirect_stdout
+
+    spec = importlib.util.spec_from_file_location(
+        "sim_attn_prefill",
+        pathlib.Path(__file__).parent.parent / "tools" / "sim_attn_prefill.py")
+    mod = importlib.util.module_from_spec(spec)
+    buf = io.StringIO()
+    with redirect_stdout(buf):
+        spec.loader.exec_module(mod)
+    assert buf.getvalue().count("OK") == 2
