@@ -188,3 +188,21 @@ def test_attn_prefill_simulation_spec():
     with redirect_stdout(buf):
         spec.loader.exec_module(mod)
     assert buf.getvalue().count("OK") == 2
+
+
+def test_gemm_tn_simulation_spec():
+    """gemm_tn staging/swizzle/fragment/epilogue index flow vs plain matmul
+    for all three LDS swizzle modes — CPU regression guard for tile edits."""
+    import importlib.util
+    import io
+    import pathlib
+    from contextlib import redirect_stdout
+
+    spec = importlib.util.spec_from_file_location(
+        "sim_gemm_tn",
+        pathlib.Path(__file__).parent.parent / "tools" / "sim_gemm_tn.py")
+    mod = importlib.util.module_from_spec(spec)
+    buf = io.StringIO()
+    with redirect_stdout(buf):
+        spec.loader.exec_module(mod)
+    assert buf.getvalue().count("OK") == 3
