@@ -52,3 +52,23 @@ def test_concurrent_mixed_jobs(local_client):
     col = "inference_result" if "inference_result" in r_emb else "outputs"
     emb0 = r_emb[col].iloc[0]
     assert hasattr(emb0, "__len__") and len(emb0) > 4
+
+
+def test_single_proc_service_with_async_decode(sutro_home):
+    """In-process EngineWorker with async_decode: lagged finishes flow
+    through the worker loop and results stay input-ordered."""
+    from sutro_amd.sdk import Sutro
+
+    so = Sutro(home=sutro_home, device="cpu",
+               engine_kwargs={"num_kv_blocks": 256, "max_model_len": 2048,
+                              "async_decode": True})
+    try:
+        rows = [f"async svc row {i}" for i in range(8)]
+        job_id = so.infer(rows, model="qwen-3.5-2b", stay_attached=False,
+                          sampling_params={"max_tokens": 6,
+                                           "temperature": 0.8})
+        df = so.await_job_completion(job_id, timeout=120)
+        assert len(df) == 8
+        assert df["inference_result"].notna().all()
+    finally:
+        so.shutdown()
