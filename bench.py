@@ -44,6 +44,8 @@ def build_args():
     p.add_argument("--async-decode", action="store_true",
                    help="one-step-lagged decode (EngineConfig.async_decode); "
                         "A/B flag until GPU-validated")
+    p.add_argument("--graph-prefill", action="store_true",
+                   help="hipGraph-captured prefill (EngineConfig.graph_prefill)")
     p.add_argument("--min-prefill", type=int, default=None,
                    help="prefill accumulation threshold (tokens); default = "
                         "engine default")
@@ -112,6 +114,7 @@ def main():
         max_num_seqs=args.batch,
         max_tokens_per_step=args.tokens_per_step,
         async_decode=args.async_decode,
+        graph_prefill=args.graph_prefill,
         **({"min_prefill_batch_tokens": args.min_prefill}
            if args.min_prefill is not None else {}),
         num_kv_blocks=args.kv_blocks if device != "cpu" else 512,

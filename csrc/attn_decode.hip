@@ -375,11 +375,11 @@ extern "C" void sutro_attn_decode(void* out, const void* q, const void* k_cache,
   // MFMA variant is the measured default at head_dim 128, bf16 AND e4m3 KV
   // (bf16: 10654 vs 10134 tok/s at batch 1024 / ctx 128; attention slope
   // halves at ctx 512; fp8 dequants with the native pk converts at the K/V
-  // fragment loads). head_dim 64 stays on the VALU kernel: the D=64 MFMA
-  // instantiation mismatches the fp32 reference (V-stage iterations >= 1
-  // corrupt — see tools/diag_hd64*.py); fix tracked in ROADMAP.md.
+  // fragment loads). head_dim 64 routes to MFMA too since the DB-bounded
+  // alpha-rescale fix (attn_decode_mfma.hip:202) — verified against the fp32
+  // reference by test_attn_head_dim_64 on hardware.
   // SUTRO_DECODE_VALU=1 falls back to the VALU kernel for A/B.
-  if (head_dim == 128 && Hq / Hk <= 8 &&
+  if ((head_dim == 128 || head_dim == 64) && Hq / Hk <= 8 &&
       getenv("SUTRO_DECODE_VALU") == nullptr) {
     sutro_attn_decode_mfma(out, q, k_cache, v_cache, block_tables, seq_lens,
                            bt_stride, n_dec, Hq, Hk, seq_offset, scale, kv_fp8,
