@@ -30,7 +30,7 @@ from .kv_cache import PagedKVCache
 from .request import FinishReason, Request, SamplingParams
 from .sampler import Sampler
 from .scheduler import Scheduler
-from .tokenizer import EOS_ID, TOKENIZER_VOCAB, get_tokenizer
+from .tokenizer import EOS_ID, get_tokenizer
 
 
 @dataclass
@@ -71,7 +71,10 @@ class LLMEngine:
         self.cfg = cfg
         self.spec: ModelSpec = cfg.spec
         self.device = cfg.device
-        self.tokenizer = get_tokenizer()
+        # tokenizer truncated to the model vocab (BPE merge-prefix property);
+        # models with vocab > the shipped tokenizer get the full tokenizer and
+        # the sampler masks the dead tail
+        self.tokenizer = get_tokenizer(self.spec.vocab_size)
         _setup_tunableop(cfg)
         from ..parallel.tp import TPContext
 
@@ -117,7 +120,8 @@ class LLMEngine:
         # block 0 is reserved scratch: hipGraph padding rows read/write it
         self.scratch_block = self.kv.allocator.allocate(1)[0]
         self.scheduler = Scheduler(cfg, self.kv)
-        self.sampler = Sampler(cfg.device, seed=cfg.seed, vocab_limit=TOKENIZER_VOCAB)
+        self.sampler = Sampler(cfg.device, seed=cfg.seed,
+                               vocab_limit=self.tokenizer.vocab_size)
 
         self.graph_runner = None
         if (cfg.device.startswith("cuda") and not cfg.enforce_eager
@@ -151,6 +155,7 @@ class LLMEngine:
                 self.prefill_graph = None
 
         self._next_req_id = 0
+        self._full_mask_row: Optional[torch.Tensor] = None
         # async decode: step N's sampled-token tensors, consumed at N+1
         # (reqs, tokens_t, lps_t, alloc_gens)
         self._pending_decode: Optional[tuple] = None
@@ -166,7 +171,8 @@ class LLMEngine:
     # ---- admission ----
 
     def register_fsm(self, schema: dict) -> int:
-        fsm = GuidedFSM.from_schema(schema, device=self.device)
+        fsm = GuidedFSM.from_schema(schema, tokenizer=self.tokenizer,
+                                    device=self.device)
         fsm_id = self._next_fsm_id
         self._next_fsm_id += 1
         self._fsms[fsm_id] = fsm
@@ -442,6 +448,10 @@ class LLMEngine:
             req.num_computed_tokens += c
 
     def _fsm_masks(self, reqs: List[Request], device) -> Optional[torch.Tensor]:
+        """Packed FSM mask rows [n, W] int32 (guided.py layout), or None when
+        no row is guided. Mixed batches (concurrent jobs with and without
+        schemas) group rows per FSM: one gather per FSM + one full-alive row
+        for the unguided rows."""
         fsm_ids = {r.fsm_id for r in reqs}
         if fsm_ids == {None}:
             return None
@@ -449,10 +459,23 @@ class LLMEngine:
             # homogeneous guided batch (the structured-job case): one gather
             fsm = self._fsms[next(iter(fsm_ids))]
             return fsm.mask_rows([r.fsm_state for r in reqs])
-        mask = torch.ones((len(reqs), TOKENIZER_VOCAB), dtype=torch.bool, device=device)
+        from .guided import full_mask_row
+
+        vocab = self.tokenizer.vocab_size
+        if self._full_mask_row is None or self._full_mask_row.device != torch.device(device):
+            self._full_mask_row = full_mask_row(vocab, device)
+        W = self._full_mask_row.shape[0]
+        mask = torch.empty((len(reqs), W), dtype=torch.int32, device=device)
+        by_fsm: Dict[Optional[int], List[int]] = {}
         for i, r in enumerate(reqs):
-            if r.fsm_id is not None:
-                mask[i] = self._fsms[r.fsm_id].mask_for(r.fsm_state).to(device)
+            by_fsm.setdefault(r.fsm_id, []).append(i)
+        for fid, rows in by_fsm.items():
+            idx = torch.tensor(rows, dtype=torch.long, device=device)
+            if fid is None:
+                mask[idx] = self._full_mask_row
+            else:
+                mask[idx] = self._fsms[fid].mask_rows(
+                    [reqs[i].fsm_state for i in rows]).to(device)
         return mask
 
     def _apply_sampled(self, req: Request, tok: int, lp: float, stats: StepStats) -> None:
@@ -471,11 +494,18 @@ class LLMEngine:
         req.cumulative_logprob += lp
         stats.output_tokens += 1
         if sp.stop and req.fsm_id is None:
-            # byte tokenizer: token == byte, so stop strings are byte suffixes
-            out = self.tokenizer.decode(req.output_token_ids[-64:])
+            # BPE tokens are multi-byte: a stop string can complete (or even
+            # be strictly inside) the newest token, so scan a decoded tail
+            # window and trim at TEXT level (req.text_override)
+            win = max(64, max(len(s) for s in sp.stop) + 8)
+            out = self.tokenizer.decode(req.output_token_ids[-win:])
             for s in sp.stop:
-                if s and out.endswith(s):
-                    del req.output_token_ids[len(req.output_token_ids) - len(s.encode()):]
+                if s and s in out:
+                    full = self.tokenizer.decode(req.output_token_ids)
+                    # first occurrence == the one just completed (earlier
+                    # ones would have finished the row on a previous step)
+                    pos = full.find(s)
+                    req.text_override = full[:pos] if pos >= 0 else full
                     self.scheduler.finish(req, FinishReason.STOP)
                     stats.finished.append(req)
                     return
@@ -529,6 +559,13 @@ class LLMEngine:
                 self.scheduler.finish(req, FinishReason.STOP)
                 stats.finished.append(req)
 
+    def output_text(self, req: Request) -> str:
+        """Final output text of a finished request (stop-string trims are
+        text-level because BPE trim points need not align to tokens)."""
+        if req.text_override is not None:
+            return req.text_override
+        return self.tokenizer.decode(req.output_token_ids)
+
     # ---- convenience: run a list of prompts to completion (tests/smoke) ----
 
     def generate(
@@ -546,4 +583,4 @@ class LLMEngine:
             reqs.append(self.add_request(ids, sp, fsm_id=fsm_id, arrival_idx=i))
         while self.has_work():
             self.step()
-        return [self.tokenizer.decode(r.output_token_ids) for r in reqs]
+        return [self.output_text(r) for r in reqs]

@@ -1,14 +1,18 @@
-"""Guided decoding: JSON schema -> regex -> byte-level DFA -> token masks.
+"""Guided decoding: JSON schema -> regex -> byte DFA -> per-state TOKEN masks.
 
 The server-side "Schema validation" behavior the reference client relies on
 (`/root/reference/sutro/sdk.py:220`, `common.py:170-181`) is implemented here
 natively: a pydantic/JSON schema is compiled to a regex over bytes, the regex
 to a DFA (Thompson NFA + subset construction), and each DFA state to an
-allowed-token bitmask the sampler applies. With the byte tokenizer one token
-is one byte, so masked sampling produces schema-valid JSON for ANY weights.
+allowed-token bitmask over the FULL BPE vocab: token t is allowed in state s
+iff walking t's bytes from s stays alive. Masks are bit-packed u32 words
+([num_states, ceil(V/32)]) so the fused HIP mask+sample kernel reads them
+directly; the vectorized builder walks every token over every state with
+batched gathers (no per-token Python loop).
 
 EOS is allowed exactly in accepting states; all other specials are never
-allowed.
+allowed. Single-byte tokens (ids 3..258) are always in the vocab, so any DFA
+byte-path is walkable by tokens — no token-level dead ends.
 """
 
 from __future__ import annotations
@@ -16,9 +20,10 @@ from __future__ import annotations
 from dataclasses import dataclass
 from typing import Dict, FrozenSet, List, Optional, Set, Tuple
 
+import numpy as np
 import torch
 
-from .tokenizer import BYTE_OFFSET, EOS_ID, TOKENIZER_VOCAB
+from .tokenizer import EOS_ID
 
 # ---------------------------------------------------------------------------
 # Regex AST + parser (small, self-contained; bytes alphabet)
@@ -367,74 +372,210 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
 # ---------------------------------------------------------------------------
 
 
-class GuidedFSM:
-    """DFA plus lazily-built per-state token masks over the tokenizer vocab."""
+# per-tokenizer cache of token byte-walk groups (tokenizer-level, shared by
+# every FSM): list of (ids int64 [T_L], bytes int64 [T_L, L]) per length L
+_TOKEN_GROUPS: Dict[int, list] = {}
 
-    def __init__(self, dfa: DFA, device: str = "cpu"):
+
+def _token_groups(tokenizer) -> list:
+    key = id(tokenizer)
+    g = _TOKEN_GROUPS.get(key)
+    if g is None:
+        by_len: Dict[int, list] = {}
+        for tid, tb in enumerate(tokenizer.token_bytes_table):
+            if tb:
+                by_len.setdefault(len(tb), []).append((tid, tb))
+        g = []
+        for L in sorted(by_len):
+            pairs = by_len[L]
+            ids = np.fromiter((p[0] for p in pairs), np.int64, len(pairs))
+            bts = np.frombuffer(b"".join(p[1] for p in pairs),
+                                dtype=np.uint8).reshape(len(pairs), L)
+            g.append((ids, bts.astype(np.int64)))
+        _TOKEN_GROUPS[key] = g
+    return g
+
+
+# states above this build their mask rows lazily (visited-state batches)
+# instead of one eager table — a 1024-char free-string property alone is
+# ~1k DFA states, and eager cost is states x total-token-bytes gathers
+_EAGER_STATE_LIMIT = 4096
+
+
+class GuidedFSM:
+    """Byte DFA plus per-state packed token bitmasks over the tokenizer vocab.
+
+    Mask rows are int32 [W] with W = ceil(vocab/32); bit (t % 32) of word
+    (t // 32) set iff token t is allowed. The fused sampler kernel consumes
+    the packed rows directly; the torch fallback expands them to bool.
+    """
+
+    def __init__(self, dfa: DFA, tokenizer, device: str = "cpu"):
         self.dfa = dfa
+        self.tokenizer = tokenizer
         self.device = device
-        self._masks: Dict[int, torch.Tensor] = {}
-        # dense device table of visited-state masks for batched gathers
-        self._table: Optional[torch.Tensor] = None
+        self.vocab = tokenizer.vocab_size
+        self.W = (self.vocab + 31) // 32
+        S = dfa.num_states
+        # dense byte-transition table; row S = dead sentinel (self-loop)
+        trans = np.full((S + 1, 256), S, dtype=np.int32)
+        for s, tr in enumerate(dfa.transitions):
+            for b, t in tr.items():
+                trans[s, b] = t
+        self._trans_np = trans
+        self._trans_t: Optional[torch.Tensor] = None  # device copy, lazy
+        self._accepting = dfa.accepting
+
+        self._packed: Optional[torch.Tensor] = None   # [rows, W] int32 device
         self._state_row: Dict[int, int] = {}
+        if S <= _EAGER_STATE_LIMIT:
+            self._build_rows(list(range(S)))
 
     @classmethod
-    def from_schema(cls, schema: dict, device: str = "cpu") -> "GuidedFSM":
-        return cls(compile_dfa(schema_to_regex(schema)), device)
+    def from_schema(cls, schema: dict, tokenizer=None,
+                    device: str = "cpu") -> "GuidedFSM":
+        if tokenizer is None:
+            from .tokenizer import get_tokenizer
+
+            tokenizer = get_tokenizer()
+        return cls(compile_dfa(schema_to_regex(schema)), tokenizer, device)
 
     def start_state(self) -> int:
         return 0
 
-    def mask_for(self, state: int) -> torch.Tensor:
-        """Bool mask [TOKENIZER_VOCAB]; True = allowed."""
-        m = self._masks.get(state)
-        if m is None:
-            mask = torch.zeros(TOKENIZER_VOCAB, dtype=torch.bool)
-            if state >= 0:
-                for b in self.dfa.transitions[state]:
-                    mask[BYTE_OFFSET + b] = True
-                if state in self.dfa.accepting:
-                    mask[EOS_ID] = True
-            m = mask.to(self.device)
-            self._masks[state] = m
-        return m
+    # ---- mask construction (vectorized token byte-walks) ----
+
+    def _allowed_bool(self, states: List[int]) -> np.ndarray:
+        """allowed [len(states), vocab] bool via batched DFA gathers (on the
+        FSM's device when it is a GPU — full-vocab x many-state builds are
+        gather-bound)."""
+        if self.device.startswith("cuda") and torch.cuda.is_available():
+            return self._allowed_bool_torch(states)
+        trans = self._trans_np
+        dead = trans.shape[0] - 1
+        st = np.asarray(states, dtype=np.int64)
+        B = len(st)
+        out = np.zeros((B, self.vocab), dtype=bool)
+        # EOS in accepting states
+        acc = np.fromiter((s in self._accepting for s in states), bool, B)
+        out[acc, EOS_ID] = True
+        for ids, bts in _token_groups(self.tokenizer):
+            # chunk tokens to bound the [B, T] gather intermediates
+            T = len(ids)
+            step = max(1, (1 << 24) // max(1, B))
+            for t0 in range(0, T, step):
+                sl = slice(t0, t0 + step)
+                cur = np.broadcast_to(st[:, None], (B, len(ids[sl]))).astype(
+                    np.int32)
+                for j in range(bts.shape[1]):
+                    cur = trans[cur, bts[sl, j][None, :]]
+                out[:, ids[sl]] = cur != dead
+        return out
+
+    def _allowed_bool_torch(self, states: List[int]) -> np.ndarray:
+        dev = self.device
+        if self._trans_t is None:
+            self._trans_t = torch.from_numpy(self._trans_np).to(dev)
+        trans = self._trans_t
+        dead = trans.shape[0] - 1
+        st = torch.tensor(states, dtype=torch.long, device=dev)
+        B = len(states)
+        out = torch.zeros((B, self.vocab), dtype=torch.bool, device=dev)
+        for ids, bts in _token_groups(self.tokenizer):
+            ids_t = torch.from_numpy(ids).to(dev)
+            bts_t = torch.from_numpy(bts).to(dev)
+            cur = st[:, None].expand(B, len(ids)).contiguous()
+            for j in range(bts_t.shape[1]):
+                cur = trans[cur, bts_t[:, j][None, :].expand_as(cur)].long()
+            out[:, ids_t] = cur != dead
+        acc = np.fromiter((s in self._accepting for s in states), bool, B)
+        res = out.cpu().numpy()
+        res[acc, EOS_ID] = True
+        return res
+
+    def _build_rows(self, states: List[int]) -> None:
+        new = [s for s in states if s not in self._state_row and s >= 0]
+        if not new:
+            return
+        allowed = self._allowed_bool(new)
+        pad = (-allowed.shape[1]) % 32
+        if pad:
+            allowed = np.pad(allowed, ((0, 0), (0, pad)))
+        packed = np.packbits(allowed, axis=1, bitorder="little")
+        packed = packed.view("<u4").astype(np.int32, copy=False)
+        rows = torch.from_numpy(np.ascontiguousarray(packed)).to(self.device)
+        if self._packed is None:
+            self._packed = rows
+        else:
+            self._packed = torch.cat([self._packed, rows], dim=0)
+        base = len(self._state_row)
+        for i, s in enumerate(new):
+            self._state_row[s] = base + i
 
     def mask_rows(self, states) -> torch.Tensor:
-        """Batched masks [n, TOKENIZER_VOCAB] via ONE device gather from a
-        growing visited-state table (the per-row `mask[i] = mask_for(s)`
-        pattern costs one device copy kernel per guided row per step)."""
-        rows = []
+        """Packed masks [n, W] int32 (one device gather); builds rows for
+        unseen states in one vectorized batch."""
+        missing = [s for s in states if s not in self._state_row]
+        if missing:
+            self._build_rows(sorted(set(missing)))
         row_of = self._state_row
-        for st in states:
-            r = row_of.get(st)
-            if r is None:
-                m = self.mask_for(st)
-                r = len(row_of)
-                row_of[st] = r
-                if self._table is None:
-                    self._table = torch.zeros(64, TOKENIZER_VOCAB,
-                                              dtype=torch.bool,
-                                              device=m.device)
-                elif r >= self._table.shape[0]:
-                    bigger = torch.zeros(self._table.shape[0] * 2,
-                                         TOKENIZER_VOCAB, dtype=torch.bool,
-                                         device=self._table.device)
-                    bigger[:self._table.shape[0]] = self._table
-                    self._table = bigger
-                self._table[r] = m
-            rows.append(r)
-        idx = torch.tensor(rows, dtype=torch.long, device=self._table.device)
-        return self._table[idx]
+        idx = torch.tensor([row_of[s] for s in states], dtype=torch.long,
+                           device=self._packed.device)
+        return self._packed[idx]
+
+    def mask_for(self, state: int) -> torch.Tensor:
+        """Bool mask [vocab]; True = allowed (tests / torch fallback)."""
+        packed = self.mask_rows([state])[0]
+        return unpack_mask(packed.unsqueeze(0), self.vocab)[0]
+
+    # ---- stepping ----
 
     def advance(self, state: int, token_id: int) -> int:
-        """Next state after a sampled token; EOS leaves the state unchanged."""
+        """Next state after a sampled token (multi-byte walk); EOS leaves the
+        state unchanged."""
         if token_id == EOS_ID:
             return state
-        return self.dfa.step(state, token_id - BYTE_OFFSET)
+        for b in self.tokenizer.token_bytes(token_id):
+            state = self.dfa.step(state, b)
+            if state < 0:
+                return -1
+        return state
 
     def is_accepting(self, state: int) -> bool:
-        return state in self.dfa.accepting
+        return state in self._accepting
 
     def must_stop(self, state: int) -> bool:
         """Accepting with no outgoing transitions: only EOS possible."""
-        return state in self.dfa.accepting and not self.dfa.transitions[state]
+        return state in self._accepting and not self.dfa.transitions[state]
+
+
+def unpack_mask(packed: torch.Tensor, vocab: int) -> torch.Tensor:
+    """[n, W] int32 packed rows -> [n, vocab] bool (torch sampler fallback)."""
+    shifts = torch.arange(32, device=packed.device, dtype=torch.int32)
+    bits = packed.unsqueeze(-1) >> shifts  # [n, W, 32]
+    return (bits & 1).to(torch.bool).reshape(packed.shape[0], -1)[:, :vocab]
+
+
+def reasoning_wrapper_schema(content_schema: Optional[dict]) -> dict:
+    """Schema forcing `-thinking` models to really generate both fields of
+    the reference's reasoning contract (`/root/reference/sutro/sdk.py:1278-
+    1320` unpacks {content, reasoning_content}, content one level deeper)."""
+    return {
+        "type": "object",
+        "properties": {
+            "reasoning_content": {"type": "string", "maxLength": 512},
+            "content": content_schema if content_schema is not None
+            else {"type": "string", "maxLength": 512},
+        },
+    }
+
+
+def full_mask_row(vocab: int, device) -> torch.Tensor:
+    """Packed all-allowed row [W] (unguided rows in mixed batches). Tail bits
+    beyond vocab are zero; specials PAD/BOS stay set — the sampler's
+    vocab-limit handles the tail and PAD/BOS carry -inf only under FSM rows."""
+    W = (vocab + 31) // 32
+    row = np.full(W * 32, True)
+    row[vocab:] = False
+    packed = np.packbits(row, bitorder="little").view("<u4").astype(np.int32)
+    return torch.from_numpy(packed).to(device)

@@ -66,6 +66,9 @@ class Request:
     alloc_gen: int = 0
     # FSM state to rewind to when the row restarts (set at admission)
     fsm_start_state: int = 0
+    # final text override: set when a stop STRING lands mid-token (BPE tokens
+    # are multi-byte, so the trim point need not be a token boundary)
+    text_override: Optional[str] = None
     _stop_ids: Optional[frozenset] = field(default=None, repr=False)
 
     def stop_ids(self, eos_id: int) -> frozenset:
@@ -100,6 +103,7 @@ class Request:
         self.num_computed_tokens = 0
         self.output_token_ids = []
         self.cumulative_logprob = 0.0
+        self.text_override = None
         self.fsm_state = self.fsm_start_state
         self.alloc_gen += 1
 

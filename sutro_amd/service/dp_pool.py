@@ -61,7 +61,7 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
         eng = LLMEngine(cfg)
         if tp > 1:
             group = eng.tp.group
-        tok = get_tokenizer()
+        tok = eng.tokenizer  # model-vocab-sized BPE
         spec = eng.spec
         tp_rank = rank % tp
         lead = tp_rank == 0
@@ -73,10 +73,13 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
         def admit(msg):
             _, job_id, rows, opts = msg
             schema = opts.get("json_schema")
-            if schema is None and spec.reasoning and not spec.embedding:
-                schema = {"type": "object", "properties": {
-                    "reasoning_content": {"type": "string", "maxLength": 512},
-                    "content": {"type": "string", "maxLength": 512}}}
+            if spec.reasoning and not spec.embedding:
+                # real two-field generation: the FSM forces the model to emit
+                # reasoning_content then content (the user schema, if any,
+                # nests under content for the client's double-unpack)
+                from ..engine.guided import reasoning_wrapper_schema
+
+                schema = reasoning_wrapper_schema(schema)
             if job_id not in fsm_cache:
                 fsm_cache[job_id] = (eng.register_fsm(schema)
                                      if schema is not None and not spec.embedding
@@ -152,7 +155,7 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
                                    "in_tokens": len(req.prompt_token_ids),
                                    "out_tokens": 0}
                     else:
-                        text = tok.decode(req.output_token_ids)
+                        text = eng.output_text(req)
                         n = max(1, len(req.output_token_ids))
                         payload = {
                             "output": text,
@@ -278,11 +281,9 @@ class MultiProcEngineWorker:
                     job.embeddings[row_idx] = payload.get("emb")
                     job.outputs[row_idx] = None
                 else:
-                    text = payload["output"]
-                    if self.spec.reasoning and job.json_schema is not None:
-                        text = json.dumps({"content": text,
-                                           "reasoning_content": ""})
-                    job.outputs[row_idx] = text
+                    # reasoning models already emit the {reasoning_content,
+                    # content} wrapper JSON via the forced FSM (admit())
+                    job.outputs[row_idx] = payload["output"]
                     job.cumulative_logprobs[row_idx] = payload.get(
                         "cumulative_logprob")
                     job.confidence_scores[row_idx] = payload.get("confidence")

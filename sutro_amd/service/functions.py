@@ -52,9 +52,8 @@ class FunctionStore:
         worker = job_service._get_local_worker(fn["model"])
         eng = worker.engine
         from ..engine.request import SamplingParams
-        from ..engine.tokenizer import get_tokenizer
 
-        tok = get_tokenizer()
+        tok = eng.tokenizer
         text = input_data if isinstance(input_data, str) else json.dumps(input_data)
         ids = tok.render_prompt(text, fn.get("system_prompt"))
         fsm_id = None
@@ -76,7 +75,7 @@ class FunctionStore:
         t0 = time.time()
         while not req.finished and time.time() - t0 < 600:
             time.sleep(0.01)
-        out_text = tok.decode(req.output_token_ids)
+        out_text = eng.output_text(req)
         n = max(1, len(req.output_token_ids))
         confidence = float(min(1.0, max(0.0, math.exp(req.cumulative_logprob / n))))
         return {

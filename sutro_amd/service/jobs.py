@@ -119,13 +119,14 @@ class EngineWorker:
         cfg = resolve_engine_config(model, device=device, **(engine_kwargs or {}))
         self.engine = LLMEngine(cfg)
         self.spec = self.engine.spec
-        self.tokenizer = get_tokenizer()
+        self.tokenizer = self.engine.tokenizer  # model-vocab-sized BPE
         self._inbox: List[tuple] = []
         self._req_meta: Dict[int, tuple] = {}  # req_id -> (job, row_idx)
         # compiled guided-decoding FSMs, shared across jobs with the same
         # schema (DFA compilation for deep schemas can take seconds)
         self._fsm_by_schema: Dict[str, int] = {}
         self._cancelled_jobs: set = set()
+        self.dead: Optional[str] = None  # failure message once the loop dies
         self._lock = threading.Lock()
         self._wake = threading.Event()
         self._stop = False
@@ -134,6 +135,13 @@ class EngineWorker:
         self._thread.start()
 
     def submit_job(self, job: JobRecord, service: "JobService") -> None:
+        if self.dead is not None:
+            with job.lock:
+                job.status = JobStatus.FAILED
+                job.failure_reason = {"message": self.dead}
+                job.datetime_completed = _now()
+            service.persist_job(job)
+            return
         with self._lock:
             self._inbox.append((job, service, None))
         self._wake.set()
@@ -167,10 +175,12 @@ class EngineWorker:
         service.persist_job(job)
         fsm_id = None
         schema = job.json_schema
-        if schema is None and self.spec.reasoning:
-            schema = {"type": "object", "properties": {
-                "reasoning_content": {"type": "string", "maxLength": 512},
-                "content": {"type": "string", "maxLength": 512}}}
+        if self.spec.reasoning and not self.spec.embedding:
+            # real two-field generation: FSM forces reasoning_content +
+            # content (user schema nests under content)
+            from ..engine.guided import reasoning_wrapper_schema
+
+            schema = reasoning_wrapper_schema(schema)
             job._auto_reasoning_schema = True  # type: ignore[attr-defined]
         if schema is not None and not self.spec.embedding:
             key = json.dumps(schema, sort_keys=True)
@@ -190,7 +200,10 @@ class EngineWorker:
                 sp.seed = i
             req = eng.add_request(ids, sp, fsm_id=fsm_id, priority=job.priority,
                                   arrival_idx=i, truncate=job.truncate_rows)
-            job.input_tokens += len(req.prompt_token_ids)
+            if rows_idx is None:
+                # resume re-admissions were already counted at original
+                # admission (persisted input_tokens) — don't double-count
+                job.input_tokens += len(req.prompt_token_ids)
             self._req_meta[req.req_id] = (job, i, service)
         job._t_start = t_start  # type: ignore[attr-defined]
         service.persist_job(job)
@@ -206,11 +219,9 @@ class EngineWorker:
                 job.embeddings[row_idx] = emb.tolist() if emb is not None else None
                 job.outputs[row_idx] = None
             else:
-                text = self.tokenizer.decode(req.output_token_ids)
-                if getattr(job, "_auto_reasoning_schema", False):
-                    pass  # text is already the {reasoning_content, content} JSON
-                elif self.spec.reasoning and job.json_schema is not None:
-                    text = json.dumps({"content": text, "reasoning_content": ""})
+                # reasoning models emit the {reasoning_content, content}
+                # wrapper JSON directly via the forced FSM (_admit)
+                text = self.engine.output_text(req)
                 job.outputs[row_idx] = text
                 n = max(1, len(req.output_token_ids))
                 job.cumulative_logprobs[row_idx] = req.cumulative_logprob
@@ -257,7 +268,25 @@ class EngineWorker:
                             self.engine.abort_request(req)
                         self._req_meta.pop(req_id, None)
             if self.engine.has_work():
-                stats = self.engine.step()
+                try:
+                    stats = self.engine.step()
+                except Exception as e:
+                    # a fatal step error (e.g. HIP OOM) must not strand
+                    # in-flight jobs as RUNNING forever: fail them all
+                    msg = f"engine step failed: {type(e).__name__}: {e}"
+                    self.dead = msg
+                    failed = {}
+                    for req_id, (job, _row, service) in list(self._req_meta.items()):
+                        failed[job.job_id] = (job, service)
+                        self._req_meta.pop(req_id, None)
+                    for job, service in failed.values():
+                        with job.lock:
+                            if not JobStatus.is_terminal(job.status):
+                                job.status = JobStatus.FAILED
+                                job.failure_reason = {"message": msg}
+                                job.datetime_completed = _now()
+                                service.persist_job(job, with_results=True)
+                    raise
                 for req in stats.finished:
                     self._finish_row(req)
             else:
@@ -299,7 +328,7 @@ class JobService:
 
     def _estimate_cost(self, job: JobRecord) -> float:
         spec = get_model_spec(job.model)
-        tok = get_tokenizer()
+        tok = get_tokenizer(spec.vocab_size)
         # large p1 estimates sample a ~1M-token prefix (README.md:173 behavior)
         budget = 1_000_000
         in_tokens = 0
@@ -353,10 +382,12 @@ class JobService:
             raise ValueError(
                 f"job exceeds row quota for priority {priority}: "
                 f"{len(inputs)} > {q['row_quota']}")
-        # token quota: byte tokenizer => input tokens ~= utf-8 bytes per row
+        # token quota: fast estimate at ~3 utf-8 bytes per BPE token (the
+        # shipped tokenizer measures 2.7-3 on English/JSON; exact counts
+        # would tokenize the whole job twice)
         est_tokens = sum(
             len(r.encode() if isinstance(r, str) else json.dumps(r).encode())
-            for r in inputs)
+            for r in inputs) // 3 + len(inputs)
         if est_tokens > q["token_quota"]:
             raise ValueError(
                 f"job exceeds token quota for priority {priority}: "

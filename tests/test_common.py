@@ -105,18 +105,32 @@ def test_colored_text_states():
 
 
 def test_tokenizer_roundtrip_property():
-    """Property: the byte tokenizer round-trips ANY unicode text exactly."""
+    """Property: the BPE tokenizer round-trips ANY unicode text exactly at
+    every truncation (byte fallback guarantees coverage)."""
     from hypothesis import given, settings, strategies as st
 
     from sutro_amd.engine.tokenizer import get_tokenizer
 
-    tok = get_tokenizer()
+    toks = [get_tokenizer(), get_tokenizer(512), get_tokenizer(32000)]
 
     @settings(max_examples=200, deadline=None)
     @given(st.text(max_size=200))
     def run(text):
-        ids = tok.encode(text)
-        assert tok.decode(ids) == text
-        assert all(0 <= t < 259 for t in ids)
+        for tok in toks:
+            ids = tok.encode(text)
+            assert tok.decode(ids) == text
+            assert all(0 <= t < tok.vocab_size for t in ids)
 
     run()
+
+
+def test_tokenizer_token_bytes_consistent():
+    """decode == concat(token_bytes) and multi-byte merges really exist."""
+    from sutro_amd.engine.tokenizer import get_tokenizer
+
+    tok = get_tokenizer()
+    s = 'The quick {"name": "value", "score": 42} fox.'
+    ids = tok.encode(s)
+    assert b"".join(tok.token_bytes(i) for i in ids).decode() == s
+    assert any(len(tok.token_bytes(i)) > 1 for i in ids)  # real BPE merges
+    assert len(ids) < len(s.encode())  # compresses vs bytes

@@ -7,7 +7,11 @@ import json
 from hypothesis import given, settings, strategies as st
 
 from sutro_amd.engine.guided import GuidedFSM
-from sutro_amd.engine.tokenizer import BYTE_OFFSET, EOS_ID
+from sutro_amd.engine.tokenizer import EOS_ID, get_tokenizer
+
+# small truncation keeps per-example FSM mask builds fast while still
+# exercising real multi-byte BPE tokens in every walk
+TOK = get_tokenizer(2048)
 
 # ---- random schema generator (the subset the engine supports) ----
 
@@ -63,7 +67,7 @@ def walk(fsm, data, max_steps=400):
         tok = data.draw(st.sampled_from(allowed))
         if tok == EOS_ID:
             return bytes(out), True
-        out.append(tok - BYTE_OFFSET)
+        out.extend(fsm.tokenizer.token_bytes(tok))
         state = fsm.advance(state, tok)
     # out of steps: walk the shortest path to an accepting state (BFS over
     # the DFA, computed once) so unbounded repeats always terminate
@@ -152,7 +156,7 @@ def validate(schema, value):
 @given(st.data())
 def test_masked_walks_produce_schema_valid_json(data):
     schema = data.draw(schemas, label="schema")
-    fsm = GuidedFSM.from_schema(schema)
+    fsm = GuidedFSM.from_schema(schema, tokenizer=TOK)
     out, clean = walk(fsm, data)
     assert clean, f"walk did not terminate: {out[:120]!r}"
     assert fsm.dfa.matches(out)
@@ -205,5 +209,5 @@ def test_valid_values_are_accepted(data):
     schema = data.draw(schemas, label="schema")
     value = sample_value(schema, data)
     text = json.dumps(value, separators=(",", ":"))
-    fsm = GuidedFSM.from_schema(schema)
+    fsm = GuidedFSM.from_schema(schema, tokenizer=TOK)
     assert fsm.dfa.matches(text.encode()), (schema, text)
