@@ -126,13 +126,15 @@ def main():
     eng = LLMEngine(cfg)
     init_s = time.time() - t_init0
 
-    # synthetic row stream: prompts of prompt_len byte-tokens, seeded per
-    # DP replica (identical within a TP group so all its ranks run lockstep)
+    # synthetic row stream: prompts of prompt_len random BPE token ids over
+    # the model's real vocab, seeded per DP replica (identical within a TP
+    # group so all its ranks run lockstep)
+    vocab_hi = eng.tokenizer.vocab_size
     rng = np.random.default_rng(1234 + dp_idx)
 
     def make_prompt():
-        body = rng.integers(3, 259, size=args.prompt_len - 1).tolist()
-        return [1] + body  # BOS + random bytes
+        body = rng.integers(3, vocab_hi, size=args.prompt_len - 1).tolist()
+        return [1] + body  # BOS + random tokens
 
     sp_kwargs = dict(max_tokens=args.max_new, temperature=0.8, top_p=0.95)
     fsm_id = None

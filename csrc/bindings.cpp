@@ -45,6 +45,9 @@ void sutro_hd64_stage_probe(float*, float*, void*, const void*, const void*,
 void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
                     const float*, const void*, const void*, float, int, int,
                     int, int, int, int, int, hipStream_t);
+void sutro_sampler_fused(const void*, int, const float*, const float*,
+                         const int*, const float*, const unsigned int*, int,
+                         long, int, int, int*, float*, hipStream_t);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -218,6 +221,38 @@ torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+void sampler_fused(torch::Tensor logits, torch::Tensor temps,
+                   torch::Tensor topps, torch::Tensor topks, torch::Tensor us,
+                   c10::optional<torch::Tensor> mask, long vl,
+                   torch::Tensor out_tok, torch::Tensor out_lp) {
+  CHECK_CUDA(logits); CHECK_CONTIG(logits);
+  const bool f32 = logits.scalar_type() == at::kFloat;
+  TORCH_CHECK(f32 || logits.scalar_type() == at::kBFloat16,
+              "logits must be bf16 or f32");
+  const long n = logits.size(0), v_row = logits.size(1);
+  TORCH_CHECK(vl <= v_row, "vocab_limit exceeds logits row");
+  TORCH_CHECK(temps.numel() >= n && topps.numel() >= n && topks.numel() >= n
+              && us.numel() >= n, "param tensors too small");
+  TORCH_CHECK(topks.scalar_type() == at::kInt, "topks must be int32");
+  const unsigned int* mptr = nullptr;
+  int w_words = 0;
+  if (mask.has_value()) {
+    auto& mt = mask.value();
+    CHECK_CUDA(mt); CHECK_CONTIG(mt);
+    TORCH_CHECK(mt.scalar_type() == at::kInt, "mask must be int32 packed");
+    TORCH_CHECK(mt.size(0) == n, "mask rows mismatch");
+    w_words = (int)mt.size(1);
+    TORCH_CHECK((long)w_words * 32 >= vl, "mask too narrow for vocab_limit");
+    mptr = (const unsigned int*)mt.data_ptr();
+  }
+  sutro_sampler_fused(logits.data_ptr(), f32 ? 1 : 0,
+                      temps.data_ptr<float>(), topps.data_ptr<float>(),
+                      topks.data_ptr<int>(), us.data_ptr<float>(), mptr,
+                      (int)n, v_row, (int)vl, w_words,
+                      out_tok.data_ptr<int>(), out_lp.data_ptr<float>(),
+                      cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, CDNA4)");
   m.def("fused_add_rmsnorm", &fused_add_rmsnorm, "residual+=x; x=rmsnorm");
@@ -229,6 +264,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
   m.def("hd64_stage_probe", &hd64_stage_probe, "D=64 decode stage dump probe");
+  m.def("sampler_fused", &sampler_fused,
+        "fused mask+temp+topk/topp+sample+logprob (radix descent, no sort)");
   m.def("gemm_tn", &gemm_tn, "bf16 TN GEMM (MFMA, glds dbuf)",
         py::arg("x"), py::arg("w"), py::arg("res") = c10::nullopt,
         py::arg("bm") = 256, py::arg("bn") = 256, py::arg("swz") = 2,

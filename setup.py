@@ -21,6 +21,7 @@ SOURCES = [
     "csrc/attn_decode_mfma.hip",
     "csrc/attn_prefill.hip",
     "csrc/gemm_tn.hip",
+    "csrc/sampler.hip",
 ]
 
 setup(

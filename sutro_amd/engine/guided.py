@@ -308,7 +308,9 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return alt(*[schema_to_regex({**schema, "type": ti}, defs, depth + 1) for ti in t])
     if t == "string":
         ml = min(int(schema.get("maxLength", _MAX_STR)), _MAX_STR)
-        return cat(cls('"'), _json_string_body(ml), cls('"'))
+        lo = max(0, min(int(schema.get("minLength", 0)), ml))
+        body = rep(alt(_STR_CHAR, _ESCAPE), lo, ml)
+        return cat(cls('"'), body, cls('"'))
     if t == "integer":
         lo, hi = schema.get("minimum"), schema.get("maximum")
         if lo is not None and hi is not None and 0 <= hi - lo <= 4096:
@@ -563,7 +565,8 @@ def reasoning_wrapper_schema(content_schema: Optional[dict]) -> dict:
     return {
         "type": "object",
         "properties": {
-            "reasoning_content": {"type": "string", "maxLength": 512},
+            "reasoning_content": {"type": "string", "minLength": 1,
+                                  "maxLength": 512},
             "content": content_schema if content_schema is not None
             else {"type": "string", "maxLength": 512},
         },

@@ -159,3 +159,11 @@ def mean_pool_normalize(hidden: torch.Tensor, query_start_locs: torch.Tensor) ->
         _require_hip().mean_pool_normalize(out, hidden, query_start_locs)
         return out
     return torch_ref.mean_pool_normalize(hidden, query_start_locs)
+
+
+def sampler_fused(logits, temps, top_ps, top_ks, u, packed_mask, vocab_limit,
+                  out_tok, out_lp):
+    """Fused mask+temperature+top-k/top-p+sample+logprob (csrc/sampler.hip).
+    GPU-only: logits must be CUDA; raises if the extension is missing."""
+    _require_hip().sampler_fused(logits, temps, top_ps, top_ks, u,
+                                 packed_mask, vocab_limit, out_tok, out_lp)

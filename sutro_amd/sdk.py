@@ -609,12 +609,20 @@ class Sutro(ObservabilityMixin, EmbeddingTemplates, ClassificationTemplates,
             inner = []
             for p in parsed:
                 c = p.get("content") if p else None
+                obj = None
                 if isinstance(c, str) and c.startswith("{"):
                     try:
-                        inner.append(json.loads(c))
+                        obj = json.loads(c)
                     except json.JSONDecodeError:
-                        inner.append({"content": c,
-                                      "reasoning_content": p.get("reasoning_content")})
+                        obj = None
+                elif isinstance(c, dict):
+                    # engine-side reasoning FSM nests the user schema directly
+                    obj = c
+                if isinstance(obj, dict):
+                    row = dict(obj)
+                    row.setdefault("reasoning_content",
+                                   p.get("reasoning_content") if p else None)
+                    inner.append(row)
                 else:
                     inner.append({"content": c,
                                   "reasoning_content": p.get("reasoning_content")

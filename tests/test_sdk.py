@@ -183,6 +183,30 @@ def test_reasoning_model_output_unpacks(local_client):
                                 sampling_params={"max_tokens": 4096})
     df = _await(local_client, job_id)
     assert "content" in df.columns and "reasoning_content" in df.columns
+    # REAL two-field generation: the FSM forces a non-empty reasoning field
+    # (not a hardcoded wrapper)
+    assert all(isinstance(r, str) and len(r) > 0
+               for r in df["reasoning_content"])
+
+
+def test_reasoning_model_schema_double_unpack(local_client):
+    """User schema nests under content; the client unpacks one level deeper
+    (reference sdk.py:1278-1320 behavior)."""
+    schema = {"type": "object", "properties": {
+        "label": {"enum": ["pos", "neg"]},
+        "score": {"type": "integer", "minimum": 0, "maximum": 9}}}
+    job_id = local_client.infer(["classify me"],
+                                model="qwen-3.5-2b-thinking",
+                                stay_attached=False,
+                                output_schema=schema,
+                                sampling_params={"max_tokens": 4096})
+    df = _await(local_client, job_id)
+    assert "label" in df.columns and "score" in df.columns
+    assert "reasoning_content" in df.columns
+    assert df["label"][0] in ("pos", "neg")
+    assert 0 <= int(df["score"][0]) <= 9
+    assert isinstance(df["reasoning_content"][0], str)
+    assert len(df["reasoning_content"][0]) > 0
 
 
 def test_progress_stream_protocol(local_client):
