@@ -163,7 +163,9 @@ def main():
                             fsm_id=fsm_id, priority=1, arrival_idx=arrival[0])
             arrival[0] += 1
 
-    refill()
+    ramp = not eng.spec.embedding and not args.no_refill
+    if not ramp:
+        refill()
 
     def sync():
         if have_gpu:
@@ -173,11 +175,29 @@ def main():
             if have_gpu:
                 torch.cuda.synchronize()
 
-    # prime: finish the initial prefill wave before warmup so the measured
-    # window reflects steady state regardless of how small W is (at batch
-    # 2048 the first fill alone is ~8 steps of pure prefill). Generative
-    # only — embedding rows never leave prefill.
-    if not eng.spec.embedding:
+    # prime to STEADY STATE. A thundering-herd start (all rows admitted at
+    # once) makes the job periodic with period ~max_new steps: short driver
+    # windows then catch a prefill-heavy or decode-only phase and report
+    # either half or double the true steady rate (r1 driver run measured
+    # 4.0k tok/s on a 20-step window vs 7.8k on 32 steps, same build).
+    # Fix: ramp admission over one row lifetime so rows finish (and refill)
+    # uniformly — every window then carries the same prefill/decode mix.
+    # Generative only — embedding rows never leave prefill.
+    if ramp:
+        per_step = max(1, args.batch // max(1, args.max_new))
+        admitted = 0
+        for _ in range(4 * args.batch):
+            if admitted >= args.batch and not any(
+                    r.in_prefill for r in eng.scheduler.running):
+                break
+            for _ in range(min(per_step, args.batch - admitted)):
+                eng.add_request(make_prompt(), SamplingParams(**sp_kwargs),
+                                fsm_id=fsm_id, priority=1,
+                                arrival_idx=arrival[0])
+                arrival[0] += 1
+                admitted += 1
+            eng.step()
+    elif not eng.spec.embedding:
         for _ in range(256):
             sch = eng.scheduler
             if not sch.running or all(not r.in_prefill for r in sch.running):

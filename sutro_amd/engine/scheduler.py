@@ -140,8 +140,14 @@ class Scheduler:
         if thr is None:
             thr = self.cfg.max_tokens_per_step
         # p0 (interactive) rows bypass accumulation — only production (p1)
-        # admission is batched for throughput
-        if decode_reqs and not prefill_reqs and thr > 0 and not self.waiting_p0:
+        # admission is batched for throughput. Hold back ONLY while the
+        # running pool is near capacity: with uniform row arrivals a fixed
+        # token threshold would starve a draining pool (rows finishing
+        # faster than the accumulation fills) — seen as a 1-row pool in the
+        # ramped bench before this condition.
+        if (decode_reqs and not prefill_reqs and thr > 0
+                and not self.waiting_p0
+                and len(self.running) >= (3 * self.cfg.max_num_seqs) // 4):
             avail = 0
             for q in (self.waiting_p0, self.waiting_p1):
                 for r in q:

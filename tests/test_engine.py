@@ -375,7 +375,10 @@ def test_prefill_accumulation_p0_bypass():
     from sutro_amd.engine.scheduler import Scheduler
 
     spec = tiny_spec_for_tests()
-    cfg = EngineConfig(spec=spec, device="cpu", max_num_seqs=16,
+    # max_num_seqs=2 so the single running decode already counts as a
+    # near-capacity pool (the hold only applies at >= 3/4 utilization; a
+    # draining pool admits immediately — see scheduler.py)
+    cfg = EngineConfig(spec=spec, device="cpu", max_num_seqs=2,
                        max_model_len=256, max_tokens_per_step=64,
                        num_kv_blocks=256)  # threshold = None -> 64 tokens
     kv = PagedKVCache(num_layers=1, num_blocks=256, num_kv_heads=1,

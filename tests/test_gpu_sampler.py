@@ -126,7 +126,7 @@ def test_sampler_fsm_mask_exact():
     # 8-token support: boundaries are far apart -> exact agreement
     assert torch.equal(tok, rtok)
     assert torch.allclose(lp, rlp, atol=2e-3, rtol=1e-3)
-    assert bool(mb.gather(1, tok.cpu().unsqueeze(1)).all())
+    assert bool(mb.cpu().gather(1, tok.cpu().unsqueeze(1)).all())
 
 
 def test_sampler_vocab_limit_tail_masked():
