@@ -48,6 +48,9 @@ void sutro_qkv_prep(const void*, void*, void*, void*, const long*, const long*,
 void sutro_sampler_fused(const void*, int, const float*, const float*,
                          const int*, const float*, const unsigned int*, int,
                          long, int, int, int*, float*, hipStream_t);
+void sutro_grouped_gemm(void*, const void*, const void*, const int*,
+                        const int*, const int*, int, int, int, long, int,
+                        hipStream_t);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -221,6 +224,33 @@ torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+void grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
+                  c10::optional<torch::Tensor> row_tok, torch::Tensor tile_off,
+                  torch::Tensor counts, long max_tiles, bool gate_silu) {
+  CHECK_CUDA(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_CUDA(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  CHECK_CONTIG(out); CHECK_BF16(out);
+  const long E = w.size(0), N = w.size(1), K = w.size(2);
+  const long n_cols = gate_silu ? N / 2 : N;
+  TORCH_CHECK(K % 64 == 0 && n_cols % 64 == 0,
+              "grouped_gemm needs K, n_cols multiples of 64 (K=", K,
+              ", n_cols=", n_cols, ")");
+  TORCH_CHECK(a.size(1) == K, "A K mismatch");
+  TORCH_CHECK(out.size(1) == n_cols, "out width mismatch");
+  TORCH_CHECK(tile_off.scalar_type() == at::kInt
+              && counts.scalar_type() == at::kInt, "tile_off/counts int32");
+  TORCH_CHECK(tile_off.numel() == E + 1 && counts.numel() == E, "seg sizes");
+  const int* rt = nullptr;
+  if (row_tok.has_value()) {
+    TORCH_CHECK(row_tok->scalar_type() == at::kInt, "row_tok int32");
+    rt = row_tok->data_ptr<int>();
+  }
+  sutro_grouped_gemm(out.data_ptr(), a.data_ptr(), w.data_ptr(), rt,
+                     tile_off.data_ptr<int>(), counts.data_ptr<int>(), (int)E,
+                     (int)max_tiles, (int)n_cols, K, gate_silu ? 1 : 0,
+                     cur_stream());
+}
+
 void sampler_fused(torch::Tensor logits, torch::Tensor temps,
                    torch::Tensor topps, torch::Tensor topks, torch::Tensor us,
                    c10::optional<torch::Tensor> mask, long vl,
@@ -264,6 +294,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
   m.def("hd64_stage_probe", &hd64_stage_probe, "D=64 decode stage dump probe");
+  m.def("grouped_gemm", &grouped_gemm,
+        "dropless MoE grouped GEMM (padded segments, static grid)");
   m.def("sampler_fused", &sampler_fused,
         "fused mask+temp+topk/topp+sample+logprob (radix descent, no sort)");
   m.def("gemm_tn", &gemm_tn, "bf16 TN GEMM (MFMA, glds dbuf)",

@@ -22,6 +22,7 @@ SOURCES = [
     "csrc/attn_prefill.hip",
     "csrc/gemm_tn.hip",
     "csrc/sampler.hip",
+    "csrc/grouped_gemm.hip",
 ]
 
 setup(

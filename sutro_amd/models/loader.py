@@ -15,9 +15,13 @@ accept split q/k/v and gate/up tensors.
 from __future__ import annotations
 
 import os
-from typing import Dict, Iterator, Tuple
+import re
+from typing import Dict, Iterator, List, Tuple
 
 import torch
+
+# HF per-expert tensors: ...mlp.experts.<E>.<gate|up|down>_proj.weight
+_EXPERT_RE = re.compile(r"^(.*\.mlp)\.experts\.(\d+)\.(gate|up|down)_proj\.weight$")
 
 
 def _iter_safetensors(path: str) -> Iterator[Tuple[str, torch.Tensor]]:
@@ -79,10 +83,19 @@ def load_weights(model: torch.nn.Module, path: str, strict: bool = True) -> int:
         "up_proj.weight": ("gate_up_proj.weight", "up"),
     }
 
+    pending_experts: Dict[str, Dict[Tuple[int, str], torch.Tensor]] = {}
+    unexpected: List[str] = []
     for raw_name, tensor in _iter_safetensors(path):
         name = _canon(raw_name)
         if name in params:
             assign(name, tensor)
+            continue
+        em = _EXPERT_RE.match(name)
+        if em and f"{em.group(1)}.gate_up" in params:
+            # per-expert stacked MoE weights ([E, N, K] layout = per-expert
+            # torch-Linear tensors stacked, so copies are direct)
+            pending_experts.setdefault(em.group(1), {})[
+                (int(em.group(2)), em.group(3))] = tensor
             continue
         # split projections to be merged
         for suffix, (target_suffix, part) in merge_map.items():
@@ -95,8 +108,7 @@ def load_weights(model: torch.nn.Module, path: str, strict: bool = True) -> int:
         else:
             if name == "lm_head.weight" and "lm_head.weight" not in params:
                 continue  # tied embeddings
-            if not strict:
-                continue
+            unexpected.append(raw_name)
 
     for target, parts in pending_merge.items():
         if target.endswith("qkv_proj.weight"):
@@ -108,6 +120,26 @@ def load_weights(model: torch.nn.Module, path: str, strict: bool = True) -> int:
                              f"{sorted(parts)}")
         assign(target, torch.cat([parts[k] for k in needed], dim=0))
 
+    for prefix, parts in pending_experts.items():
+        n_exp = max(e for e, _ in parts) + 1
+        for what, pname in (("gate/up", f"{prefix}.gate_up"),
+                            ("down", f"{prefix}.down")):
+            need = (("gate", "up") if what == "gate/up" else ("down",))
+            full_e = []
+            for e in range(n_exp):
+                try:
+                    full_e.append(torch.cat([parts[(e, k)] for k in need],
+                                            dim=0))
+                except KeyError:
+                    raise ValueError(
+                        f"incomplete expert tensors for {prefix} expert {e}")
+            assign(pname, torch.stack(full_e))
+
+    if unexpected and strict:
+        raise ValueError(
+            f"checkpoint has {len(unexpected)} tensors that map to no "
+            f"parameter (strict=True): {unexpected[:6]}"
+            f"{'...' if len(unexpected) > 6 else ''}")
     missing = [n for n, p in params.items()
                if n not in loaded and p.dim() >= 2]
     if missing and strict:

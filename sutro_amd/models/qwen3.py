@@ -141,31 +141,34 @@ class Qwen3MoE(nn.Module):
         self.top_k = spec.experts_per_token
         h, m = spec.hidden_size, spec.moe_intermediate_size
         self.router = nn.Linear(h, spec.num_experts, bias=False, dtype=dtype)
+        # weights are [E, N, K] K-contiguous — per-expert torch-Linear
+        # layout: direct MFMA K-fragments in the grouped kernel AND direct
+        # per-expert copies from HF checkpoints (experts.N.gate_proj etc.)
         if self.ep:
             assert spec.num_experts % tp.size == 0
             e_l = spec.num_experts // tp.size
             self.experts_per_rank = e_l
             self.expert_base = tp.rank * e_l
-            self.gate_up = nn.Parameter(torch.empty(e_l, h, 2 * m, dtype=dtype))
-            _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 0, tp)
-            self.down = nn.Parameter(torch.empty(e_l, m, h, dtype=dtype))
-            _mark_shard(self.down, (spec.num_experts, m, h), 0, tp)
+            self.gate_up = nn.Parameter(torch.empty(e_l, 2 * m, h, dtype=dtype))
+            _mark_shard(self.gate_up, (spec.num_experts, 2 * m, h), 0, tp)
+            self.down = nn.Parameter(torch.empty(e_l, h, m, dtype=dtype))
+            _mark_shard(self.down, (spec.num_experts, h, m), 0, tp)
         else:
             assert m % tp.size == 0
             m_l = m // tp.size
             self.experts_per_rank = spec.num_experts
             self.expert_base = 0
             self.gate_up = nn.Parameter(
-                torch.empty(spec.num_experts, h, 2 * m_l, dtype=dtype))
-            _mark_shard(self.gate_up, (spec.num_experts, h, 2 * m), 2, tp,
+                torch.empty(spec.num_experts, 2 * m_l, h, dtype=dtype))
+            _mark_shard(self.gate_up, (spec.num_experts, 2 * m, h), 1, tp,
                         row_sections=[(0, m), (m, m)])
             self.down = nn.Parameter(
-                torch.empty(spec.num_experts, m_l, h, dtype=dtype))
-            _mark_shard(self.down, (spec.num_experts, m, h), 1, tp)
+                torch.empty(spec.num_experts, h, m_l, dtype=dtype))
+            _mark_shard(self.down, (spec.num_experts, h, m), 2, tp)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.is_cuda:
-            return self.tp.all_reduce(self._forward_batched(x))
+            return self.tp.all_reduce(self._forward_grouped(x))
         return self.tp.all_reduce(self._forward_loop(x))
 
     def _forward_loop(self, x: torch.Tensor) -> torch.Tensor:
@@ -189,8 +192,8 @@ class Qwen3MoE(nn.Module):
                 continue
             if base <= e < base + e_l:  # EP: other ranks own the rest
                 seg = gathered[start:start + n]
-                act = ops.silu_mul(seg @ self.gate_up[e - base])
-                out_sorted[start:start + n] = act @ self.down[e - base]
+                act = ops.silu_mul(seg @ self.gate_up[e - base].T)
+                out_sorted[start:start + n] = act @ self.down[e - base].T
             else:
                 out_sorted[start:start + n] = 0
             start += n
@@ -199,50 +202,65 @@ class Qwen3MoE(nn.Module):
         out.index_add_(0, sorted_tok, out_sorted.float() * w_sorted)
         return out.to(x.dtype)
 
-    def _forward_batched(self, x: torch.Tensor,
-                         capacity_factor: float = 2.0) -> torch.Tensor:
-        """Capacity-padded batched-GEMM execution (GPU hot path).
+    _GG_BM = 64  # grouped-gemm tile height (csrc/grouped_gemm.hip GG_BM)
 
-        Tokens are scattered into per-expert slots of a fixed-capacity buffer
-        and run through two hipBLASLt batched GEMMs. Shapes are static in T,
-        with no host synchronization, so decode steps stay hipGraph-capturable.
-        Assignments beyond an expert's capacity (cap = T*top_k/E *
-        capacity_factor) are dropped and their routing weight renormalized
-        away — with top-k renormalized router weights this is the standard
-        capacity-factor semantics."""
-        import math as _math
+    def _forward_grouped(self, x: torch.Tensor) -> torch.Tensor:
+        """EXACT dropless top-k execution on the GPU hot path.
 
+        Assignments are sorted by expert into per-expert segments padded to
+        the 64-row kernel tile, so the grouped-GEMM grid is a static upper
+        bound (hipGraph-capturable) and NO assignment is ever dropped —
+        replaces the round-1 capacity-factor bmm path that silently dropped
+        over-capacity tokens. All shaping tensors stay
+        on device (no host sync)."""
         T, h = x.shape
         k = self.top_k
         E_l, base = self.experts_per_rank, self.expert_base
-        cap = max(8, int(_math.ceil(T * k / self.num_experts * capacity_factor)))
-        cap = min(cap, T)  # an expert can't hold more than every token once
+        BM = self._GG_BM
+        dev = x.device
         weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
         flat_e = idx.reshape(-1)                          # [T*k]
-        flat_tok = torch.arange(T, device=x.device).repeat_interleave(k)
+        flat_tok = torch.arange(T, device=dev).repeat_interleave(k)
         order = torch.argsort(flat_e, stable=True)
         sorted_e = flat_e[order]
         sorted_tok = flat_tok[order]
+        e_loc = sorted_e - base
+        valid = (e_loc >= 0) & (e_loc < E_l)              # EP: local only
+        counts = torch.bincount(e_loc[valid], minlength=E_l).to(torch.int32)
+        padded = (counts + (BM - 1)) // BM * BM
+        pad_off = torch.zeros(E_l + 1, dtype=torch.int32, device=dev)
+        pad_off[1:] = torch.cumsum(padded, 0)
+        tile_off = pad_off // BM
         # rank of each assignment within its expert segment
         first = torch.searchsorted(sorted_e, sorted_e, side="left")
-        pos = torch.arange(T * k, device=x.device) - first
-        e_loc = sorted_e - base
-        valid = (pos < cap) & (e_loc >= 0) & (e_loc < E_l)  # EP: local only
-        # overflow/non-local assignments land in a trash row (static shapes:
-        # boolean-mask indexing would break hipGraph capture)
-        trash = E_l * cap
-        slot = torch.where(valid, e_loc * cap + pos,
-                           torch.full_like(pos, trash))
-        buf = x.new_zeros(E_l * cap + 1, h)
-        buf.index_put_((slot,), x[sorted_tok])
-        act = ops.silu_mul(
-            torch.bmm(buf[:E_l * cap].view(E_l, cap, h), self.gate_up)
-            .view(E_l * cap, -1))
-        out_buf = torch.bmm(act.view(E_l, cap, -1), self.down).view(E_l * cap, h)
-        gathered = out_buf[slot.clamp(max=E_l * cap - 1)].float()
-        out = torch.zeros(T, h, dtype=torch.float32, device=x.device)
-        w_sorted = weights.reshape(-1)[order] * valid.float()
-        out.index_add_(0, sorted_tok, gathered * w_sorted.unsqueeze(-1))
+        pos = torch.arange(T * k, device=dev) - first
+        rows_max = T * k + E_l * BM                       # static bound
+        row_tok = torch.full((rows_max,), -1, dtype=torch.int32, device=dev)
+        padpos_sorted = (pad_off[e_loc.clamp(0, E_l - 1).long()].long()
+                         + pos)
+        padpos_sorted = torch.where(valid, padpos_sorted,
+                                    torch.full_like(padpos_sorted,
+                                                    rows_max - 1))
+        row_tok[padpos_sorted[valid]] = sorted_tok[valid].to(torch.int32)
+        max_tiles = rows_max // BM
+
+        m_l = self.gate_up.shape[1] // 2
+        act = x.new_empty(rows_max, m_l)
+        ops.grouped_gemm(act, x, self.gate_up, row_tok, tile_off, counts,
+                         max_tiles, True)
+        out_sorted = x.new_empty(rows_max, h)
+        ops.grouped_gemm(out_sorted, act, self.down, None, tile_off, counts,
+                         max_tiles, False)
+        # deterministic combine in fixed k-order (no float atomics): map each
+        # original (token, j) assignment back to its padded row
+        padpos = torch.empty(T * k, dtype=torch.long, device=dev)
+        padpos[order] = padpos_sorted
+        w_flat = weights.reshape(-1).clone()
+        inv_valid = torch.zeros(T * k, dtype=torch.bool, device=dev)
+        inv_valid[order] = valid
+        w_flat = torch.where(inv_valid, w_flat, torch.zeros_like(w_flat))
+        contrib = out_sorted[padpos.view(T, k)].float()
+        out = (contrib * w_flat.view(T, k, 1)).sum(dim=1)
         return out.to(x.dtype)
 
 

@@ -167,3 +167,14 @@ def sampler_fused(logits, temps, top_ps, top_ks, u, packed_mask, vocab_limit,
     GPU-only: logits must be CUDA; raises if the extension is missing."""
     _require_hip().sampler_fused(logits, temps, top_ps, top_ks, u,
                                  packed_mask, vocab_limit, out_tok, out_lp)
+
+
+def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu):
+    """Dropless-MoE grouped GEMM: csrc/grouped_gemm.hip on GPU, torch
+    reference on CPU (same padded-segment layout)."""
+    if _use_hip(a):
+        _require_hip().grouped_gemm(out, a, w, row_tok, tile_off, counts,
+                                    max_tiles, gate_silu)
+        return out
+    return torch_ref.grouped_gemm(out, a, w, row_tok, tile_off, counts,
+                                  max_tiles, gate_silu)

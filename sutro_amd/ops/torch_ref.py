@@ -138,3 +138,30 @@ def topk_softmax_router(
     w, idx = probs.topk(top_k, dim=-1)
     w = w / w.sum(dim=-1, keepdim=True)
     return w, idx
+
+
+def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu,
+                 bm: int = 64):
+    """Reference for csrc/grouped_gemm.hip: per-expert segments padded to the
+    bm tile; gate_silu fuses silu(gate)*up over the two N-halves of w.
+    Padding rows of `out` are left untouched (the kernel writes deterministic
+    garbage there; consumers must ignore them either way)."""
+    E, N, K = w.shape
+    n_cols = N // 2 if gate_silu else N
+    for e in range(E):
+        c = int(counts[e])
+        if c == 0:
+            continue
+        s0 = int(tile_off[e]) * bm
+        if row_tok is not None:
+            toks = row_tok[s0:s0 + c].long().clamp(min=0)
+            rows = a[toks].float()
+        else:
+            rows = a[s0:s0 + c].float()
+        if gate_silu:
+            g = rows @ w[e, :n_cols].T.float()
+            u = rows @ w[e, n_cols:].T.float()
+            out[s0:s0 + c] = (torch.nn.functional.silu(g) * u).to(out.dtype)
+        else:
+            out[s0:s0 + c] = (rows @ w[e].T.float()).to(out.dtype)
+    return out

@@ -159,9 +159,10 @@ def test_embedding_mode():
         assert abs(np.linalg.norm(v) - 1.0) < 1e-5
 
 
-def test_moe_batched_matches_loop():
-    """Capacity-padded batched MoE == exact per-expert loop (no drops when
-    capacity_factor covers everything)."""
+def test_moe_grouped_matches_loop():
+    """EXACT dropless grouped execution == per-expert loop, including under
+    adversarial routing skew (one expert receiving far beyond the old
+    capacity cap) — no token is ever dropped."""
     import torch
 
     from sutro_amd.models.qwen3 import Qwen3MoE
@@ -177,7 +178,14 @@ def test_moe_batched_matches_loop():
         torch.nn.init.normal_(p, std=0.1)
     x = torch.randn(17, 32)
     ref = moe._forward_loop(x)
-    got = moe._forward_batched(x, capacity_factor=100.0)
+    got = moe._forward_grouped(x)
+    torch.testing.assert_close(got, ref, atol=1e-4, rtol=1e-4)
+    # adversarial skew: bias the router so EVERY token routes to expert 0
+    # (old capacity path would have dropped ~7/8 of the assignments)
+    with torch.no_grad():
+        moe.router.weight[0] += 50.0
+    ref = moe._forward_loop(x)
+    got = moe._forward_grouped(x)
     torch.testing.assert_close(got, ref, atol=1e-4, rtol=1e-4)
 
 
@@ -650,10 +658,9 @@ def test_async_decode_fsm_rows_fall_back_to_sync():
         assert 0 <= parsed["v"] <= 9
 
 
-def test_moe_batched_property():
-    """Property: for random shapes/routings, the capacity-padded batched MoE
-    equals the exact loop whenever capacity covers all tokens, and never
-    produces NaN/inf even when tokens overflow into the trash slot."""
+def test_moe_grouped_property():
+    """Property: for random shapes and routings the grouped padded-segment
+    execution equals the exact loop — ALWAYS (dropless), finite outputs."""
     from hypothesis import given, settings, strategies as st
 
     from sutro_amd.models.qwen3 import Qwen3MoE
@@ -675,12 +682,8 @@ def test_moe_batched_property():
             torch.nn.init.normal_(p, std=0.1)
         x = torch.randn(T, 16)
         ref = moe._forward_loop(x)
-        # ample capacity: exact equality with the loop
-        got = moe._forward_batched(x, capacity_factor=float(E))
+        got = moe._forward_grouped(x)
         torch.testing.assert_close(got, ref, atol=1e-4, rtol=1e-4)
-        # tight capacity: overflow tokens drop to the trash slot — output
-        # must stay finite and drop-free tokens must still match
-        tight = moe._forward_batched(x, capacity_factor=0.25)
-        assert torch.isfinite(tight).all()
+        assert torch.isfinite(got).all()
 
     run()

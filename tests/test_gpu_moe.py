@@ -1,0 +1,128 @@
+"""GPU tests for the dropless grouped-GEMM MoE kernel
+(csrc/grouped_gemm.hip) vs the torch reference layout semantics and the
+exact per-expert loop."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from sutro_amd import ops
+    from sutro_amd.ops import torch_ref
+
+DEV = "cuda"
+BM = 64
+
+
+def require_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+
+
+def _segments(counts):
+    counts_t = torch.tensor(counts, dtype=torch.int32, device=DEV)
+    padded = (counts_t + BM - 1) // BM * BM
+    pad_off = torch.zeros(len(counts) + 1, dtype=torch.int32, device=DEV)
+    pad_off[1:] = torch.cumsum(padded, 0)
+    return counts_t, pad_off // BM, int(pad_off[-1]) // BM
+
+
+def test_grouped_gemm_plain_vs_ref():
+    require_gpu()
+    torch.manual_seed(0)
+    E, K, N = 5, 128, 192
+    counts = [7, 0, 130, 64, 1]  # empty expert + multi-tile + exact-tile
+    counts_t, tile_off, max_tiles = _segments(counts)
+    rows_max = int(tile_off[-1]) * BM + BM
+    a = torch.randn(rows_max, K).to(torch.bfloat16).to(DEV)
+    w = (torch.randn(E, N, K) * 0.2).to(torch.bfloat16).to(DEV)
+    out = torch.zeros(rows_max, N, dtype=torch.bfloat16, device=DEV)
+    ops.grouped_gemm(out, a, w, None, tile_off, counts_t, max_tiles, False)
+    ref = torch.zeros_like(out)
+    torch_ref.grouped_gemm(ref, a, w, None, tile_off.cpu(), counts_t.cpu(),
+                           max_tiles, False)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * BM
+        torch.testing.assert_close(out[s0:s0 + c].float(),
+                                   ref[s0:s0 + c].float(),
+                                   atol=5e-2, rtol=5e-2)
+
+
+def test_grouped_gemm_gather_silu_vs_ref():
+    require_gpu()
+    torch.manual_seed(1)
+    E, K, m = 4, 192, 128
+    T = 100
+    counts = [65, 3, 0, 90]
+    counts_t, tile_off, max_tiles = _segments(counts)
+    rows_max = int(tile_off[-1]) * BM + BM
+    x = torch.randn(T, K).to(torch.bfloat16).to(DEV)
+    w = (torch.randn(E, 2 * m, K) * 0.2).to(torch.bfloat16).to(DEV)
+    rng = np.random.default_rng(5)
+    row_tok = torch.full((rows_max,), -1, dtype=torch.int32, device=DEV)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * BM
+        row_tok[s0:s0 + c] = torch.from_numpy(
+            rng.integers(0, T, size=c).astype(np.int32)).to(DEV)
+    out = torch.zeros(rows_max, m, dtype=torch.bfloat16, device=DEV)
+    ops.grouped_gemm(out, x, w, row_tok, tile_off, counts_t, max_tiles, True)
+    ref = torch.zeros_like(out)
+    torch_ref.grouped_gemm(ref, x, w, row_tok.cpu(), tile_off.cpu(),
+                           counts_t.cpu(), max_tiles, True)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * BM
+        torch.testing.assert_close(out[s0:s0 + c].float(),
+                                   ref[s0:s0 + c].float(),
+                                   atol=5e-2, rtol=5e-2)
+
+
+def test_moe_grouped_gpu_matches_loop_skewed():
+    """Full MoE layer on GPU: dropless grouped kernel path == exact loop,
+    under adversarial skew (all tokens to one expert)."""
+    require_gpu()
+    from sutro_amd.models.qwen3 import Qwen3MoE
+    from sutro_amd.models.registry import ModelSpec
+
+    spec = ModelSpec(name="tiny-moe-gpu", hidden_size=256, num_layers=1,
+                     num_heads=4, num_kv_heads=2, head_dim=64,
+                     intermediate_size=0, vocab_size=512, num_experts=8,
+                     experts_per_token=2, moe_intermediate_size=256)
+    torch.manual_seed(2)
+    moe = Qwen3MoE(spec, torch.bfloat16).to(DEV)
+    for p in moe.parameters():
+        torch.nn.init.normal_(p, std=0.05)
+    x = (torch.randn(333, 256) * 0.5).to(torch.bfloat16).to(DEV)
+    ref = moe._forward_loop(x).float()
+    got = moe._forward_grouped(x).float()
+    torch.testing.assert_close(got, ref, atol=3e-2, rtol=3e-2)
+    with torch.no_grad():
+        moe.router.weight[3] += 50.0  # every token -> expert 3 (skew)
+    ref = moe._forward_loop(x).float()
+    got = moe._forward_grouped(x).float()
+    torch.testing.assert_close(got, ref, atol=3e-2, rtol=3e-2)
+
+
+def test_moe_model_generates_gpu():
+    """qwen-3-30b-a3b-shaped tiny MoE model decodes end-to-end on the GPU
+    grouped path (through the engine, hipGraph decode included)."""
+    require_gpu()
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import ModelSpec
+
+    spec = ModelSpec(name="tiny-moe-e2e", hidden_size=256, num_layers=2,
+                     num_heads=4, num_kv_heads=2, head_dim=64,
+                     intermediate_size=0, vocab_size=2048, max_context=512,
+                     tie_embeddings=True, num_experts=4, experts_per_token=2,
+                     moe_intermediate_size=192)
+    cfg = EngineConfig(spec=spec, device=DEV, max_model_len=256,
+                       num_kv_blocks=128, max_tokens_per_step=512)
+    eng = LLMEngine(cfg)
+    outs = eng.generate(["moe gpu row one", "moe gpu row two"],
+                        sampling=SamplingParams(max_tokens=12,
+                                                temperature=0.7))
+    assert len(outs) == 2
+    assert all(isinstance(o, str) for o in outs)

@@ -71,3 +71,62 @@ def test_load_split_projections(tmp_path):
     for (na, pa), (nb, pb) in zip(sorted(src.named_parameters()),
                                   sorted(dst.named_parameters())):
         torch.testing.assert_close(pa, pb, msg=na)
+
+
+def _mk_moe(seed=3, tp=None, ep=False):
+    from sutro_amd.models.registry import ModelSpec
+
+    spec = ModelSpec(name="tiny-moe-load", hidden_size=64, num_layers=2,
+                     num_heads=4, num_kv_heads=2, head_dim=16,
+                     intermediate_size=0, vocab_size=512, max_context=512,
+                     tie_embeddings=True, num_experts=4, experts_per_token=2,
+                     moe_intermediate_size=64)
+    m = Qwen3Model(spec, torch.float32, 128, tp, moe_ep=ep)
+    m.init_random_weights(seed)
+    return m
+
+
+def test_load_per_expert_hf_names(tmp_path):
+    """HF MoE checkpoints ship mlp.experts.N.{gate,up,down}_proj tensors;
+    the [E, N, K] parameter layout stacks them directly."""
+    from safetensors.torch import save_file
+
+    src = _mk_moe(seed=11)
+    state = {}
+    for name, p in src.named_parameters():
+        t = p.detach().clone()
+        if name.endswith(".mlp.gate_up"):
+            prefix = "model." + name[: -len("gate_up")]
+            m = t.shape[1] // 2
+            for e in range(t.shape[0]):
+                state[f"{prefix}experts.{e}.gate_proj.weight"] = t[e, :m]
+                state[f"{prefix}experts.{e}.up_proj.weight"] = t[e, m:]
+        elif name.endswith(".mlp.down"):
+            prefix = "model." + name[: -len("down")]
+            for e in range(t.shape[0]):
+                state[f"{prefix}experts.{e}.down_proj.weight"] = t[e]
+        else:
+            state["model." + name] = t
+    save_file(state, str(tmp_path / "model.safetensors"))
+    dst = _mk_moe(seed=0)
+    load_weights(dst, str(tmp_path))
+    for (na, pa), (nb, pb) in zip(sorted(src.named_parameters()),
+                                  sorted(dst.named_parameters())):
+        assert na == nb
+        torch.testing.assert_close(pa, pb)
+
+
+def test_load_strict_rejects_unknown_tensors(tmp_path):
+    from safetensors.torch import save_file
+
+    import pytest
+
+    src = _mk(seed=2)
+    state = {"model." + n: p.detach().clone()
+             for n, p in src.named_parameters()}
+    state["model.layers.0.totally_unknown.weight"] = torch.zeros(3, 3)
+    save_file(state, str(tmp_path / "model.safetensors"))
+    dst = _mk(seed=0)
+    with pytest.raises(ValueError, match="map to no parameter"):
+        load_weights(dst, str(tmp_path))
+    load_weights(dst, str(tmp_path), strict=False)  # non-strict: ignored
