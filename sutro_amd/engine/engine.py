@@ -188,7 +188,10 @@ class LLMEngine:
         truncate: bool = True,
     ) -> Request:
         sampling = sampling or SamplingParams(max_tokens=self.cfg.default_max_new_tokens)
-        limit = self.cfg.max_model_len - max(1, min(sampling.max_tokens, 64))
+        # truncation reserves the FULL requested generation budget (capped so
+        # a huge max_tokens still leaves a 16-token prompt window) — a 64-cap
+        # here made near-limit rows finish early with LENGTH
+        limit = max(16, self.cfg.max_model_len - max(1, sampling.max_tokens))
         if len(prompt_token_ids) > limit:
             if truncate:
                 prompt_token_ids = prompt_token_ids[:limit]

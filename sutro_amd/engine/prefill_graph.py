@@ -44,6 +44,8 @@ class PrefillGraphRunner:
         self.tile_max = self.s_max + self.t_pad // 32
         # replay only when the wave is big enough that padding waste is small
         self.min_tokens = cfg.graph_prefill_min_tokens
+        self._h2d_done = (torch.cuda.Event()
+                          if torch.cuda.is_available() else None)
         dev = cfg.device
         T, S, W, TL = self.t_pad, self.s_max, self.bt_width, self.tile_max
         self.d_ids = torch.zeros(T, dtype=torch.long, device=dev)
@@ -166,7 +168,14 @@ class PrefillGraphRunner:
     @torch.no_grad()
     def run(self, sb: ScheduledBatch) -> torch.Tensor:
         """Replay the padded prefill pass; returns hidden[:real_tokens]."""
+        # guard: the previous replay's non_blocking H2D copies must have
+        # consumed the pinned host staging before we overwrite it (mirrors
+        # DecodeGraphRunner._h2d_done; ADVICE.md r1 item 1)
+        if self._h2d_done is not None:
+            self._h2d_done.synchronize()
         real_t = self._fill_host(sb)
         self._h2d()
+        if self._h2d_done is not None:
+            self._h2d_done.record()
         self.graph.replay()
         return self.hidden_out[:real_t]
