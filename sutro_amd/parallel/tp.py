@@ -5,6 +5,16 @@ all-reduce per block half (after o_proj and after down_proj), none elsewhere.
 Column-parallel layers shard the output dim; row-parallel layers shard the
 input dim and all-reduce the partial sums.
 
+Message-size-aware reduction (SURVEY.md §5 xGMI note):
+- decode steps (small, latency-bound activations) issue ONE all-reduce —
+  RCCL's low-latency algorithms handle the small-message regime, and the
+  hipGraph-captured decode path stays a single enqueued op;
+- prefill waves (large activations) run CHUNK-PIPELINED in
+  RowParallelLinear: the GEMM is computed in token chunks and each chunk's
+  all-reduce is issued async (RCCL executes on its own HIP stream) while the
+  next chunk's GEMM occupies the MFMA units — the exposure left is one
+  chunk's reduction instead of the whole wave's.
+
 All TP ranks of a group run the engine in lockstep: after the final
 all-reduce every rank holds identical hidden states, so logits, sampling
 streams and scheduler decisions stay rank-identical with no extra
@@ -13,12 +23,18 @@ synchronization (see LLMEngine docstring).
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
 import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
+
+# tokens at/above which a row-parallel matmul pipelines GEMM chunks with
+# async all-reduces; below it (decode) a single collective is issued
+PIPELINE_MIN_TOKENS = int(os.environ.get("SUTRO_TP_PIPELINE_MIN", "4096"))
+PIPELINE_CHUNKS = int(os.environ.get("SUTRO_TP_PIPELINE_CHUNKS", "4"))
 
 
 class TPContext:
@@ -88,5 +104,26 @@ class RowParallelLinear(nn.Module):
             torch.empty(out_features, self.in_per_rank, dtype=dtype))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        tp = self.tp
+        T = x.shape[0]
+        if (tp.enabled and x.dim() == 2 and T >= PIPELINE_MIN_TOKENS
+                and not (x.is_cuda
+                         and torch.cuda.is_current_stream_capturing())):
+            # prefill pipelining: chunk i's all-reduce (async, RCCL stream)
+            # overlaps chunk i+1's GEMM on the compute stream
+            n = min(PIPELINE_CHUNKS, T)
+            y = x.new_empty(T, self.weight.shape[0])
+            bounds = [(T * i) // n for i in range(n + 1)]
+            works = []
+            wt = self.weight.detach().t()  # inference-only framework: the
+            # out= mm below rejects autograd-tracked operands
+            for i in range(n):
+                a, b = bounds[i], bounds[i + 1]
+                torch.mm(x[a:b], wt, out=y[a:b])
+                works.append(dist.all_reduce(y[a:b], group=tp.group,
+                                             async_op=True))
+            for wk in works:
+                wk.wait()
+            return y
         y = F.linear(x, self.weight)
-        return self.tp.all_reduce(y)
+        return tp.all_reduce(y)
