@@ -166,8 +166,19 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
                             "out_tokens": len(req.output_token_ids),
                         }
                     out_q.put(("row_done", rank, job_id, (row_idx, payload)))
+            elif tp > 1:
+                # lockstep groups must keep polling the broadcast channel
+                time.sleep(0.02)
             else:
-                time.sleep(0.02 if tp > 1 else 0.05)
+                # idle single-rank worker: BLOCK on the inbox instead of
+                # sleep-polling (p0 interactive latency ate up to 50 ms of
+                # poll jitter per hop — VERDICT r1 weak item 9)
+                try:
+                    m = in_q.get(timeout=0.5)
+                except queue_mod.Empty:
+                    continue
+                if not handle(m):
+                    running = False
     except Exception as e:  # pragma: no cover
         out_q.put(("worker_error", rank, None, f"{type(e).__name__}: {e}"))
 

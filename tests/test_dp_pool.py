@@ -244,3 +244,41 @@ def test_restart_resume_embedding_job(sutro_home, monkeypatch):
         assert len(res["outputs"]) == 4
     finally:
         svc2.shutdown()
+
+
+def test_dp8_soak_sharded_and_ordered(sutro_home, monkeypatch):
+    """SCALE-readiness soak: 8 engine worker processes (the full-node DP
+    shape the driver benches sight-unseen), 64 rows, ordered results,
+    per-row seeds deterministic."""
+    monkeypatch.setenv("SUTRO_AMD_NUM_WORKERS", "8")
+    from sutro_amd.service.jobs import JobService
+
+    svc = JobService(home=sutro_home, device="cpu",
+                     engine_kwargs={"num_kv_blocks": 64,
+                                    "max_model_len": 256})
+    try:
+        rows = [f"soak row {i} " + "x" * (i % 17) for i in range(64)]
+        out = svc.submit_job({
+            "model": "qwen-3.5-2b", "inputs": rows,
+            "random_seed_per_input": True,
+            "sampling_params": {"max_tokens": 5, "temperature": 0.9},
+        })
+        job_id = out["results"]
+        assert _await(svc, job_id, timeout=600) == "SUCCEEDED"
+        assert len(svc.workers["qwen-3.5-2b"].procs) == 8
+        res = svc.job_results(job_id, include_inputs=True)["results"]
+        assert len(res["outputs"]) == 64
+        assert all(o is not None for o in res["outputs"])
+        assert res["inputs"] == rows  # input-ordered merge across 8 shards
+        # determinism: resubmit, same seeded outputs
+        out2 = svc.submit_job({
+            "model": "qwen-3.5-2b", "inputs": rows,
+            "random_seed_per_input": True,
+            "sampling_params": {"max_tokens": 5, "temperature": 0.9},
+        })
+        jid2 = out2["results"]
+        assert _await(svc, jid2, timeout=600) == "SUCCEEDED"
+        res2 = svc.job_results(jid2)["results"]
+        assert res2["outputs"] == res["outputs"]
+    finally:
+        svc.shutdown()
