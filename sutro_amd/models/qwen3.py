@@ -12,6 +12,7 @@ import math
 from typing import Optional
 
 import torch
+import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
 
@@ -201,6 +202,81 @@ class Qwen3MoE(nn.Module):
         w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
         out.index_add_(0, sorted_tok, out_sorted.float() * w_sorted)
         return out.to(x.dtype)
+
+    def _run_local_experts(self, rows: torch.Tensor,
+                           e_loc: torch.Tensor) -> torch.Tensor:
+        """rows[i] through LOCAL expert e_loc[i] (0..experts_per_rank-1);
+        returns [n, h]. Segment loop (the a2a path already host-synced for
+        split sizes, so dynamic per-expert GEMMs are free here)."""
+        n, h = rows.shape
+        order = torch.argsort(e_loc, stable=True)
+        sorted_rows = rows[order]
+        sorted_e = e_loc[order]
+        out_sorted = torch.empty_like(sorted_rows)
+        counts = torch.bincount(sorted_e, minlength=self.experts_per_rank)
+        start = 0
+        for e, c in enumerate(counts.tolist()):
+            if c == 0:
+                continue
+            seg = sorted_rows[start:start + c]
+            act = ops.silu_mul(seg @ self.gate_up[e].T)
+            out_sorted[start:start + c] = act @ self.down[e].T
+            start += c
+        out = torch.empty_like(rows)
+        out[order] = out_sorted
+        return out
+
+    def forward_a2a(self, x_local: torch.Tensor) -> torch.Tensor:
+        """EP with ALL-TO-ALL dispatch over the xGMI mesh: each rank holds a
+        DIFFERENT token shard (sequence/token-sharded group, NOT the engine's
+        lockstep TP groups) and E/size experts (ep=True layout). Tokens are
+        routed, exchanged to their experts' owner ranks (all_to_all_single),
+        computed locally, and returned by the reverse exchange.
+
+        Exact (dropless); split sizes are exchanged first (one tiny a2a +
+        host sync — inherent to dynamic routing). Lockstep TP groups keep
+        the all-reduce combine in forward(): every rank already holds every
+        token there, so a2a would move rows for no reduction in traffic.
+        """
+        assert self.ep, "forward_a2a needs expert-sharded (ep=True) layout"
+        tp = self.tp
+        world, rank = tp.size, tp.rank
+        e_l = self.experts_per_rank
+        T, h = x_local.shape
+        k = self.top_k
+        weights, idx = torch_ref.topk_softmax_router(self.router(x_local), k)
+        flat_e = idx.reshape(-1)
+        flat_tok = torch.arange(T, device=x_local.device).repeat_interleave(k)
+        dest = torch.div(flat_e, e_l, rounding_mode="floor")
+        order = torch.argsort(dest, stable=True)
+        send_rows = x_local[flat_tok[order]].contiguous()
+        send_eloc = (flat_e - dest * e_l)[order].contiguous()
+        send_counts = torch.bincount(dest, minlength=world)
+
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=tp.group)
+        s_splits = send_counts.tolist()
+        r_splits = recv_counts.tolist()
+        n_recv = sum(r_splits)
+
+        recv_rows = x_local.new_empty(n_recv, h)
+        dist.all_to_all_single(recv_rows, send_rows, output_split_sizes=r_splits,
+                               input_split_sizes=s_splits, group=tp.group)
+        recv_eloc = send_eloc.new_empty(n_recv)
+        dist.all_to_all_single(recv_eloc, send_eloc, output_split_sizes=r_splits,
+                               input_split_sizes=s_splits, group=tp.group)
+
+        computed = self._run_local_experts(recv_rows, recv_eloc)
+
+        back_rows = x_local.new_empty(T * k, h)
+        dist.all_to_all_single(back_rows, computed.contiguous(),
+                               output_split_sizes=s_splits,
+                               input_split_sizes=r_splits, group=tp.group)
+        # back_rows is in `order`; combine per token with routing weights
+        out = torch.zeros(T, h, dtype=torch.float32, device=x_local.device)
+        w_sorted = weights.reshape(-1)[order].unsqueeze(-1)
+        out.index_add_(0, flat_tok[order], back_rows.float() * w_sorted)
+        return out.to(x_local.dtype)
 
     _GG_BM = 64  # grouped-gemm tile height (csrc/grouped_gemm.hip GG_BM)
 

@@ -256,3 +256,64 @@ def test_tp_pipelined_all_reduce_matches_dense():
         p.join(timeout=60)
     assert kind == "ok", y
     torch.testing.assert_close(y, ref, atol=1e-5, rtol=1e-5)
+
+
+def _run_a2a_worker(rank, world, port, result_q):
+    import os
+
+    import torch.distributed as dist
+
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        from sutro_amd.models.qwen3 import Qwen3MoE
+        from sutro_amd.models.registry import ModelSpec
+        from sutro_amd.parallel.tp import TPContext
+
+        spec = ModelSpec(name="tiny-moe-a2a", hidden_size=32, num_layers=1,
+                         num_heads=2, num_kv_heads=1, head_dim=16,
+                         intermediate_size=0, vocab_size=128, num_experts=4,
+                         experts_per_token=2, moe_intermediate_size=32)
+        tp = TPContext(size=world, rank=rank, group=None)
+        torch.manual_seed(0)  # identical router/experts on every rank...
+        full = Qwen3MoE(spec, torch.float32)  # ...single-rank reference
+        for p in full.parameters():
+            torch.nn.init.normal_(p, std=0.1)
+        shard = Qwen3MoE(spec, torch.float32, tp, ep=True)
+        with torch.no_grad():
+            shard.router.weight.copy_(full.router.weight)
+            e_l = spec.num_experts // world
+            shard.gate_up.copy_(full.gate_up[rank * e_l:(rank + 1) * e_l])
+            shard.down.copy_(full.down[rank * e_l:(rank + 1) * e_l])
+
+        torch.manual_seed(7)
+        x_all = torch.randn(26, 32)  # uneven shards: 13 tokens each
+        x_local = x_all[rank * 13:(rank + 1) * 13]
+        with torch.no_grad():
+            y_local = shard.forward_a2a(x_local)
+            y_ref = full._forward_loop(x_all)[rank * 13:(rank + 1) * 13]
+        result_q.put(("ok", rank, y_local.detach(), y_ref.detach()))
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        result_q.put(("err", rank, repr(e), None))
+
+
+def test_moe_all_to_all_dispatch_matches_single_rank():
+    """EP all-to-all: token-sharded ranks exchange assignments to expert
+    owners and back; every rank's output equals the single-module result on
+    the concatenated tokens (gloo, world 2)."""
+    import torch.multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_run_a2a_worker, args=(r, 2, 29517, q))
+          for r in range(2)]
+    for p in ps:
+        p.start()
+    got = [q.get(timeout=180) for _ in range(2)]
+    for p in ps:
+        p.join(timeout=60)
+    for kind, rank, y, ref in got:
+        assert kind == "ok", y
+        torch.testing.assert_close(y, ref, atol=1e-5, rtol=1e-5)
