@@ -16,12 +16,19 @@
 // across replays — preemption restarts and async/sync A-B runs regenerate
 // identical tokens. The value thresholds (top-k count, top-p mass) resolve
 // by a 4-level radix descent over the monotonic u32 transform of s_i
-// (256-bucket count+mass histograms per level; boundary bucket recursed),
-// touching the logits row ~6x total:
-//   A: max/argmax   B: Z + level-0 histogram   R1-R3: refinements
-//   F: chunked index-order scan (wave chunk -> lane subchunk -> element)
-// At n=2048 rows x 151,936 bf16 logits that is ~6 x 622 MB ~ 450 us/step at
-// HBM3E rate, vs multi-ms for a torch sort-based path.
+// (256-bucket count+mass histograms per level; boundary bucket recursed).
+//
+// Throughput notes (r2 rocprof: first version ran 2.33 ms per 2048x151,936
+// bf16 call ~ 5x the pass-count HBM bound):
+//  - all full-vocab passes consume u16x8 (16 B) per lane per step — the
+//    scalar-u16 version dispatched 8x the instructions for the same bytes;
+//  - the level-0 histogram is 8-way lane-replicated (hist[lane&7][bucket]):
+//    scaled logits cluster in a handful of exponent buckets, so per-element
+//    atomics serialized on a few LDS addresses; replication cuts that 8x
+//    and the replicas are reduced once at scan time (integer = order-free);
+//  - the final index-order selection streams the crossing chunk with a
+//    whole-wave u64 scan (ballot early-exit) instead of a single-lane walk.
+//   Passes: A max/argmax | B Z + level-0 hist | R1-R3 refine | F select.
 
 #include <hip/hip_runtime.h>
 
@@ -29,6 +36,7 @@
 
 #define SMP_THREADS 256
 #define SMP_WAVES (SMP_THREADS / 64)
+#define HREP 8  // level-0 histogram replicas
 
 // monotonic u32 key: key(a) < key(b)  <=>  a < b (floats, no NaN)
 __device__ __forceinline__ u32 f32key(float s) {
@@ -54,8 +62,10 @@ struct SmpShared {
   int argmax;
   float zf;       // float Z (logprob)
   u64 zfix;       // fixed-point Z (thresholds)
-  u32 hist_cnt[2][256];
-  u64 hist_mass[2][256];
+  // level 0 uses all HREP replicas (reduced into replica 0 at scan time);
+  // refinement levels use replicas 0 (count descent) and 1 (mass descent)
+  u32 hist_cnt[HREP][256];
+  u64 hist_mass[HREP][256];
   Boundary bk, bp;
   u32 tau;        // combined threshold key
   u64 kept_mass;  // M
@@ -72,8 +82,33 @@ __device__ __forceinline__ float load_logit(const void* row, int i) {
   return ((const float*)row)[i];
 }
 
+// 8 consecutive logits starting at i0 (16B/32B aligned vector loads)
+template <bool BF16>
+__device__ __forceinline__ void load_logit8(const void* row, int i0,
+                                            float* out) {
+  if (BF16) {
+    const u16x8 v = *(const u16x8*)((const u16*)row + i0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = bf2f(v[j]);
+  } else {
+    const f32x4 a = *(const f32x4*)((const float*)row + i0);
+    const f32x4 b = *(const f32x4*)((const float*)row + i0 + 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      out[j] = a[j];
+      out[4 + j] = b[j];
+    }
+  }
+}
+
 __device__ __forceinline__ bool alive_bit(const u32* mrow, int i) {
   return mrow == nullptr || ((mrow[i >> 5] >> (i & 31)) & 1u);
+}
+
+// mask bits for 8 consecutive tokens at i0 (i0 % 8 == 0: never crosses a word)
+__device__ __forceinline__ u32 alive_bits8(const u32* mrow, int i0) {
+  if (mrow == nullptr) return 0xFFu;
+  return (mrow[i0 >> 5] >> (i0 & 31)) & 0xFFu;
 }
 
 // deterministic block max+argmax (lowest index wins ties)
@@ -150,18 +185,31 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
   const float T = temps[row];
   const bool greedy = T < 1e-5f;
   const float inv_t = greedy ? 1.0f : 1.0f / T;
+  const int vl8 = vl & ~7;
 
-  // ---- pass A: max + argmax over alive ----
+  // ---- pass A: max + argmax over alive (vectorized) ----
   float lmax = -INFINITY;
   int larg = -1;
-  for (int i = tid; i < vl; i += SMP_THREADS) {
-    if (!alive_bit(mrow, i)) continue;
-    float v = load_logit<BF16>(lrow, i);
-    if (v > lmax) {
-      lmax = v;
-      larg = i;
-    }
+  for (int i0 = tid * 8; i0 < vl8; i0 += SMP_THREADS * 8) {
+    const u32 ab = alive_bits8(mrow, i0);
+    if (ab == 0) continue;
+    float v[8];
+    load_logit8<BF16>(lrow, i0, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      if (((ab >> j) & 1u) && v[j] > lmax) {
+        lmax = v[j];
+        larg = i0 + j;
+      }
   }
+  for (int i = vl8 + tid; i < vl; i += SMP_THREADS)
+    if (alive_bit(mrow, i)) {
+      float v = load_logit<BF16>(lrow, i);
+      if (v > lmax) {
+        lmax = v;
+        larg = i;
+      }
+    }
   block_argmax(&sm, lmax, larg, tid);
   const float m = (sm.argmax >= 0) ? sm.m : 0.0f;
   if (sm.argmax < 0) {  // no alive token (should not happen): emit 0
@@ -172,25 +220,47 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
     return;
   }
 
-  // ---- pass B: Z (float + fixed) and level-0 histograms ----
-  for (int i = tid; i < 256; i += SMP_THREADS) {
-    sm.hist_cnt[0][i] = 0;
-    sm.hist_mass[0][i] = 0;
+  // ---- pass B: Z (float + fixed) and 8-way replicated level-0 hist ----
+  for (int i = tid; i < HREP * 256; i += SMP_THREADS) {
+    ((u32*)sm.hist_cnt)[i] = 0;
+    ((u64*)sm.hist_mass)[i] = 0;
   }
   __syncthreads();
+  const int rep = tid & (HREP - 1);
   float zf_part = 0.0f;
   u64 zx_part = 0;
-  for (int i = tid; i < vl; i += SMP_THREADS) {
+  for (int i0 = tid * 8; i0 < vl8; i0 += SMP_THREADS * 8) {
+    const u32 ab = alive_bits8(mrow, i0);
+    if (ab == 0) continue;
+    float v[8];
+    load_logit8<BF16>(lrow, i0, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (!((ab >> j) & 1u)) continue;
+      const float s = (v[j] - m) * inv_t;
+      const float p = __expf(s);
+      zf_part += p;
+      if (!greedy) {
+        const u64 q = (u64)(p * 4294967296.0f);  // exact exponent shift
+        zx_part += q;
+        const u32 bu = f32key(s) >> 24;
+        atomicAdd(&sm.hist_cnt[rep][bu], 1u);
+        atomicAdd((unsigned long long*)&sm.hist_mass[rep][bu],
+                  (unsigned long long)q);
+      }
+    }
+  }
+  for (int i = vl8 + tid; i < vl; i += SMP_THREADS) {
     if (!alive_bit(mrow, i)) continue;
-    float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
-    float p = __expf(s);
+    const float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
+    const float p = __expf(s);
     zf_part += p;
     if (!greedy) {
-      u64 q = (u64)(p * 4294967296.0f);  // exact: *2^32 is an exponent shift
+      const u64 q = (u64)(p * 4294967296.0f);
       zx_part += q;
-      u32 key = f32key(s);
-      atomicAdd(&sm.hist_cnt[0][key >> 24], 1u);
-      atomicAdd((unsigned long long*)&sm.hist_mass[0][key >> 24],
+      const u32 bu = f32key(s) >> 24;
+      atomicAdd(&sm.hist_cnt[rep][bu], 1u);
+      atomicAdd((unsigned long long*)&sm.hist_mass[rep][bu],
                 (unsigned long long)q);
     }
   }
@@ -205,6 +275,20 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
     sm.wfix[tid >> 6] = zx_part;
   }
   __syncthreads();
+  // reduce histogram replicas into replica 0 (integers: order-free)
+  if (!greedy) {
+    for (int b = tid; b < 256; b += SMP_THREADS) {
+      u32 c = 0;
+      u64 q = 0;
+#pragma unroll
+      for (int r = 0; r < HREP; ++r) {
+        c += sm.hist_cnt[r][b];
+        q += sm.hist_mass[r][b];
+      }
+      sm.hist_cnt[0][b] = c;
+      sm.hist_mass[0][b] = q;
+    }
+  }
   if (tid == 0) {
     float z = 0.0f;
     u64 zx = 0;
@@ -239,9 +323,6 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
   }
   __syncthreads();
 
-  const u32 p_thr_words = 0;  // (placeholder keeps layout honest)
-  (void)p_thr_words;
-
   for (int level = 1; level < 4 && !sm.done; ++level) {
     const int shift_prev = 32 - 8 * level;
     const int shift_cur = shift_prev - 8;
@@ -254,21 +335,46 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
     __syncthreads();
     const int ka = sm.bk.active, pa = sm.bp.active;
     const u32 kpref = sm.bk.pref, ppref = sm.bp.pref;
-    for (int i = tid; i < vl; i += SMP_THREADS) {
+    for (int i0 = tid * 8; i0 < vl8; i0 += SMP_THREADS * 8) {
+      const u32 ab = alive_bits8(mrow, i0);
+      if (ab == 0) continue;
+      float v[8];
+      load_logit8<BF16>(lrow, i0, v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (!((ab >> j) & 1u)) continue;
+        const float s = (v[j] - m) * inv_t;
+        const u32 key = f32key(s);
+        const u32 hp = key >> shift_prev;
+        if (hp != kpref && hp != ppref) continue;
+        const u64 q = (u64)(__expf(s) * 4294967296.0f);
+        const u32 bu = (key >> shift_cur) & 0xFFu;
+        if (ka && hp == kpref) {
+          atomicAdd(&sm.hist_cnt[0][bu], 1u);
+          atomicAdd((unsigned long long*)&sm.hist_mass[0][bu],
+                    (unsigned long long)q);
+        }
+        if (pa && hp == ppref) {
+          atomicAdd(&sm.hist_cnt[1][bu], 1u);
+          atomicAdd((unsigned long long*)&sm.hist_mass[1][bu],
+                    (unsigned long long)q);
+        }
+      }
+    }
+    for (int i = vl8 + tid; i < vl; i += SMP_THREADS) {
       if (!alive_bit(mrow, i)) continue;
-      float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
-      u32 key = f32key(s);
-      if ((key >> shift_prev) != kpref && (key >> shift_prev) != ppref)
-        continue;
-      float p = __expf(s);
-      u64 q = (u64)(p * 4294967296.0f);
-      u32 bu = (key >> shift_cur) & 0xFFu;
-      if (ka && (key >> shift_prev) == kpref) {
+      const float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
+      const u32 key = f32key(s);
+      const u32 hp = key >> shift_prev;
+      if (hp != kpref && hp != ppref) continue;
+      const u64 q = (u64)(__expf(s) * 4294967296.0f);
+      const u32 bu = (key >> shift_cur) & 0xFFu;
+      if (ka && hp == kpref) {
         atomicAdd(&sm.hist_cnt[0][bu], 1u);
         atomicAdd((unsigned long long*)&sm.hist_mass[0][bu],
                   (unsigned long long)q);
       }
-      if (pa && (key >> shift_prev) == ppref) {
+      if (pa && hp == ppref) {
         atomicAdd(&sm.hist_cnt[1][bu], 1u);
         atomicAdd((unsigned long long*)&sm.hist_mass[1][bu],
                   (unsigned long long)q);
@@ -308,18 +414,39 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
   const u64 target = sm.target;
 
   // ---- pass F: index-order inverse CDF, hierarchical first-crossing ----
-  // phase 1: per-wave contiguous chunks, lane-strided (coalesced)
-  const int chunk = (vl + SMP_WAVES - 1) / SMP_WAVES;
+  // phase 1: per-wave contiguous chunks, lane-strided vectorized sums
+  const int chunk = ((vl + SMP_WAVES * 8 - 1) / (SMP_WAVES * 8)) * 8;
   const int wv = tid >> 6, lane = tid & 63;
-  const int c0 = wv * chunk, c1 = min(vl, c0 + chunk);
+  const int c0 = min(vl, wv * chunk), c1 = min(vl, c0 + chunk);
   u64 wsum = 0;
   int wlast = -1;
-  for (int i = c0 + lane; i < c1; i += 64) {
-    if (!alive_bit(mrow, i)) continue;
-    float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
-    if (f32key(s) < tau) continue;
-    wsum += (u64)(__expf(s) * 4294967296.0f);
-    wlast = i;
+  for (int i0 = c0 + lane * 8; i0 + 8 <= c1; i0 += 64 * 8) {
+    const u32 ab = alive_bits8(mrow, i0);
+    if (ab == 0) continue;
+    float v[8];
+    load_logit8<BF16>(lrow, i0, v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (!((ab >> j) & 1u)) continue;
+      const float s = (v[j] - m) * inv_t;
+      if (f32key(s) < tau) continue;
+      wsum += (u64)(__expf(s) * 4294967296.0f);
+      wlast = i0 + j;
+    }
+  }
+  // scalar remainder: the vectorized loop covers every FULL 8-block of
+  // [c0, c1) (block b -> lane b%64, iteration b/64); at most 7 elements of
+  // the last (vl-tail) chunk remain
+  {
+    const int rem0 = c0 + ((c1 - c0) & ~7);
+    const int i = rem0 + lane;
+    if (i < c1 && alive_bit(mrow, i)) {
+      const float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
+      if (f32key(s) >= tau) {
+        wsum += (u64)(__expf(s) * 4294967296.0f);
+        wlast = max(wlast, i);
+      }
+    }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) {
@@ -353,46 +480,36 @@ sampler_kernel(const void* __restrict__ logits, const float* __restrict__ temps,
   __syncthreads();
 
   if (sm.token < 0 && wv == 0 && s_chunk >= 0) {
-    // phase 2: wave 0 re-scans the chosen chunk; lanes own contiguous
-    // subranges (cache-hot after phase 1)
-    const int p0 = s_chunk * chunk, p1 = min(vl, p0 + chunk);
-    const int sub = (p1 - p0 + 63) / 64;
-    const int a0 = p0 + lane * sub, a1 = min(p1, a0 + sub);
-    u64 lsum = 0;
-    for (int i = a0; i < a1; ++i) {
-      if (!alive_bit(mrow, i)) continue;
-      float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
-      if (f32key(s) < tau) continue;
-      lsum += (u64)(__expf(s) * 4294967296.0f);
-    }
-    // inclusive scan across lanes (deterministic integer)
-    u64 inc = lsum;
-    for (int off = 1; off < 64; off <<= 1) {
-      u64 t = __shfl_up(inc, off, 64);
-      if (lane >= off) inc += t;
-    }
+    // phase 2: wave 0 streams the chosen chunk with a whole-wave u64 scan
+    // (cache-hot after phase 1); ballot early-exit at the crossing
+    const int p0 = min(vl, s_chunk * chunk), p1 = min(vl, p0 + chunk);
     const u64 rem = target - s_base;
-    const u64 exc = inc - lsum;
-    const bool mine = (exc <= rem) && (rem < inc);
-    const u64 bal = __ballot(mine);
-    const int l_star = (bal == 0) ? -1 : __ffsll((unsigned long long)bal) - 1;
-    if (lane == l_star) {
-      // phase 3: walk my subrange to the exact element
-      u64 acc = exc;
-      int tok = -1;
-      for (int i = a0; i < a1; ++i) {
-        if (!alive_bit(mrow, i)) continue;
-        float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
-        if (f32key(s) < tau) continue;
-        acc += (u64)(__expf(s) * 4294967296.0f);
-        if (acc > rem) {
-          tok = i;
-          break;
-        }
+    u64 running = 0;
+    for (int j = p0; j < p1; j += 64) {
+      const int i = j + lane;
+      u64 q = 0;
+      if (i < p1 && alive_bit(mrow, i)) {
+        const float s = (load_logit<BF16>(lrow, i) - m) * inv_t;
+        if (f32key(s) >= tau) q = (u64)(__expf(s) * 4294967296.0f);
       }
-      sm.token = (tok >= 0) ? tok : s_lastk;
+      u64 inc = q;
+      for (int off = 1; off < 64; off <<= 1) {
+        const u64 t = __shfl_up(inc, off, 64);
+        if (lane >= off) inc += t;
+      }
+      const bool mine = q > 0 && (running + inc - q) <= rem
+                        && rem < (running + inc);
+      const u64 total = __shfl(inc, 63, 64);
+      const u64 found = __ballot(mine);
+      if (found) {  // wave-uniform: broadcast the winner's index and stop
+        const int src = __ffsll((unsigned long long)found) - 1;
+        const int tok = __shfl(mine ? i : -1, src, 64);
+        if (lane == 0) sm.token = tok;
+        break;
+      }
+      running += total;
     }
-    if (l_star < 0 && lane == 0) sm.token = s_lastk;
+    if (lane == 0 && sm.token < 0) sm.token = s_lastk;
   }
   __syncthreads();
   if (tid == 0) {
