@@ -158,3 +158,44 @@ def test_seeded_sampling_independent_of_batch_composition():
     ref = draw(0)
     for extra in (1, 3, 7):
         assert draw(extra) == ref
+
+
+def test_threshold_semantics_ties_all_kept():
+    """Value-threshold top-k/top-p: ALL ties at a boundary are kept (the
+    sort-free semantics the fused kernel implements — no sort-order
+    tie-breaking)."""
+    from sutro_amd.engine.sampler import sample_torch_reference
+
+    # four-way tie at the top; top_k=2 must keep all four tied tokens
+    logits = torch.tensor([[5.0, 5.0, 5.0, 5.0, 1.0, 0.0]]).repeat(512, 1)
+    n = logits.shape[0]
+    temps = torch.ones(n)
+    top_ps = torch.ones(n)
+    top_ks = torch.full((n,), 2, dtype=torch.int32)
+    g = torch.Generator().manual_seed(0)
+    u = torch.rand(n, generator=g)
+    toks, _ = sample_torch_reference(logits, temps, top_ps, top_ks, u, 6)
+    seen = set(toks.tolist())
+    assert seen <= {0, 1, 2, 3}          # never the non-tied tail
+    assert len(seen) == 4                 # every tied token reachable
+
+
+def test_threshold_semantics_top_p_mass_gt():
+    """keep_i iff mass{s_j > s_i} < top_p * Z: the boundary token itself is
+    kept even when the strictly-greater mass is just below the threshold."""
+    from sutro_amd.engine.sampler import sample_torch_reference
+
+    # probs ~ [0.6, 0.3, 0.1]: top_p=0.7 -> mass_gt(token1)=0.6 < 0.7 keeps
+    # token 1; mass_gt(token2)=0.9 >= 0.7 drops token 2
+    p = torch.tensor([0.6, 0.3, 0.1])
+    logits = torch.log(p).unsqueeze(0).repeat(2048, 1)
+    n = logits.shape[0]
+    temps = torch.ones(n)
+    top_ps = torch.full((n,), 0.7)
+    top_ks = torch.full((n,), 3, dtype=torch.int32)
+    g = torch.Generator().manual_seed(1)
+    u = torch.rand(n, generator=g)
+    toks, _ = sample_torch_reference(logits, temps, top_ps, top_ks, u, 3)
+    seen = set(toks.tolist())
+    assert 2 not in seen
+    assert seen == {0, 1}
