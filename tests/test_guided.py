@@ -142,3 +142,68 @@ def test_unique_items_permutation():
     assert d.matches(b'["z","x","y"]')
     assert not d.matches(b'["x","x","y"]')
     assert not d.matches(b'["x","y"]')
+
+
+def test_mixed_guided_and_free_rows_one_batch():
+    """Concurrent jobs with and without schemas share engine steps: the
+    packed-mask grouping path (engine._fsm_masks mixed case) must mask ONLY
+    the guided rows."""
+    import json
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import tiny_spec_for_tests
+
+    cfg = EngineConfig(spec=tiny_spec_for_tests(), device="cpu",
+                       max_model_len=512, num_kv_blocks=128,
+                       max_tokens_per_step=256)
+    eng = LLMEngine(cfg)
+    schema = {"type": "object", "properties": {
+        "label": {"enum": ["a", "b"]}, "n": {"type": "integer",
+                                             "minimum": 0, "maximum": 9}}}
+    fsm_id = eng.register_fsm(schema)
+    guided = [eng.add_request(eng.tokenizer.encode(f"guided {i}"),
+                              SamplingParams(max_tokens=64, temperature=0.9),
+                              fsm_id=fsm_id) for i in range(3)]
+    free = [eng.add_request(eng.tokenizer.encode(f"free {i}"),
+                            SamplingParams(max_tokens=6, temperature=0.9))
+            for i in range(3)]
+    while eng.has_work():
+        eng.step()
+    for r in guided:
+        obj = json.loads(eng.output_text(r))
+        assert obj["label"] in ("a", "b")
+        assert 0 <= obj["n"] <= 9
+    for r in free:
+        assert r.finished  # free rows unaffected by the guided rows' masks
+        assert len(r.output_token_ids) <= 6
+
+
+def test_two_different_schemas_same_batch():
+    """Two FSMs in one batch: per-FSM row grouping gathers the right masks."""
+    import json
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import tiny_spec_for_tests
+
+    cfg = EngineConfig(spec=tiny_spec_for_tests(), device="cpu",
+                       max_model_len=512, num_kv_blocks=128,
+                       max_tokens_per_step=256)
+    eng = LLMEngine(cfg)
+    f1 = eng.register_fsm({"enum": ["red", "green"]})
+    f2 = eng.register_fsm({"type": "integer", "minimum": 10, "maximum": 19})
+    r1 = [eng.add_request(eng.tokenizer.encode("x"),
+                          SamplingParams(max_tokens=32, temperature=1.0),
+                          fsm_id=f1) for _ in range(2)]
+    r2 = [eng.add_request(eng.tokenizer.encode("y"),
+                          SamplingParams(max_tokens=32, temperature=1.0),
+                          fsm_id=f2) for _ in range(2)]
+    while eng.has_work():
+        eng.step()
+    for r in r1:
+        assert json.loads(eng.output_text(r)) in ("red", "green")
+    for r in r2:
+        assert 10 <= int(eng.output_text(r)) <= 19
