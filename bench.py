@@ -107,16 +107,36 @@ def run_e2e(args, rank: int, world: int, local_rank: int, have_gpu: bool):
              for _ in range(rows)]
     sp = {"max_tokens": max_new, "temperature": 0.8, "top_p": 0.95}
 
+    schema = None
+    if args.schema:
+        schema = {"type": "object", "properties": {
+            "name": {"type": "string", "maxLength": 48},
+            "category": {"enum": ["news", "review", "spam", "other"]},
+            "sentiment": {"enum": ["positive", "negative", "neutral"]},
+            "score": {"type": "integer", "minimum": 0, "maximum": 100}}}
+
     # warmup job: engine init, hipGraph capture, first GEMM algo selection
     wid = client.infer(texts[:8], model=model, job_priority=1,
-                       stay_attached=False, sampling_params=sp)
+                       stay_attached=False, sampling_params=sp,
+                       output_schema=schema)
     client.await_job_completion(wid, obtain_results=False, timeout=600)
+
+    # p0 interactive latency: single row submit -> materialized result
+    # (admission bypasses the p1 accumulation hold)
+    t_p0 = time.time()
+    pid = client.infer([texts[0]], model=model, job_priority=0,
+                       stay_attached=False, output_schema=schema,
+                       sampling_params={**sp, "max_tokens": 16})
+    client.await_job_completion(pid, obtain_results=False, timeout=300)
+    client.get_job_results(pid)
+    p0_latency_ms = (time.time() - t_p0) * 1000.0
 
     if dist is not None:
         dist.barrier()
     t0 = time.time()
     job_id = client.infer(texts, model=model, job_priority=1,
-                          stay_attached=False, sampling_params=sp)
+                          stay_attached=False, sampling_params=sp,
+                          output_schema=schema)
     client.await_job_completion(job_id, obtain_results=False, timeout=3600)
     df = client.get_job_results(job_id)
     t1 = time.time()
@@ -151,6 +171,8 @@ def run_e2e(args, rank: int, world: int, local_rank: int, have_gpu: bool):
                 "prompt_len": args.prompt_len, "max_new_tokens": max_new,
                 "parallelism": f"dp{n_gpus}",
                 "elapsed_s": round(elapsed, 1),
+                "p0_row_latency_ms": round(p0_latency_ms, 1),
+                "guided": bool(schema),
             },
         }
         print(json.dumps(result))
