@@ -687,3 +687,44 @@ def test_moe_grouped_property():
         assert torch.isfinite(got).all()
 
     run()
+
+
+def test_model_vocab_larger_than_tokenizer():
+    """gemma/gpt-oss vocabs exceed the shipped tokenizer: the engine caps
+    the sampler at the tokenizer vocab and never emits dead-tail ids."""
+    from sutro_amd.models.registry import ModelSpec
+
+    spec = ModelSpec(name="tiny-bigvocab", hidden_size=64, num_layers=2,
+                     num_heads=4, num_kv_heads=2, head_dim=16,
+                     intermediate_size=128, vocab_size=200_000,
+                     max_context=512, tie_embeddings=True)
+    cfg = EngineConfig(spec=spec, device="cpu", max_model_len=256,
+                       num_kv_blocks=64, max_tokens_per_step=128)
+    eng = LLMEngine(cfg)
+    assert eng.tokenizer.vocab_size == 151_936  # capped at the full BPE
+    assert eng.sampler.vocab_limit == 151_936
+    r = eng.add_request(eng.tokenizer.encode("big vocab row"),
+                        SamplingParams(max_tokens=6, temperature=1.0))
+    while eng.has_work():
+        eng.step()
+    assert all(t < 151_936 for t in r.output_token_ids)
+
+
+def test_sampler_all_masked_row_falls_back():
+    """A fully-masked row (no alive token) must not crash or emit junk ids
+    beyond the vocab — it falls back to the greedy choice."""
+    import torch
+
+    from sutro_amd.engine.request import Request, SamplingParams as SP
+    from sutro_amd.engine.sampler import Sampler
+
+    s = Sampler("cpu", vocab_limit=16)
+    logits = torch.randn(2, 16)
+    mask = torch.zeros(2, 16, dtype=torch.bool)
+    mask[1, 3] = True  # row 0 fully masked, row 1 single-token
+    reqs = [Request(req_id=i, prompt_token_ids=[3],
+                    sampling=SP(max_tokens=4, temperature=1.0))
+            for i in range(2)]
+    toks, lps = s.sample(logits, reqs, fsm_mask=mask)
+    assert 0 <= toks[0] < 16
+    assert toks[1] == 3
