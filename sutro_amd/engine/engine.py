@@ -123,6 +123,18 @@ class LLMEngine:
         self.sampler = Sampler(cfg.device, seed=cfg.seed,
                                vocab_limit=self.tokenizer.vocab_size)
 
+        # size the shared MoE workspace BEFORE any hipGraph capture: big
+        # transients allocated inside a capture become graph-owned per
+        # bucket per layer (measured ~277 GB on qwen-3-30b-a3b, call 9)
+        if cfg.device.startswith("cuda"):
+            from ..models.qwen3 import Qwen3MoE
+
+            for mod in self.model.modules():
+                if isinstance(mod, Qwen3MoE):
+                    mod.prealloc_workspace(max(cfg.max_tokens_per_step,
+                                               cfg.max_num_seqs))
+                    break  # workspace is shared across layers
+
         self.graph_runner = None
         if (cfg.device.startswith("cuda") and not cfg.enforce_eager
                 and not self.spec.embedding):

@@ -24,6 +24,12 @@ from ..parallel.tp import TPContext
 from .registry import ModelSpec
 
 
+# shared grouped-MoE workspaces: one per (device, dtype, h, m_l, E_local) —
+# all layers of a model (and all models with identical shapes) reuse them;
+# sized before hipGraph capture so captured decode steps never allocate
+_MOE_WS: dict = {}
+
+
 def _mark_shard(p: nn.Parameter, full_shape, dim: int, tp: TPContext,
                 row_sections=None) -> None:
     """Record how a parameter shards so init can slice a deterministic full
@@ -280,6 +286,46 @@ class Qwen3MoE(nn.Module):
 
     _GG_BM = 64  # grouped-gemm tile height (csrc/grouped_gemm.hip GG_BM)
 
+    def prealloc_workspace(self, max_tokens: int) -> None:
+        """Size the shared grouped-MoE workspace BEFORE hipGraph capture.
+
+        The big per-forward transients (act / out_sorted / gather+combine
+        buffers) must not be allocated inside a captured decode step: every
+        capture would own its own copy per layer per batch bucket — measured
+        ~277 GB of graph-pool memory on qwen-3-30b-a3b before this fix
+        (gpurun call 9). The workspace is module-level (one set per
+        (device, dtype, h, m_l) — all 48 layers share it: execution is
+        sequential) and grow-only."""
+        dev = next(self.parameters()).device
+        self._workspace(max_tokens, dev)
+
+    def _workspace(self, T: int, dev) -> dict:
+        h = self.down.shape[1]
+        m_l = self.gate_up.shape[1] // 2
+        rows_max = T * self.top_k + self.experts_per_rank * self._GG_BM
+        key = (str(dev), self.gate_up.dtype, h, m_l, self.experts_per_rank)
+        ws = _MOE_WS.get(key)
+        if ws is None or ws["rows"] < rows_max:
+            ws = {
+                "rows": rows_max,
+                "act": torch.empty(rows_max, m_l, dtype=self.gate_up.dtype,
+                                   device=dev),
+                "out_sorted": torch.empty(rows_max, h,
+                                          dtype=self.gate_up.dtype,
+                                          device=dev),
+                "row_tok": torch.empty(rows_max, dtype=torch.int32,
+                                       device=dev),
+            }
+            ws["gather"] = torch.empty(T, h, dtype=self.gate_up.dtype,
+                                       device=dev)
+            ws["gather_f32"] = torch.empty(T, h, dtype=torch.float32,
+                                           device=dev)
+            ws["out_f32"] = torch.empty(T, h, dtype=torch.float32, device=dev)
+            ws["max_T"] = T
+            _MOE_WS[key] = ws
+        return ws
+
+    @torch.no_grad()  # inference-only: out= gathers reject autograd operands
     def _forward_grouped(self, x: torch.Tensor) -> torch.Tensor:
         """EXACT dropless top-k execution on the GPU hot path.
 
@@ -288,12 +334,14 @@ class Qwen3MoE(nn.Module):
         bound (hipGraph-capturable) and NO assignment is ever dropped —
         replaces the round-1 capacity-factor bmm path that silently dropped
         over-capacity tokens. All shaping tensors stay
-        on device (no host sync)."""
+        on device (no host sync); the large transients live in the shared
+        prealloc'd workspace (see prealloc_workspace)."""
         T, h = x.shape
         k = self.top_k
         E_l, base = self.experts_per_rank, self.expert_base
         BM = self._GG_BM
         dev = x.device
+        ws = self._workspace(T, dev)
         weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
         flat_e = idx.reshape(-1)                          # [T*k]
         flat_tok = torch.arange(T, device=dev).repeat_interleave(k)
@@ -311,7 +359,8 @@ class Qwen3MoE(nn.Module):
         first = torch.searchsorted(sorted_e, sorted_e, side="left")
         pos = torch.arange(T * k, device=dev) - first
         rows_max = T * k + E_l * BM                       # static bound
-        row_tok = torch.full((rows_max,), -1, dtype=torch.int32, device=dev)
+        row_tok = ws["row_tok"][:rows_max]
+        row_tok.fill_(-1)
         padpos_sorted = (pad_off[e_loc.clamp(0, E_l - 1).long()].long()
                          + pos)
         padpos_sorted = torch.where(valid, padpos_sorted,
@@ -321,10 +370,10 @@ class Qwen3MoE(nn.Module):
         max_tiles = rows_max // BM
 
         m_l = self.gate_up.shape[1] // 2
-        act = x.new_empty(rows_max, m_l)
+        act = ws["act"][:rows_max]
         ops.grouped_gemm(act, x, self.gate_up, row_tok, tile_off, counts,
                          max_tiles, True)
-        out_sorted = x.new_empty(rows_max, h)
+        out_sorted = ws["out_sorted"][:rows_max]
         ops.grouped_gemm(out_sorted, act, self.down, None, tile_off, counts,
                          max_tiles, False)
         # deterministic combine in fixed k-order (no float atomics): map each
@@ -334,9 +383,17 @@ class Qwen3MoE(nn.Module):
         w_flat = weights.reshape(-1).clone()
         inv_valid = torch.zeros(T * k, dtype=torch.bool, device=dev)
         inv_valid[order] = valid
-        w_flat = torch.where(inv_valid, w_flat, torch.zeros_like(w_flat))
-        contrib = out_sorted[padpos.view(T, k)].float()
-        out = (contrib * w_flat.view(T, k, 1)).sum(dim=1)
+        w_flat = torch.where(inv_valid, w_flat,
+                             torch.zeros_like(w_flat)).view(T, k)
+        out = ws["out_f32"][:T]
+        out.zero_()
+        gbuf = ws["gather"][:T]
+        gf32 = ws["gather_f32"][:T]
+        pp = padpos.view(T, k)
+        for j in range(k):
+            torch.index_select(out_sorted, 0, pp[:, j], out=gbuf)
+            gf32.copy_(gbuf)
+            out.addcmul_(gf32, w_flat[:, j].unsqueeze(1))
         return out.to(x.dtype)
 
 
