@@ -51,6 +51,8 @@ void sutro_sampler_fused(const void*, int, const float*, const float*,
 void sutro_grouped_gemm(void*, const void*, const void*, const int*,
                         const int*, const int*, int, int, int, long, int,
                         hipStream_t);
+void sutro_moe_combine(void*, const void*, const long*, const float*, int,
+                       int, int, hipStream_t);
 }
 
 void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w, double eps) {
@@ -251,6 +253,20 @@ void grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
                      cur_stream());
 }
 
+void moe_combine(torch::Tensor out, torch::Tensor rows, torch::Tensor padpos,
+                 torch::Tensor w) {
+  CHECK_CUDA(rows); CHECK_CONTIG(rows); CHECK_BF16(rows);
+  CHECK_CONTIG(out); CHECK_BF16(out);
+  const long T = out.size(0), h = out.size(1), k = padpos.size(1);
+  TORCH_CHECK(h % 8 == 0, "h must be a multiple of 8");
+  TORCH_CHECK(padpos.scalar_type() == at::kLong, "padpos must be int64");
+  TORCH_CHECK(w.scalar_type() == at::kFloat, "w must be f32");
+  TORCH_CHECK(padpos.is_contiguous() && w.is_contiguous(), "contig");
+  sutro_moe_combine(out.data_ptr(), rows.data_ptr(),
+                    padpos.data_ptr<long>(), w.data_ptr<float>(), (int)T,
+                    (int)h, (int)k, cur_stream());
+}
+
 void sampler_fused(torch::Tensor logits, torch::Tensor temps,
                    torch::Tensor topps, torch::Tensor topks, torch::Tensor us,
                    c10::optional<torch::Tensor> mask, long vl,
@@ -294,6 +310,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma32_probe", &mfma32_probe, "MFMA fragment-layout probe");
   m.def("mfma16_probe", &mfma16_probe, "16x16 MFMA fragment-layout probe");
   m.def("hd64_stage_probe", &hd64_stage_probe, "D=64 decode stage dump probe");
+  m.def("moe_combine", &moe_combine, "fused MoE weighted gather-combine");
   m.def("grouped_gemm", &grouped_gemm,
         "dropless MoE grouped GEMM (padded segments, static grid)");
   m.def("sampler_fused", &sampler_fused,

@@ -228,3 +228,43 @@ extern "C" void sutro_grouped_gemm(void* out, const void* a, const void* w,
                        tile_off, counts, n_experts, n_cols, K);
   HIP_CHECK_LAUNCH();
 }
+
+// ---- fused MoE combine: out[t] = sum_j w[t,j] * rows[padpos[t,j]] ----
+// One thread per (token, 8-element h-chunk); k reads + 1 write per thread,
+// f32 accumulate, bf16 store. Replaces a k-long index_select/addcmul loop
+// (8x3 kernels per layer inside the decode graph).
+__global__ void __launch_bounds__(256) moe_combine_kernel(
+    u16* __restrict__ out,          // [T, h] bf16
+    const u16* __restrict__ rows,   // [rows_max, h] bf16
+    const long* __restrict__ padpos,  // [T, k]
+    const float* __restrict__ w,    // [T, k]
+    int T, int h, int k) {
+  const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long chunks = (long)h / 8;
+  if (idx >= (long)T * chunks) return;
+  const int t = (int)(idx / chunks);
+  const int h0 = (int)(idx % chunks) * 8;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int j = 0; j < k; ++j) {
+    const float wj = w[(long)t * k + j];
+    const long r = padpos[(long)t * k + j];
+    const u16x8 v = *(const u16x8*)(rows + r * h + h0);
+#pragma unroll
+    for (int e = 0; e < 8; ++e) acc[e] += wj * bf2f(v[e]);
+  }
+  u16x8 o;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) o[e] = f2bf(acc[e]);
+  *(u16x8*)(out + (long)t * h + h0) = o;
+}
+
+extern "C" void sutro_moe_combine(void* out, const void* rows,
+                                  const long* padpos, const float* w, int T,
+                                  int h, int k, hipStream_t stream) {
+  const long total = (long)T * (h / 8);
+  if (total == 0) return;
+  const long blocks = (total + 255) / 256;
+  hipLaunchKernelGGL(moe_combine_kernel, dim3((unsigned)blocks), dim3(256), 0,
+                     stream, (u16*)out, (const u16*)rows, padpos, w, T, h, k);
+  HIP_CHECK_LAUNCH();
+}

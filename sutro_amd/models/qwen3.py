@@ -305,7 +305,9 @@ class Qwen3MoE(nn.Module):
         rows_max = T * self.top_k + self.experts_per_rank * self._GG_BM
         key = (str(dev), self.gate_up.dtype, h, m_l, self.experts_per_rank)
         ws = _MOE_WS.get(key)
-        if ws is None or ws["rows"] < rows_max:
+        if ws is None or ws["rows"] < rows_max or ws["max_T"] < T:
+            rows_max = max(rows_max, ws["rows"] if ws else 0)
+            T = max(T, ws["max_T"] if ws else 0)
             ws = {
                 "rows": rows_max,
                 "act": torch.empty(rows_max, m_l, dtype=self.gate_up.dtype,
@@ -316,11 +318,13 @@ class Qwen3MoE(nn.Module):
                 "row_tok": torch.empty(rows_max, dtype=torch.int32,
                                        device=dev),
             }
-            ws["gather"] = torch.empty(T, h, dtype=self.gate_up.dtype,
-                                       device=dev)
-            ws["gather_f32"] = torch.empty(T, h, dtype=torch.float32,
-                                           device=dev)
-            ws["out_f32"] = torch.empty(T, h, dtype=torch.float32, device=dev)
+            ws["out"] = torch.empty(T, h, dtype=self.gate_up.dtype,
+                                    device=dev)
+            # zero once: rows past the padded total are never written, but
+            # EP-invalid assignments gather them with weight 0 — torch.empty
+            # garbage could be inf/NaN and 0*NaN = NaN
+            ws["act"].zero_()
+            ws["out_sorted"].zero_()
             ws["max_T"] = T
             _MOE_WS[key] = ws
         return ws
@@ -385,16 +389,16 @@ class Qwen3MoE(nn.Module):
         inv_valid[order] = valid
         w_flat = torch.where(inv_valid, w_flat,
                              torch.zeros_like(w_flat)).view(T, k)
-        out = ws["out_f32"][:T]
-        out.zero_()
-        gbuf = ws["gather"][:T]
-        gf32 = ws["gather_f32"][:T]
         pp = padpos.view(T, k)
-        for j in range(k):
-            torch.index_select(out_sorted, 0, pp[:, j], out=gbuf)
-            gf32.copy_(gbuf)
-            out.addcmul_(gf32, w_flat[:, j].unsqueeze(1))
-        return out.to(x.dtype)
+        if x.is_cuda:
+            # fused weighted gather-combine into the workspace (no allocs
+            # inside captured decode steps)
+            out = ws["out"][:T]
+            ops.moe_combine(out, out_sorted, pp.contiguous(),
+                            w_flat.contiguous())
+            return out
+        contrib = out_sorted[pp].float()
+        return (contrib * w_flat.view(T, k, 1)).sum(dim=1).to(x.dtype)
 
 
 class Qwen3Block(nn.Module):

@@ -178,3 +178,10 @@ def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu):
         return out
     return torch_ref.grouped_gemm(out, a, w, row_tok, tile_off, counts,
                                   max_tiles, gate_silu)
+
+
+def moe_combine(out, rows, padpos, w):
+    """out[t] = sum_j w[t,j] * rows[padpos[t,j]] (fused bf16 gather-combine,
+    csrc/grouped_gemm.hip). GPU-only: the CPU MoE path combines in torch."""
+    _require_hip().moe_combine(out, rows, padpos, w)
+    return out
