@@ -747,7 +747,13 @@ class Sutro(ObservabilityMixin, EmbeddingTemplates, ClassificationTemplates,
                         f"Job {job_id} finished with status {status}: {reason}",
                         "fail"))
                 return None
-            time.sleep(POLL_INTERVAL)
+            if self._is_local:
+                # event-driven: the in-process service notifies on terminal
+                # transitions (no poll-interval latency for p0 rows)
+                self._local_transport().service.wait_terminal(
+                    job_id, min(30.0, timeout - (time.time() - start)))
+            else:
+                time.sleep(POLL_INTERVAL)
         raise TimeoutError(f"job {job_id} did not complete within {timeout}s")
 
     # ---- cache management (CLI) ----

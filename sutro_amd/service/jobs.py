@@ -316,6 +316,9 @@ class JobService:
         self.workers: Dict[str, EngineWorker] = {}
         self.quotas = [dict(q) for q in DEFAULT_QUOTAS]
         self._lock = threading.Lock()
+        # event-driven completion waits (p0 interactive latency: clients
+        # block here instead of sleep-polling status)
+        self._done_cv = threading.Condition()
         self._load_persisted_jobs()
 
     # ---- cost model ----
@@ -589,7 +592,25 @@ class JobService:
 
     # ---- persistence (detach/reattach + engine-restart resume) ----
 
+    def wait_terminal(self, job_id: str, timeout: float) -> Optional[str]:
+        """Block until the job reaches a terminal status (or timeout);
+        returns the status or None. Notified by persist_job; a short
+        condition-wait cap keeps missed notifications harmless."""
+        deadline = time.time() + timeout
+        while True:
+            job = self.jobs.get(job_id)
+            if job is not None and JobStatus.is_terminal(job.status):
+                return job.status
+            rem = deadline - time.time()
+            if rem <= 0:
+                return None
+            with self._done_cv:
+                self._done_cv.wait(min(rem, 0.25))
+
     def persist_job(self, job: JobRecord, with_results: bool = False) -> None:
+        if JobStatus.is_terminal(job.status):
+            with self._done_cv:
+                self._done_cv.notify_all()
         path = os.path.join(self.home, "jobs", f"{job.job_id}.json")
         with open(path, "w") as f:
             json.dump(job.to_public(), f)
