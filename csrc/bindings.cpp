@@ -50,7 +50,7 @@ void sutro_sampler_fused(const void*, int, const float*, const float*,
                          long, int, int, int*, float*, hipStream_t);
 void sutro_grouped_gemm(void*, const void*, const void*, const int*,
                         const int*, const int*, int, int, int, long, int,
-                        hipStream_t);
+                        int, hipStream_t);
 void sutro_moe_combine(void*, const void*, const long*, const float*, int,
                        int, int, hipStream_t);
 }
@@ -228,7 +228,9 @@ torch::Tensor mfma16_probe(torch::Tensor a, torch::Tensor b) {
 
 void grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
                   c10::optional<torch::Tensor> row_tok, torch::Tensor tile_off,
-                  torch::Tensor counts, long max_tiles, bool gate_silu) {
+                  torch::Tensor counts, long max_tiles, bool gate_silu,
+                  long bm) {
+  TORCH_CHECK(bm == 64 || bm == 128, "bm must be 64 or 128");
   CHECK_CUDA(a); CHECK_CONTIG(a); CHECK_BF16(a);
   CHECK_CUDA(w); CHECK_CONTIG(w); CHECK_BF16(w);
   CHECK_CONTIG(out); CHECK_BF16(out);
@@ -250,7 +252,7 @@ void grouped_gemm(torch::Tensor out, torch::Tensor a, torch::Tensor w,
   sutro_grouped_gemm(out.data_ptr(), a.data_ptr(), w.data_ptr(), rt,
                      tile_off.data_ptr<int>(), counts.data_ptr<int>(), (int)E,
                      (int)max_tiles, (int)n_cols, K, gate_silu ? 1 : 0,
-                     cur_stream());
+                     (int)bm, cur_stream());
 }
 
 void moe_combine(torch::Tensor out, torch::Tensor rows, torch::Tensor padpos,

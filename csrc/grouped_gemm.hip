@@ -41,20 +41,26 @@ __device__ __forceinline__ u32 gswz(u32 byte) {
   return byte ^ (((byte >> 8) & 7u) << 4);
 }
 
-#define GG_BM 64
-#define GG_BN 64
-#define GG_WAVES 4
-#define GG_THREADS (GG_WAVES * 64)
+// Tile configurations (all reuse the verified gemm_tn staging recipe):
+//   BM=64,  BN=64,  4 waves 2x2  — small batches (avg segment < 256 rows):
+//                                  least padding waste
+//   BM=128, BN=128, 8 waves 2x4  — big batches: 4x the MFMA work per
+//                                  barrier, B tiles shared across 4 M-rows
+//                                  of waves (measured 3-5x kernel speedup
+//                                  at the b8192 operating point)
+//   BM=128, BN=64,  8 waves 4x2  — big batches whose n_cols % 128 != 0
+//                                  (gpt-oss m=2880)
+// The segment padding (and tile_off table) is computed host-side with the
+// SAME BM the kernel is launched with.
 
-// Stage a 64x64 bf16 tile (64 rows of 128B) with 4 waves; each wave issues
-// 2 x 1KB chunks. Row source resolved by the caller-provided row pointer
-// function: direct rows (weights / act) or gathered rows (row_tok).
-// gsrc==nullptr rows (padding) load row0 of `fallback`.
-template <bool GATHER>
+// Stage a ROWS x 64 bf16 tile (ROWS rows of 128B) with WAVES waves; each
+// wave issues ROWS/(8*WAVES) x 1KB chunks.
+template <int ROWS, int WAVES, bool GATHER>
 __device__ __forceinline__ void gg_stage(
     u16* lds, const u16* gsrc, long row_stride_b, const int* row_tok,
     int m0, int rows_valid, int wave, int lane) {
-  constexpr int CH = 2;  // 8KB tile / 1KB chunks / 4 waves
+  constexpr int CH = ROWS / (8 * WAVES);
+  static_assert(CH >= 1, "tile too small for wave count");
 #pragma unroll
   for (int j = 0; j < CH; ++j) {
     const int c = wave * CH + j;
@@ -83,8 +89,9 @@ __device__ __forceinline__ const bf16x8* gg_frag(const u16* lds, int row,
 
 // GATE_SILU=1: B has two streams (gate/up at N-offset 0 and n_cols), output
 // n_cols wide with silu(g)*u. GATE_SILU=0: plain B, plain store.
-template <int GATE_SILU>
-__global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
+// Waves arranged WROWS(M) x WCOLS(N); wave tile = (BM/WROWS) x (BN/WCOLS).
+template <int GATE_SILU, int BM, int BN, int WROWS, int WCOLS>
+__global__ __launch_bounds__(WROWS* WCOLS * 64) void grouped_gemm_kernel(
     u16* __restrict__ out,        // [rows_max, n_cols] bf16
     const u16* __restrict__ a,    // GATHER: [T, K]; else [rows_max, K]
     const u16* __restrict__ w,    // [E, N, K] bf16 (N = n_cols or 2*n_cols)
@@ -92,8 +99,13 @@ __global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
     const int* __restrict__ tile_off,   // [E+1] padded-offset / BM
     const int* __restrict__ counts,     // [E] live rows per expert
     int n_experts, int n_cols, long K) {
+  constexpr int WAVES = WROWS * WCOLS;
+  constexpr int WM = BM / WROWS;
+  constexpr int WN = BN / WCOLS;
+  constexpr int MF = WM / 16;
+  constexpr int NF = WN / 16;
   const int t = blockIdx.x;
-  const int n0 = blockIdx.y * GG_BN;
+  const int n0 = blockIdx.y * BN;
   if (t >= tile_off[n_experts]) return;
   // binary search: expert e with tile_off[e] <= t < tile_off[e+1]
   int lo = 0, hi = n_experts - 1;
@@ -103,18 +115,17 @@ __global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
     else hi = mid - 1;
   }
   const int e = lo;
-  const int m0 = (t - tile_off[e]) * GG_BM;           // within-segment row
-  const int seg0 = tile_off[e] * GG_BM;               // segment global base
+  const int m0 = (t - tile_off[e]) * BM;              // within-segment row
+  const int seg0 = tile_off[e] * BM;                  // segment global base
   const int rows_valid = counts[e] - m0;              // live rows this tile
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wm = wave >> 1, wn = wave & 1;            // 2(M) x 2(N)
-  constexpr int MF = 2, NF = 2;                       // 32x32 wave tile
+  const int wm = wave / WCOLS, wn = wave % WCOLS;
 
-  __shared__ u16 smem[2][(GG_BM + (GATE_SILU ? 2 : 1) * GG_BN) * 64];
-  const int b_off = GG_BM * 64;
-  const int b2_off = (GG_BM + GG_BN) * 64;
+  __shared__ u16 smem[2][(BM + (GATE_SILU ? 2 : 1) * BN) * 64];
+  const int b_off = BM * 64;
+  const int b2_off = (BM + BN) * 64;
 
   const long Kb = K * 2;
   const u16* wa = w + (long)e * (GATE_SILU ? 2 : 1) * n_cols * K;
@@ -123,23 +134,25 @@ __global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
   const u16* aa = GATE_SILU ? a : a + (long)seg0 * K;
   const int* rt = row_tok ? row_tok + seg0 : nullptr;
 
-  f32x4 accg[MF][NF], accu[MF][NF];
+  f32x4 accg[MF][NF], accu[GATE_SILU ? MF : 1][GATE_SILU ? NF : 1];
 #pragma unroll
   for (int i = 0; i < MF; ++i)
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
       accg[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
-      accu[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+      if (GATE_SILU) accu[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
     }
 
   // prologue
   if (GATE_SILU)
-    gg_stage<true>(smem[0], aa, Kb, rt, m0, rows_valid, wave, lane);
+    gg_stage<BM, WAVES, true>(smem[0], aa, Kb, rt, m0, rows_valid, wave, lane);
   else
-    gg_stage<false>(smem[0], aa, Kb, nullptr, m0, GG_BM, wave, lane);
-  gg_stage<false>(smem[0] + b_off, wg, Kb, nullptr, 0, GG_BN, wave, lane);
+    gg_stage<BM, WAVES, false>(smem[0], aa, Kb, nullptr, m0, BM, wave, lane);
+  gg_stage<BN, WAVES, false>(smem[0] + b_off, wg, Kb, nullptr, 0, BN, wave,
+                             lane);
   if (GATE_SILU)
-    gg_stage<false>(smem[0] + b2_off, wu, Kb, nullptr, 0, GG_BN, wave, lane);
+    gg_stage<BN, WAVES, false>(smem[0] + b2_off, wu, Kb, nullptr, 0, BN, wave,
+                               lane);
   __syncthreads();
 
   const int KT = (int)(K >> 6);
@@ -151,30 +164,31 @@ __global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
     if (kt + 1 < KT) {
       const long ko = (long)(kt + 1) * 64;
       if (GATE_SILU)
-        gg_stage<true>(smem[p ^ 1], aa + ko, Kb, rt, m0, rows_valid, wave,
-                       lane);
+        gg_stage<BM, WAVES, true>(smem[p ^ 1], aa + ko, Kb, rt, m0,
+                                  rows_valid, wave, lane);
       else
-        gg_stage<false>(smem[p ^ 1], aa + ko, Kb, nullptr, m0, GG_BM, wave,
-                        lane);
-      gg_stage<false>(smem[p ^ 1] + b_off, wg + ko, Kb, nullptr, 0, GG_BN,
-                      wave, lane);
+        gg_stage<BM, WAVES, false>(smem[p ^ 1], aa + ko, Kb, nullptr, m0, BM,
+                                   wave, lane);
+      gg_stage<BN, WAVES, false>(smem[p ^ 1] + b_off, wg + ko, Kb, nullptr, 0,
+                                 BN, wave, lane);
       if (GATE_SILU)
-        gg_stage<false>(smem[p ^ 1] + b2_off, wu + ko, Kb, nullptr, 0, GG_BN,
-                        wave, lane);
+        gg_stage<BN, WAVES, false>(smem[p ^ 1] + b2_off, wu + ko, Kb, nullptr,
+                                   0, BN, wave, lane);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       const int kk16 = kk * 4 + (lane >> 4);
-      bf16x8 bg[NF], bu[NF];
+      bf16x8 bg[NF], bu[GATE_SILU ? NF : 1];
 #pragma unroll
       for (int j = 0; j < NF; ++j) {
-        bg[j] = *gg_frag(bgT, wn * 32 + j * 16 + (lane & 15), kk16);
+        bg[j] = *gg_frag(bgT, wn * WN + j * 16 + (lane & 15), kk16);
         if (GATE_SILU)
-          bu[j] = *gg_frag(buT, wn * 32 + j * 16 + (lane & 15), kk16);
+          bu[j] = *gg_frag(buT, wn * WN + j * 16 + (lane & 15), kk16);
       }
 #pragma unroll
       for (int i = 0; i < MF; ++i) {
-        const bf16x8 af = *gg_frag(aT, wm * 32 + i * 16 + (lane & 15), kk16);
+        const bf16x8 af =
+            *gg_frag(aT, wm * WM + i * 16 + (lane & 15), kk16);
 #pragma unroll
         for (int j = 0; j < NF; ++j) {
           accg[i][j] = mfma16g(af, bg[j], accg[i][j]);
@@ -192,12 +206,11 @@ __global__ __launch_bounds__(GG_THREADS) void grouped_gemm_kernel(
   for (int i = 0; i < MF; ++i) {
 #pragma unroll
     for (int j = 0; j < NF; ++j) {
-      const int gn = n0 + wn * 32 + j * 16 + dcol;
-      const int grow0 = wm * 32 + i * 16 + drow;
+      const int gn = n0 + wn * WN + j * 16 + dcol;
+      const int grow0 = wm * WM + i * 16 + drow;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int grow = grow0 + r;
-        if (grow >= GG_BM) continue;  // (never: drow+r < 16*MF)
         const long o = (long)(seg0 + m0 + grow) * n_cols + gn;
         float v = accg[i][j][r];
         if (GATE_SILU) {
@@ -214,19 +227,31 @@ extern "C" void sutro_grouped_gemm(void* out, const void* a, const void* w,
                                    const int* row_tok, const int* tile_off,
                                    const int* counts, int n_experts,
                                    int max_tiles, int n_cols, long K,
-                                   int gate_silu, hipStream_t stream) {
+                                   int gate_silu, int bm,
+                                   hipStream_t stream) {
   if (max_tiles == 0) return;
-  dim3 grid((unsigned)max_tiles, (unsigned)(n_cols / GG_BN)),
-      block(GG_THREADS);
-  if (gate_silu)
-    hipLaunchKernelGGL((grouped_gemm_kernel<1>), grid, block, 0, stream,
-                       (u16*)out, (const u16*)a, (const u16*)w, row_tok,
-                       tile_off, counts, n_experts, n_cols, K);
-  else
-    hipLaunchKernelGGL((grouped_gemm_kernel<0>), grid, block, 0, stream,
-                       (u16*)out, (const u16*)a, (const u16*)w, row_tok,
-                       tile_off, counts, n_experts, n_cols, K);
-  HIP_CHECK_LAUNCH();
+#define GG_LAUNCH(GS, BM, BN, WR, WC)                                       \
+  {                                                                         \
+    dim3 grid((unsigned)max_tiles, (unsigned)(n_cols / BN)),                \
+        block(WR* WC * 64);                                                 \
+    hipLaunchKernelGGL((grouped_gemm_kernel<GS, BM, BN, WR, WC>), grid,     \
+                       block, 0, stream, (u16*)out, (const u16*)a,          \
+                       (const u16*)w, row_tok, tile_off, counts, n_experts, \
+                       n_cols, K);                                          \
+    HIP_CHECK_LAUNCH();                                                     \
+    return;                                                                 \
+  }
+  if (bm == 128 && n_cols % 128 == 0) {
+    if (gate_silu) GG_LAUNCH(1, 128, 128, 2, 4)
+    else GG_LAUNCH(0, 128, 128, 2, 4)
+  }
+  if (bm == 128) {
+    if (gate_silu) GG_LAUNCH(1, 128, 64, 4, 2)
+    else GG_LAUNCH(0, 128, 64, 4, 2)
+  }
+  if (gate_silu) GG_LAUNCH(1, 64, 64, 2, 2)
+  else GG_LAUNCH(0, 64, 64, 2, 2)
+#undef GG_LAUNCH
 }
 
 // ---- fused MoE combine: out[t] = sum_j w[t,j] * rows[padpos[t,j]] ----

@@ -284,7 +284,9 @@ class Qwen3MoE(nn.Module):
         out.index_add_(0, flat_tok[order], back_rows.float() * w_sorted)
         return out.to(x_local.dtype)
 
-    _GG_BM = 64  # grouped-gemm tile height (csrc/grouped_gemm.hip GG_BM)
+    _GG_BM = 128  # MAX grouped-gemm tile height (workspace sizing); the
+    # per-call tile is 128 when average segments fill it (big batches),
+    # else 64 (csrc/grouped_gemm.hip tile configurations)
 
     def prealloc_workspace(self, max_tokens: int) -> None:
         """Size the shared grouped-MoE workspace BEFORE hipGraph capture.
@@ -343,7 +345,9 @@ class Qwen3MoE(nn.Module):
         T, h = x.shape
         k = self.top_k
         E_l, base = self.experts_per_rank, self.expert_base
-        BM = self._GG_BM
+        # big batches pad segments to 128 (8-wave 128-row tiles run ~4x the
+        # MFMA work per barrier); small batches keep 64 to bound padding
+        BM = 128 if T * k >= E_l * 256 else 64
         dev = x.device
         ws = self._workspace(T, dev)
         weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
@@ -376,10 +380,10 @@ class Qwen3MoE(nn.Module):
         m_l = self.gate_up.shape[1] // 2
         act = ws["act"][:rows_max]
         ops.grouped_gemm(act, x, self.gate_up, row_tok, tile_off, counts,
-                         max_tiles, True)
+                         max_tiles, True, bm=BM)
         out_sorted = ws["out_sorted"][:rows_max]
         ops.grouped_gemm(out_sorted, act, self.down, None, tile_off, counts,
-                         max_tiles, False)
+                         max_tiles, False, bm=BM)
         # deterministic combine in fixed k-order (no float atomics): map each
         # original (token, j) assignment back to its padded row
         padpos = torch.empty(T * k, dtype=torch.long, device=dev)

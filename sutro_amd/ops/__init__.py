@@ -169,15 +169,17 @@ def sampler_fused(logits, temps, top_ps, top_ks, u, packed_mask, vocab_limit,
                                  packed_mask, vocab_limit, out_tok, out_lp)
 
 
-def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu):
+def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu,
+                 bm=64):
     """Dropless-MoE grouped GEMM: csrc/grouped_gemm.hip on GPU, torch
-    reference on CPU (same padded-segment layout)."""
+    reference on CPU (same padded-segment layout). bm = segment tile height
+    (64 for small batches, 128 for large — host pads with the same value)."""
     if _use_hip(a):
         _require_hip().grouped_gemm(out, a, w, row_tok, tile_off, counts,
-                                    max_tiles, gate_silu)
+                                    max_tiles, gate_silu, bm)
         return out
     return torch_ref.grouped_gemm(out, a, w, row_tok, tile_off, counts,
-                                  max_tiles, gate_silu)
+                                  max_tiles, gate_silu, bm)
 
 
 def moe_combine(out, rows, padpos, w):

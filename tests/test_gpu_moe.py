@@ -126,3 +126,85 @@ def test_moe_model_generates_gpu():
                                                 temperature=0.7))
     assert len(outs) == 2
     assert all(isinstance(o, str) for o in outs)
+
+
+def _segments_bm(counts, bm):
+    counts_t = torch.tensor(counts, dtype=torch.int32, device=DEV)
+    padded = (counts_t + bm - 1) // bm * bm
+    pad_off = torch.zeros(len(counts) + 1, dtype=torch.int32, device=DEV)
+    pad_off[1:] = torch.cumsum(padded, 0)
+    return counts_t, pad_off // bm, int(pad_off[-1]) // bm
+
+
+@pytest.mark.parametrize("n_cols", [256, 192])  # 128-divisible and not
+def test_grouped_gemm_bm128_vs_ref(n_cols):
+    """The 128-row tile configs (8 waves; BN=128 and the BN=64 fallback for
+    n_cols % 128 != 0) match the reference on skewed segments."""
+    require_gpu()
+    torch.manual_seed(3)
+    E, K = 4, 256
+    counts = [300, 0, 129, 64]
+    counts_t, tile_off, max_tiles = _segments_bm(counts, 128)
+    rows_max = int(tile_off[-1]) * 128 + 128
+    a = torch.randn(rows_max, K).to(torch.bfloat16).to(DEV)
+    w = (torch.randn(E, n_cols, K) * 0.2).to(torch.bfloat16).to(DEV)
+    out = torch.zeros(rows_max, n_cols, dtype=torch.bfloat16, device=DEV)
+    ops.grouped_gemm(out, a, w, None, tile_off, counts_t, max_tiles, False,
+                     bm=128)
+    ref = torch.zeros_like(out)
+    torch_ref.grouped_gemm(ref, a, w, None, tile_off.cpu(), counts_t.cpu(),
+                           max_tiles, False, 128)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * 128
+        torch.testing.assert_close(out[s0:s0 + c].float(),
+                                   ref[s0:s0 + c].float(),
+                                   atol=5e-2, rtol=5e-2)
+
+
+def test_grouped_gemm_bm128_gather_silu_vs_ref():
+    require_gpu()
+    torch.manual_seed(4)
+    E, K, m = 4, 192, 256
+    T = 600
+    counts = [513, 0, 128, 77]
+    counts_t, tile_off, max_tiles = _segments_bm(counts, 128)
+    rows_max = int(tile_off[-1]) * 128 + 128
+    x = torch.randn(T, K).to(torch.bfloat16).to(DEV)
+    w = (torch.randn(E, 2 * m, K) * 0.2).to(torch.bfloat16).to(DEV)
+    rng = np.random.default_rng(9)
+    row_tok = torch.full((rows_max,), -1, dtype=torch.int32, device=DEV)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * 128
+        row_tok[s0:s0 + c] = torch.from_numpy(
+            rng.integers(0, T, size=c).astype(np.int32)).to(DEV)
+    out = torch.zeros(rows_max, m, dtype=torch.bfloat16, device=DEV)
+    ops.grouped_gemm(out, x, w, row_tok, tile_off, counts_t, max_tiles, True,
+                     bm=128)
+    ref = torch.zeros_like(out)
+    torch_ref.grouped_gemm(ref, x, w, row_tok.cpu(), tile_off.cpu(),
+                           counts_t.cpu(), max_tiles, True, 128)
+    for e, c in enumerate(counts):
+        s0 = int(tile_off[e]) * 128
+        torch.testing.assert_close(out[s0:s0 + c].float(),
+                                   ref[s0:s0 + c].float(),
+                                   atol=5e-2, rtol=5e-2)
+
+
+def test_moe_grouped_gpu_big_batch_tiles():
+    """Full layer at a T large enough to select the 128-row tiles."""
+    require_gpu()
+    from sutro_amd.models.qwen3 import Qwen3MoE
+    from sutro_amd.models.registry import ModelSpec
+
+    spec = ModelSpec(name="tiny-moe-big", hidden_size=256, num_layers=1,
+                     num_heads=4, num_kv_heads=2, head_dim=64,
+                     intermediate_size=0, vocab_size=512, num_experts=4,
+                     experts_per_token=2, moe_intermediate_size=256)
+    torch.manual_seed(5)
+    moe = Qwen3MoE(spec, torch.bfloat16).to(DEV)
+    for p in moe.parameters():
+        torch.nn.init.normal_(p, std=0.05)
+    x = (torch.randn(1024, 256) * 0.5).to(torch.bfloat16).to(DEV)  # T*k=2048 >= E*256
+    ref = moe._forward_loop(x).float()
+    got = moe._forward_grouped(x).float()
+    torch.testing.assert_close(got, ref, atol=3e-2, rtol=3e-2)
