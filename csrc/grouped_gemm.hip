@@ -238,9 +238,10 @@ extern "C" void sutro_grouped_gemm(void* out, const void* a, const void* w,
                        block, 0, stream, (u16*)out, (const u16*)a,          \
                        (const u16*)w, row_tok, tile_off, counts, n_experts, \
                        n_cols, K);                                          \
-    HIP_CHECK_LAUNCH();                                                     \
     return;                                                                 \
   }
+  /* no hipGetLastError in this launcher: error-state queries are rejected
+     while a stream is capturing (decode steps replay under hipGraphs) */
   if (bm == 128 && n_cols % 128 == 0) {
     if (gate_silu) GG_LAUNCH(1, 128, 128, 2, 4)
     else GG_LAUNCH(0, 128, 128, 2, 4)
@@ -289,7 +290,8 @@ extern "C" void sutro_moe_combine(void* out, const void* rows,
   const long total = (long)T * (h / 8);
   if (total == 0) return;
   const long blocks = (total + 255) / 256;
+  // no hipGetLastError here: error-state queries are rejected while a
+  // stream is capturing (decode steps replay this inside hipGraphs)
   hipLaunchKernelGGL(moe_combine_kernel, dim3((unsigned)blocks), dim3(256), 0,
                      stream, (u16*)out, (const u16*)rows, padpos, w, T, h, k);
-  HIP_CHECK_LAUNCH();
 }

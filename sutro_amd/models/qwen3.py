@@ -358,7 +358,12 @@ class Qwen3MoE(nn.Module):
         sorted_tok = flat_tok[order]
         e_loc = sorted_e - base
         valid = (e_loc >= 0) & (e_loc < E_l)              # EP: local only
-        counts = torch.bincount(e_loc[valid], minlength=E_l).to(torch.int32)
+        # capture-clean counting: bincount and boolean-mask indexing host-
+        # sync (output size), which silently broke hipGraph capture of MoE
+        # decode steps — every shaping op below is static-shaped
+        counts = torch.zeros(E_l, dtype=torch.int32, device=dev)
+        counts.index_add_(0, e_loc.clamp(0, E_l - 1),
+                          valid.to(torch.int32))
         padded = (counts + (BM - 1)) // BM * BM
         pad_off = torch.zeros(E_l + 1, dtype=torch.int32, device=dev)
         pad_off[1:] = torch.cumsum(padded, 0)
@@ -371,10 +376,13 @@ class Qwen3MoE(nn.Module):
         row_tok.fill_(-1)
         padpos_sorted = (pad_off[e_loc.clamp(0, E_l - 1).long()].long()
                          + pos)
+        # invalid (non-local EP) assignments redirect to the last slot,
+        # which lies beyond every padded segment (padded_total < rows_max)
+        # so no tile ever reads it — scatter_ keeps shapes static
         padpos_sorted = torch.where(valid, padpos_sorted,
                                     torch.full_like(padpos_sorted,
                                                     rows_max - 1))
-        row_tok[padpos_sorted[valid]] = sorted_tok[valid].to(torch.int32)
+        row_tok.scatter_(0, padpos_sorted, sorted_tok.to(torch.int32))
         max_tiles = rows_max // BM
 
         m_l = self.gate_up.shape[1] // 2
