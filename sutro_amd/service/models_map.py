@@ -41,5 +41,14 @@ def resolve_engine_config(
         device = "cuda" if torch.cuda.is_available() else "cpu"
     if device == "cpu" and spec.param_count() > _CPU_PARAM_LIMIT:
         spec = dev_proxy(spec)
+    if (device.startswith("cuda") and spec.num_experts > 0
+            and "max_num_seqs" not in kwargs):
+        # MoE splits the batch across experts: bigger resident batches fill
+        # the grouped-GEMM tiles (a3b measures 19.4k tok/s at 2048 rows vs
+        # 30.8k at 8192; mixtral 11.7k at 4096 — PROFILES.md r2 captures
+        # 8-10). max_num_seqs is an admission CAP — the scheduler only
+        # admits rows whose KV fits, so heavier models simply run fewer.
+        kwargs["max_num_seqs"] = 8192 if spec.param_count() < 40e9 else 4096
+        kwargs.setdefault("max_tokens_per_step", 65536)
     return EngineConfig(spec=spec, device=device,
                         max_model_len=min(max_model_len, spec.max_context), **kwargs)
