@@ -126,8 +126,9 @@ class Qwen3MLP(nn.Module):
 
 
 class Qwen3MoE(nn.Module):
-    """Top-k routed SwiGLU experts. Round-1 execution: sort tokens by expert and
-    run per-expert GEMMs on contiguous segments (grouped-GEMM kernel later)."""
+    """Top-k routed SwiGLU experts, executed EXACTLY (dropless) through the
+    grouped-GEMM kernel (csrc/grouped_gemm.hip) on GPU with hipGraph-safe
+    static shaping; per-expert segment loop as the CPU reference."""
 
     def __init__(self, spec: ModelSpec, dtype: torch.dtype,
                  tp: Optional[TPContext] = None, ep: bool = False):
