@@ -149,6 +149,13 @@ class EngineWorker:
     def submit_partial(self, job: JobRecord, service: "JobService",
                        rows_idx) -> None:
         """Admit only the given row indices (engine-restart resume)."""
+        if self.dead is not None:
+            with job.lock:
+                job.status = JobStatus.FAILED
+                job.failure_reason = {"message": self.dead}
+                job.datetime_completed = _now()
+            service.persist_job(job)
+            return
         with self._lock:
             self._inbox.append((job, service, list(rows_idx)))
         self._wake.set()
