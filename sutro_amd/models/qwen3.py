@@ -176,6 +176,11 @@ class Qwen3MoE(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if x.is_cuda:
+            if (self._PREFILL_BLASLT_MIN
+                    and x.shape[0] >= self._PREFILL_BLASLT_MIN
+                    and not torch.cuda.is_current_stream_capturing()):
+                # big eager waves: library GEMMs on true segment sizes
+                return self.tp.all_reduce(self._forward_prefill_blaslt(x))
             return self.tp.all_reduce(self._forward_grouped(x))
         return self.tp.all_reduce(self._forward_loop(x))
 
@@ -331,6 +336,58 @@ class Qwen3MoE(nn.Module):
             ws["max_T"] = T
             _MOE_WS[key] = ws
         return ws
+
+    # tokens at/above which the prefill path hands expert segments to
+    # hipBLASLt (library GEMMs run ~1.2 PF/s vs the grouped kernel's ~0.4
+    # at wave shapes; prefill is never hipGraph-captured so the host sync
+    # for dynamic per-expert M is free). 0 disables.
+    import os as _os
+    _PREFILL_BLASLT_MIN = int(_os.environ.get("SUTRO_MOE_PREFILL_BLASLT_MIN",
+                                              "8192"))
+    del _os
+
+    @torch.no_grad()
+    def _forward_prefill_blaslt(self, x: torch.Tensor) -> torch.Tensor:
+        """Exact dropless MoE for big (eager) prefill waves: sort by expert,
+        per-expert hipBLASLt GEMMs on the true segment sizes, deterministic
+        fused combine (moe_combine — index_add_ would race float atomics)."""
+        T, h = x.shape
+        k = self.top_k
+        E_l, base = self.experts_per_rank, self.expert_base
+        dev = x.device
+        weights, idx = torch_ref.topk_softmax_router(self.router(x), k)
+        flat_e = idx.reshape(-1)
+        flat_tok = torch.arange(T, device=dev).repeat_interleave(k)
+        order = torch.argsort(flat_e, stable=True)
+        sorted_e = flat_e[order]
+        sorted_tok = flat_tok[order]
+        e_loc = sorted_e - base
+        valid = (e_loc >= 0) & (e_loc < E_l)
+        counts = torch.zeros(E_l, dtype=torch.int64, device=dev)
+        counts.index_add_(0, e_loc.clamp(0, E_l - 1), valid.to(torch.int64))
+        gathered = x[sorted_tok]                      # [T*k, h], sorted order
+        out_sorted = torch.zeros_like(gathered)
+        counts_l = counts.tolist()                    # host sync (eager only)
+        # EP: non-local assignments sort to the edges; find the local span
+        start = int(torch.searchsorted(sorted_e, base).item()) if base else 0
+        for e in range(E_l):
+            n = counts_l[e]
+            if n:
+                seg = gathered[start:start + n]
+                act = ops.silu_mul(F.linear(seg, self.gate_up[e]))
+                out_sorted[start:start + n] = F.linear(act, self.down[e])
+            start += n
+        w_flat = weights.reshape(-1)
+        inv_valid = torch.zeros(T * k, dtype=torch.bool, device=dev)
+        inv_valid[order] = valid
+        w_flat = torch.where(inv_valid, w_flat,
+                             torch.zeros_like(w_flat)).view(T, k)
+        padpos = torch.empty(T * k, dtype=torch.long, device=dev)
+        padpos[order] = torch.arange(T * k, device=dev)
+        out = torch.empty_like(x)
+        ops.moe_combine(out, out_sorted, padpos.view(T, k).contiguous(),
+                        w_flat.contiguous())
+        return out
 
     @torch.no_grad()  # inference-only: out= gathers reject autograd operands
     def _forward_grouped(self, x: torch.Tensor) -> torch.Tensor:

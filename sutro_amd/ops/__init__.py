@@ -183,7 +183,11 @@ def grouped_gemm(out, a, w, row_tok, tile_off, counts, max_tiles, gate_silu,
 
 
 def moe_combine(out, rows, padpos, w):
-    """out[t] = sum_j w[t,j] * rows[padpos[t,j]] (fused bf16 gather-combine,
-    csrc/grouped_gemm.hip). GPU-only: the CPU MoE path combines in torch."""
-    _require_hip().moe_combine(out, rows, padpos, w)
+    """out[t] = sum_j w[t,j] * rows[padpos[t,j]]: fused bf16 gather-combine
+    (csrc/grouped_gemm.hip) on GPU, deterministic torch gather on CPU."""
+    if _use_hip(rows):
+        _require_hip().moe_combine(out, rows, padpos, w)
+        return out
+    contrib = rows[padpos].float()          # [T, k, h]
+    out.copy_((contrib * w.unsqueeze(-1)).sum(dim=1).to(out.dtype))
     return out
