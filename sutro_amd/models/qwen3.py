@@ -337,13 +337,16 @@ class Qwen3MoE(nn.Module):
             _MOE_WS[key] = ws
         return ws
 
-    # tokens at/above which the prefill path hands expert segments to
-    # hipBLASLt (library GEMMs run ~1.2 PF/s vs the grouped kernel's ~0.4
-    # at wave shapes; prefill is never hipGraph-captured so the host sync
-    # for dynamic per-expert M is free). 0 disables.
+    # A/B knob (default OFF): hand prefill-wave expert segments to
+    # per-expert hipBLASLt GEMMs instead of the grouped kernel. Measured
+    # NET SLOWER at the a3b b8192 operating point (28.1k vs 30.8k tok/s):
+    # 128 experts x (2 GEMM + silu) launches + a host sync per layer
+    # outweigh the library's higher per-GEMM rate at ~4096-row segments
+    # (PROFILES.md r2 capture 13). Kept for future shapes where segments
+    # are far larger; set SUTRO_MOE_PREFILL_BLASLT_MIN=8192 to re-enable.
     import os as _os
     _PREFILL_BLASLT_MIN = int(_os.environ.get("SUTRO_MOE_PREFILL_BLASLT_MIN",
-                                              "8192"))
+                                              "0"))
     del _os
 
     @torch.no_grad()
