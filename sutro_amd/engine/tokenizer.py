@@ -139,6 +139,14 @@ class BPETokenizer:
         parts.append(f"<|user|>\n{user}\n<|assistant|>\n")
         return self.encode("".join(parts), add_bos=True)
 
+    def render_prompts(self, users: List[str],
+                       system: Optional[str] = None) -> List[List[int]]:
+        """Batch chat-template render (one Rust batch-encode call — large
+        job admissions tokenize thousands of rows)."""
+        pre = f"<|system|>\n{system}\n" if system else ""
+        texts = [f"{pre}<|user|>\n{u}\n<|assistant|>\n" for u in users]
+        return self.encode_batch(texts, add_bos=True)
+
 
 class ByteTokenizer:
     """Legacy lossless byte tokenizer (ids 3..258 = bytes). Kept for unit
@@ -175,6 +183,10 @@ class ByteTokenizer:
             parts.append(f"<|system|>\n{system}\n")
         parts.append(f"<|user|>\n{user}\n<|assistant|>\n")
         return self.encode("".join(parts), add_bos=True)
+
+    def render_prompts(self, users: List[str],
+                       system: Optional[str] = None) -> List[List[int]]:
+        return [self.render_prompt(u, system) for u in users]
 
 
 _LOCK = threading.Lock()

@@ -86,8 +86,9 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
                                      else None)
             fsm_id = fsm_cache[job_id]
             default_max = 1024 if schema is not None else cfg.default_max_new_tokens
-            for row_idx, text in rows:
-                ids = tok.render_prompt(text, opts.get("system_prompt"))
+            all_ids = tok.render_prompts([t for _, t in rows],
+                                         opts.get("system_prompt"))
+            for (row_idx, text), ids in zip(rows, all_ids):
                 sp = SamplingParams.from_dict(opts.get("sampling_params"),
                                               default_max)
                 if opts.get("random_seed_per_input"):

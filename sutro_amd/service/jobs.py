@@ -197,11 +197,12 @@ class EngineWorker:
                 self._fsm_by_schema[key] = fsm_id
         default_max = 1024 if schema is not None else eng.cfg.default_max_new_tokens
         t_start = time.time()
-        indices = rows_idx if rows_idx is not None else range(len(job.inputs))
-        for i in indices:
-            row = job.inputs[i]
-            text = row if isinstance(row, str) else json.dumps(row)
-            ids = self.tokenizer.render_prompt(text, job.system_prompt)
+        indices = list(rows_idx if rows_idx is not None
+                       else range(len(job.inputs)))
+        texts = [job.inputs[i] if isinstance(job.inputs[i], str)
+                 else json.dumps(job.inputs[i]) for i in indices]
+        all_ids = self.tokenizer.render_prompts(texts, job.system_prompt)
+        for i, ids in zip(indices, all_ids):
             sp = SamplingParams.from_dict(job.sampling_params, default_max)
             if job.random_seed_per_input:
                 sp.seed = i
