@@ -527,6 +527,13 @@ class LLMEngine:
         if req.fsm_id is not None:
             fsm = self._fsms[req.fsm_id]
             req.fsm_state = fsm.advance(req.fsm_state, tok)
+            if req.fsm_state < 0:
+                # defensive: a token escaped the mask (cannot happen while
+                # FSM support is non-empty; guards a corrupted state from
+                # ever reaching mask_rows where it would KeyError)
+                self.scheduler.finish(req, FinishReason.STOP)
+                stats.finished.append(req)
+                return
             if fsm.must_stop(req.fsm_state):
                 self.scheduler.finish(req, FinishReason.STOP)
                 stats.finished.append(req)
