@@ -199,3 +199,51 @@ def test_threshold_semantics_top_p_mass_gt():
     seen = set(toks.tolist())
     assert 2 not in seen
     assert seen == {0, 1}
+
+
+def test_reference_matches_bruteforce_oracle():
+    """sample_torch_reference vs a direct O(V^2) transcription of the
+    documented semantics (independent oracle; quantized logits keep float
+    boundaries well-separated so comparisons are exact)."""
+    import numpy as np
+
+    from sutro_amd.engine.sampler import sample_torch_reference
+
+    rng = np.random.default_rng(7)
+    for trial in range(200):
+        vl = int(rng.integers(3, 24))
+        logits = torch.tensor(
+            rng.integers(-6, 7, size=vl).astype(np.float32) / 2.0)
+        temp = float(rng.choice([0.5, 1.0, 1.7]))
+        top_p = float(rng.choice([0.3, 0.7, 0.95, 1.0]))
+        top_k = int(rng.choice([1, 2, 5, vl]))
+        u = float(rng.uniform(0.01, 0.99))
+
+        # oracle
+        s = (logits - logits.max()) / temp
+        p = torch.exp(s)
+        z = float(p.sum())
+        keep = []
+        for i in range(vl):
+            cnt_gt = sum(1 for j in range(vl) if s[j] > s[i])
+            mass_gt = sum(float(p[j]) for j in range(vl) if s[j] > s[i])
+            keep.append(cnt_gt < top_k and mass_gt < top_p * z)
+        cum, tok_oracle = 0.0, None
+        target = u * sum(float(p[i]) for i in range(vl) if keep[i])
+        for i in range(vl):
+            if keep[i]:
+                cum += float(p[i])
+                if cum > target:
+                    tok_oracle = i
+                    break
+        if tok_oracle is None:
+            tok_oracle = max(i for i in range(vl) if keep[i])
+
+        toks, _ = sample_torch_reference(
+            logits.unsqueeze(0),
+            torch.tensor([temp]), torch.tensor([top_p]),
+            torch.tensor([top_k], dtype=torch.int32),
+            torch.tensor([u]), vl)
+        assert int(toks[0]) == tok_oracle, (
+            f"trial {trial}: vl={vl} T={temp} p={top_p} k={top_k} u={u} "
+            f"logits={logits.tolist()}")
