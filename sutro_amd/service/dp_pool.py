@@ -80,11 +80,15 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
                 from ..engine.guided import reasoning_wrapper_schema
 
                 schema = reasoning_wrapper_schema(schema)
-            if job_id not in fsm_cache:
-                fsm_cache[job_id] = (eng.register_fsm(schema)
-                                     if schema is not None and not spec.embedding
-                                     else None)
-            fsm_id = fsm_cache[job_id]
+            skey = (json.dumps(schema, sort_keys=True)
+                    if schema is not None else None)
+            if skey not in fsm_cache:
+                # keyed by schema (not job): repeated-schema jobs reuse the
+                # compiled DFA + mask table
+                fsm_cache[skey] = (eng.register_fsm(schema)
+                                   if schema is not None and not spec.embedding
+                                   else None)
+            fsm_id = fsm_cache[skey]
             default_max = 1024 if schema is not None else cfg.default_max_new_tokens
             all_ids = tok.render_prompts([t for _, t in rows],
                                          opts.get("system_prompt"))
