@@ -86,9 +86,13 @@ def run_e2e(args, rank: int, world: int, local_rank: int, have_gpu: bool):
 
     from sutro_amd.sdk import Sutro
 
+    from sutro_amd.models.registry import get_model_spec as _gms
+
     device = (f"cuda:{local_rank}" if have_gpu else "cpu")
     dist = dist_mod if world > 1 else None
-    model = args.model if have_gpu else "qwen-3-0.6b"
+    model = args.model
+    if not have_gpu and not _gms(model).embedding:
+        model = "qwen-3-0.6b"  # CPU smoke: small generative stand-in
     rows = args.e2e_rows if have_gpu else 24
     max_new = args.max_new if have_gpu else 8
     home = tempfile.mkdtemp(prefix=f"sutro-bench-{rank}-")
@@ -106,6 +110,39 @@ def run_e2e(args, rank: int, world: int, local_rank: int, have_gpu: bool):
     texts = [" ".join(rng.choice(words) for _ in range(args.prompt_len // 2))
              for _ in range(rows)]
     sp = {"max_tokens": max_new, "temperature": 0.8, "top_p": 0.95}
+
+    if _gms(model).embedding:
+        # BASELINE config 3 shape: embedding job through so.embed()
+        _ = client.embed(texts[:8], model=model)      # warmup
+        if dist is not None:
+            dist.barrier()
+        t0 = time.time()
+        df = client.embed(texts, model=model)
+        t1 = time.time()
+        assert len(df) == rows
+        elapsed = t1 - t0
+        if dist is not None:
+            te = torch.tensor([elapsed], dtype=torch.float64)
+            dist.all_reduce(te, op=dist.ReduceOp.MAX)
+            elapsed = float(te.item())
+            tr = torch.tensor([float(rows)], dtype=torch.float64)
+            dist.all_reduce(tr)
+            rows = int(tr.item())
+        n_gpus = world if world > 1 else args.gpus
+        if rank == 0:
+            print(json.dumps({
+                "metric": "embedding_rows_per_sec",
+                "value": round(rows / elapsed, 2), "unit": "rows/s",
+                "n_gpus": n_gpus, "steps": args.steps, "warmup": args.warmup,
+                "ms_per_step": None, "higher_is_better": True,
+                "scaling": "weak", "vs_baseline": None,
+                "dtype": "bf16" if have_gpu else "fp32", "data": "synthetic",
+                "config": {"model": model, "mode": "e2e_service_so.embed",
+                           "rows": rows, "parallelism": f"dp{n_gpus}",
+                           "rows_per_hour": round(rows / elapsed * 3600, 1),
+                           "elapsed_s": round(elapsed, 1)}}))
+        client.shutdown()
+        return
 
     schema = None
     if args.schema:
