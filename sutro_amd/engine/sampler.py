@@ -135,7 +135,6 @@ class Sampler:
         self._u_pinned = [None, None]
         self._u_flip = 0
         self._out_bufs: Optional[tuple] = None
-        self._use_hip = None  # resolved on first CUDA sample
 
     def _param_tensors(self, reqs: List[Request], device) -> tuple:
         # req_ids are unique for the engine's lifetime and a request's sampling

@@ -35,7 +35,6 @@ def _worker_main(rank: int, world: int, tp: int, port: int, model: str,
 
         from ..engine.engine import LLMEngine
         from ..engine.request import SamplingParams
-        from ..engine.tokenizer import get_tokenizer
         from .models_map import resolve_engine_config
 
         n_gpus = torch.cuda.device_count() if torch.cuda.is_available() else 0
