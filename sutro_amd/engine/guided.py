@@ -271,6 +271,119 @@ def _json_string_body(max_len: int = _MAX_STR) -> _Node:
     return rep(alt(_STR_CHAR, _ESCAPE), 0, max_len)
 
 
+# ---- ranged integers (digit-wise regex construction) ----
+
+def _digits_ge(s: str, allow_longer: bool = True) -> _Node:
+    """Unsigned integers >= int(s) (no leading zeros). Equal-length numbers
+    match digit by digit; optionally any number with MORE digits."""
+    n = len(s)
+    opts: List[_Node] = []
+    # equal length, >= s: for each position i, prefix s[:i], digit > s[i],
+    # free digits after; plus s itself
+    for i in range(n):
+        d = int(s[i])
+        if d < 9:
+            lead = [lit(s[:i])] if i else []
+            first = crange(str(d + 1), "9")
+            if i == 0 and n > 1:
+                pass  # first digit 1..9 handled by crange (d>=1 there)
+            opts.append(cat(*lead, first, rep(crange("0", "9"), n - 1 - i,
+                                              n - 1 - i)))
+    opts.append(lit(s))
+    if allow_longer:
+        # any number with more digits (no leading zero), up to 18 digits
+        if n < 18:
+            opts.append(cat(crange("1", "9"),
+                            rep(crange("0", "9"), n, 17)))
+    return alt(*opts)
+
+
+def _digits_le(s: str, fixed: bool = False) -> _Node:
+    """Unsigned integers <= int(s). Standalone: no leading zeros, shorter
+    numbers (and "0") allowed. fixed=True: exactly len(s) digit characters,
+    leading zeros allowed (digit-"rest" comparisons inside a range)."""
+    n = len(s)
+    opts: List[_Node] = [lit(s)]
+    # equal length, < s at position i
+    for i in range(n):
+        d = int(s[i])
+        floor = 1 if (not fixed and i == 0 and n > 1) else 0
+        if d > floor:
+            lead = [lit(s[:i])] if i else []
+            opts.append(cat(*lead, crange(str(floor), str(d - 1)),
+                            rep(crange("0", "9"), n - 1 - i, n - 1 - i)))
+    if not fixed:
+        # fewer digits
+        if n > 1:
+            opts.append(alt(lit("0"), cat(crange("1", "9"),
+                                          rep(crange("0", "9"), 0, n - 2))))
+        else:
+            opts.append(crange("0", s))  # n == 1: 0..s (supersedes lit(s))
+    return alt(*opts)
+
+
+def _digits_range(a: str, b: str) -> _Node:
+    """Unsigned integers in [int(a), int(b)] (0 <= a <= b)."""
+    if a == b:
+        return lit(a)
+    if len(a) == len(b):
+        # common prefix, then split on the first differing digit
+        i = 0
+        while a[i] == b[i]:
+            i += 1
+        pre = [lit(a[:i])] if i else []
+        n_rest = len(a) - i - 1
+        da, db = int(a[i]), int(b[i])
+        opts: List[_Node] = []
+        # first differing digit == da: rest >= a-rest
+        opts.append(cat(lit(a[i]), _digits_ge(a[i + 1:], allow_longer=False))
+                    if n_rest else lit(a[i]))
+        # strictly between
+        if db - da >= 2:
+            opts.append(cat(crange(str(da + 1), str(db - 1)),
+                            rep(crange("0", "9"), n_rest, n_rest)))
+        # first differing digit == db: rest <= b-rest (fixed width:
+        # leading zeros are legal in rest digits)
+        opts.append(cat(lit(b[i]), _digits_le(b[i + 1:], fixed=True))
+                    if n_rest else lit(b[i]))
+        return cat(*pre, alt(*opts))
+    # different lengths: [a, 10^len(a)-1] + full middle lengths + [10^(len(b)-1), b]
+    opts = [_digits_ge(a, allow_longer=False)]
+    for ln in range(len(a) + 1, len(b)):
+        opts.append(cat(crange("1", "9"), rep(crange("0", "9"), ln - 1,
+                                              ln - 1)))
+    opts.append(_digits_range("1" + "0" * (len(b) - 1), b))
+    return alt(*opts)
+
+
+def _int_range(lo, hi) -> _Node:
+    """Integers with optional inclusive bounds — exact at ANY magnitude
+    (the old path only enforced bounds it could enumerate)."""
+    if lo is None and hi is None:
+        return _INT
+    lo_i = None if lo is None else int(lo)
+    hi_i = None if hi is None else int(hi)
+    opts: List[_Node] = []
+    # negative side: -x with x in [max(1,-hi), -lo] (digit ranges flipped)
+    if lo_i is None or lo_i < 0:
+        neg_hi = None if lo_i is None else -lo_i          # largest magnitude
+        neg_lo = 1 if (hi_i is None or hi_i >= 0) else -hi_i
+        if neg_hi is None:
+            opts.append(cat(cls("-"), _digits_ge(str(neg_lo))))
+        elif neg_hi >= neg_lo:
+            opts.append(cat(cls("-"), _digits_range(str(neg_lo),
+                                                    str(neg_hi))))
+    # non-negative side: x in [max(0,lo), hi]
+    if hi_i is None or hi_i >= 0:
+        nn_lo = 0 if (lo_i is None or lo_i < 0) else lo_i
+        if hi_i is None:
+            opts.append(_digits_ge(str(nn_lo)) if nn_lo > 0
+                        else alt(lit("0"), _digits_ge("1")))
+        else:
+            opts.append(_digits_range(str(nn_lo), str(hi_i)))
+    return alt(*opts)
+
+
 def _escape_json(s: str) -> str:
     return s.replace("\\", "\\\\").replace('"', '\\"')
 
@@ -313,9 +426,11 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return cat(cls('"'), body, cls('"'))
     if t == "integer":
         lo, hi = schema.get("minimum"), schema.get("maximum")
-        if lo is not None and hi is not None and 0 <= hi - lo <= 4096:
+        if lo is not None and hi is not None and 0 <= hi - lo <= 256:
+            # tiny ranges enumerate (smallest DFA); anything else gets the
+            # exact digit-wise construction — bounds hold at ANY magnitude
             return alt(*[lit(str(v)) for v in range(int(lo), int(hi) + 1)])
-        return _INT
+        return _int_range(lo, hi)
     if t == "number":
         return _NUMBER
     if t == "boolean":

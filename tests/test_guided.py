@@ -207,3 +207,55 @@ def test_two_different_schemas_same_batch():
         assert json.loads(eng.output_text(r)) in ("red", "green")
     for r in r2:
         assert 10 <= int(eng.output_text(r)) <= 19
+
+
+def test_integer_range_any_magnitude():
+    """Ranged integers are exact at magnitudes far beyond enumeration —
+    digit-wise construction (the old path ignored un-enumerable bounds)."""
+    import random
+
+    from sutro_amd.engine.guided import _int_range, compile_dfa
+
+    rng = random.Random(42)
+    for _ in range(30):
+        lo = rng.randint(-10**12, 10**12)
+        hi = lo + rng.randint(0, 10**12)
+        dfa = compile_dfa(_int_range(lo, hi))
+        for _ in range(50):
+            v = rng.randint(lo, hi)
+            assert dfa.matches(str(v).encode()), (lo, hi, v)
+        for off in (1, 17, 10**6):
+            assert not dfa.matches(str(hi + off).encode()), (lo, hi, hi + off)
+            assert not dfa.matches(str(lo - off).encode()), (lo, hi, lo - off)
+        assert not dfa.matches(b"-")
+        assert not dfa.matches(b"007")
+
+
+def test_integer_range_guided_generation():
+    """End to end: masked walks over a big-range integer schema always emit
+    in-bounds values."""
+    import json
+
+    from sutro_amd.engine.config import EngineConfig
+    from sutro_amd.engine.engine import LLMEngine
+    from sutro_amd.engine.request import SamplingParams
+    from sutro_amd.models.registry import tiny_spec_for_tests
+
+    cfg = EngineConfig(spec=tiny_spec_for_tests(), device="cpu",
+                       max_model_len=256, num_kv_blocks=64,
+                       max_tokens_per_step=128)
+    eng = LLMEngine(cfg)
+    schema = {"type": "object", "properties": {
+        "qty": {"type": "integer", "minimum": 1500, "maximum": 2_000_000},
+        "delta": {"type": "integer", "minimum": -5_000_000, "maximum": -999}}}
+    fsm_id = eng.register_fsm(schema)
+    reqs = [eng.add_request(eng.tokenizer.encode(f"r{i}"),
+                            SamplingParams(max_tokens=64, temperature=1.2,
+                                           seed=i),
+                            fsm_id=fsm_id) for i in range(8)]
+    while eng.has_work():
+        eng.step()
+    for r in reqs:
+        obj = json.loads(eng.output_text(r))
+        assert 1500 <= obj["qty"] <= 2_000_000, obj
+        assert -5_000_000 <= obj["delta"] <= -999, obj
