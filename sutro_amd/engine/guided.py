@@ -271,6 +271,207 @@ def _json_string_body(max_len: int = _MAX_STR) -> _Node:
     return rep(alt(_STR_CHAR, _ESCAPE), 0, max_len)
 
 
+# ---- string "pattern" support (subset of Python/ECMA regex) ----
+
+# characters a generated JSON string may contain without escaping
+_SAFE_CHARS = frozenset(b for b in range(0x20, 0x7F) if b not in (0x22, 0x5C))
+# unbounded quantifiers are capped so guided generation always terminates
+# (random-weight models would otherwise run strings to max_tokens and emit
+# unterminated JSON) — mirrors the _MAX_STR cap on free strings
+_PATTERN_REP_CAP = 32
+# free-character slack on each un-anchored side of a pattern. Small on
+# purpose: two bounded wings over alphabets that overlap the pattern make
+# subset construction track (wing-pos x match-progress x wing-pos) — wing 16
+# already blows past 10^5 DFA states for patterns like \d+ (measured), wing
+# 8 stays ~10^3. Anchor patterns (^...$) for exact control.
+_PATTERN_WING = 8
+
+_CLASS_ESCAPES = {
+    "d": frozenset(range(0x30, 0x3A)),
+    "D": _SAFE_CHARS - frozenset(range(0x30, 0x3A)),
+    "w": frozenset(list(range(0x30, 0x3A)) + list(range(0x41, 0x5B))
+                   + list(range(0x61, 0x7B)) + [0x5F]),
+    "s": frozenset([0x20]),
+}
+_CLASS_ESCAPES["W"] = _SAFE_CHARS - _CLASS_ESCAPES["w"]
+_CLASS_ESCAPES["S"] = _SAFE_CHARS - _CLASS_ESCAPES["s"]
+
+
+class PatternError(ValueError):
+    pass
+
+
+def parse_pattern(pattern: str) -> Tuple[_Node, bool, bool]:
+    """Parse a (subset of) regex into the AST. Returns (node, anchored_start,
+    anchored_end). Supported: literals, '.', classes [] (ranges, ^negation),
+    \\d \\w \\s (+ negations), | ( ) ? * + {m} {m,n} {m,}, ^ $ at the ends.
+    Unsupported (raises PatternError): backrefs, lookaround, \\b, inline
+    flags, characters outside printable ASCII, literal '"' or '\\\\' (they
+    would need JSON escape interplay). Unbounded reps cap at 64."""
+    s = pattern
+    i = 0
+    anchored_start = s.startswith("^")
+    if anchored_start:
+        i = 1
+    anchored_end = s.endswith("$") and not s.endswith("\\$")
+    end = len(s) - 1 if anchored_end else len(s)
+
+    def peek():
+        return s[i] if i < end else None
+
+    def atom_char() -> FrozenSet[int]:
+        nonlocal i
+        c = s[i]
+        if c == "\\":
+            i += 1
+            if i >= end:
+                raise PatternError("trailing backslash")
+            e = s[i]
+            i += 1
+            if e in _CLASS_ESCAPES:
+                return _CLASS_ESCAPES[e]
+            if e in "bBAZ1234567890":
+                raise PatternError(f"unsupported escape \\{e}")
+            if e == "n" or e == "t" or e == "r":
+                raise PatternError("control characters not valid in the "
+                                   "compact JSON string charset")
+            b = ord(e)
+            if b not in _SAFE_CHARS:
+                raise PatternError(f"character {e!r} outside the JSON-safe "
+                                   f"charset")
+            return frozenset([b])
+        i += 1
+        if c == ".":
+            return _SAFE_CHARS
+        b = ord(c)
+        if b not in _SAFE_CHARS:
+            raise PatternError(f"character {c!r} outside the JSON-safe charset")
+        return frozenset([b])
+
+    def parse_class() -> FrozenSet[int]:
+        nonlocal i
+        assert s[i] == "["
+        i += 1
+        neg = peek() == "^"
+        if neg:
+            i += 1
+        out: Set[int] = set()
+        first = True
+        while True:
+            if i >= end:
+                raise PatternError("unterminated character class")
+            if s[i] == "]" and not first:
+                i += 1
+                break
+            first = False
+            if s[i] == "\\":
+                chars = atom_char()
+                out |= chars
+                continue
+            a = s[i]
+            i += 1
+            if peek() == "-" and i + 1 < end and s[i + 1] != "]":
+                i += 1
+                b = s[i]
+                i += 1
+                for v in range(ord(a), ord(b) + 1):
+                    if v in _SAFE_CHARS:
+                        out.add(v)
+            else:
+                if ord(a) not in _SAFE_CHARS:
+                    raise PatternError(f"character {a!r} outside the "
+                                       f"JSON-safe charset")
+                out.add(ord(a))
+        if neg:
+            out = set(_SAFE_CHARS) - out
+        if not out:
+            raise PatternError("empty character class")
+        return frozenset(out)
+
+    def parse_quant(node: _Node) -> _Node:
+        nonlocal i
+        c = peek()
+        if c == "*":
+            i += 1
+            return rep(node, 0, _PATTERN_REP_CAP)
+        if c == "+":
+            i += 1
+            return rep(node, 1, _PATTERN_REP_CAP)
+        if c == "?":
+            i += 1
+            return rep(node, 0, 1)
+        if c == "{":
+            j = s.index("}", i)
+            body = s[i + 1:j]
+            i = j + 1
+            if "," in body:
+                lo_s, hi_s = body.split(",", 1)
+                lo = int(lo_s or 0)
+                hi = int(hi_s) if hi_s else lo + _PATTERN_REP_CAP
+            else:
+                lo = hi = int(body)
+            if hi < lo or hi - lo > 4096:
+                raise PatternError(f"bad repetition {{{body}}}")
+            return rep(node, lo, hi)
+        return node
+
+    def parse_alt() -> _Node:
+        nonlocal i
+        options = [parse_cat()]
+        while peek() == "|":
+            i += 1
+            options.append(parse_cat())
+        return options[0] if len(options) == 1 else alt(*options)
+
+    def parse_cat() -> _Node:
+        nonlocal i
+        parts: List[_Node] = []
+        while i < end and s[i] not in "|)":
+            c = s[i]
+            if c in "^$":
+                raise PatternError("anchors only supported at the pattern "
+                                   "ends")
+            if c == "(":
+                i += 1
+                if peek() == "?":
+                    if i + 1 < end and s[i + 1] == ":":
+                        i += 2  # non-capturing group
+                    else:
+                        raise PatternError("lookaround/flags unsupported")
+                node = parse_alt()
+                if peek() != ")":
+                    raise PatternError("unbalanced parentheses")
+                i += 1
+            elif c == "[":
+                node = _Lit(parse_class())
+            else:
+                node = _Lit(atom_char())
+            parts.append(parse_quant(node))
+        return cat(*parts) if parts else cat()
+
+    node = parse_alt()
+    if i != end:
+        raise PatternError(f"unexpected {s[i]!r} at {i}")
+    return node, anchored_start, anchored_end
+
+
+def pattern_string_body(pattern: str, max_len: int = _MAX_STR) -> _Node:
+    """JSON-string body constrained by a schema "pattern". JSON Schema
+    patterns are UNANCHORED (re.search semantics): un-anchored ends get
+    bounded free-character wings of _PATTERN_WING chars (bounded so
+    generation terminates; small so the wing x pattern subset construction
+    stays ~10^3 states — anchor with ^...$ for exact strings)."""
+    inner, a_start, a_end = parse_pattern(pattern)
+    wing = rep(_Lit(_SAFE_CHARS), 0, min(_PATTERN_WING, max_len))
+    parts: List[_Node] = []
+    if not a_start:
+        parts.append(wing)
+    parts.append(inner)
+    if not a_end:
+        parts.append(wing)
+    return cat(*parts)
+
+
 # ---- ranged integers (digit-wise regex construction) ----
 
 def _digits_ge(s: str, allow_longer: bool = True) -> _Node:
@@ -421,6 +622,11 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return alt(*[schema_to_regex({**schema, "type": ti}, defs, depth + 1) for ti in t])
     if t == "string":
         ml = min(int(schema.get("maxLength", _MAX_STR)), _MAX_STR)
+        if "pattern" in schema:
+            # pattern strings: the regex constrains the body; the wing cap
+            # (unanchored ends) uses maxLength as the bound
+            return cat(cls('"'), pattern_string_body(schema["pattern"], ml),
+                       cls('"'))
         lo = max(0, min(int(schema.get("minLength", 0)), ml))
         body = rep(alt(_STR_CHAR, _ESCAPE), lo, ml)
         return cat(cls('"'), body, cls('"'))
