@@ -26,8 +26,20 @@ leaf = st.one_of(
     st.builds(lambda lo, w: {"type": "integer", "minimum": lo,
                              "maximum": lo + w},
               st.integers(-20, 50), st.integers(0, 40)),
+    # huge ranges: the digit-wise construction (no enumeration possible)
+    st.builds(lambda lo, w: {"type": "integer", "minimum": lo,
+                             "maximum": lo + w},
+              st.integers(-10**9, 10**9), st.integers(5000, 10**10)),
     st.just({"type": "number"}),
     st.just({"type": "string"}),
+    st.sampled_from([
+        {"type": "string", "pattern": r"^[A-Z]{2}-\d{3}$"},
+        {"type": "string", "pattern": r"^v\d+\.\d+$"},
+        {"type": "string", "pattern": r"^[a-z_][a-z0-9_]{0,11}$"},
+        {"type": "string", "format": "date"},
+        {"type": "string", "format": "uuid"},
+        {"type": "string", "format": "ipv4"},
+    ]),
     st.builds(lambda v: {"enum": v}, enum_vals),
     # nullable union + ranked permutation list (rank-template shapes)
     st.builds(lambda v: {"anyOf": [{"enum": v}, {"type": "null"}]}, enum_vals),
@@ -135,6 +147,24 @@ def validate(schema, value):
         assert isinstance(value, (int, float)) and not isinstance(value, bool)
     elif t == "string":
         assert isinstance(value, str)
+        if "pattern" in schema:
+            import re as _re
+
+            assert _re.search(schema["pattern"], value), (schema, value)
+        fmt = schema.get("format")
+        if fmt == "date":
+            import datetime as _dt
+
+            if not value.endswith("02-29"):
+                _dt.date.fromisoformat(value)
+        elif fmt == "uuid":
+            import uuid as _uuid
+
+            _uuid.UUID(value)
+        elif fmt == "ipv4":
+            import ipaddress as _ip
+
+            _ip.IPv4Address(value)
     elif t == "array":
         assert isinstance(value, list)
         if "minItems" in schema:
@@ -182,6 +212,34 @@ def sample_value(schema, data):
     if t == "number":
         return data.draw(st.integers(-99, 99))  # ints are valid numbers
     if t == "string":
+        if "pattern" in schema or "format" in schema:
+            gens = {
+                r"^[A-Z]{2}-\d{3}$": lambda d: (
+                    "".join(d.draw(st.sampled_from("ABCXYZ"))
+                            for _ in range(2)) + "-" +
+                    "".join(str(d.draw(st.integers(0, 9)))
+                            for _ in range(3))),
+                r"^v\d+\.\d+$": lambda d: (
+                    f"v{d.draw(st.integers(0, 99))}."
+                    f"{d.draw(st.integers(0, 99))}"),
+                r"^[a-z_][a-z0-9_]{0,11}$": lambda d: (
+                    d.draw(st.sampled_from("az_")) +
+                    "".join(d.draw(st.sampled_from("a0z_9"))
+                            for _ in range(d.draw(st.integers(0, 5))))),
+            }
+            fmts = {
+                "date": lambda d: (f"{d.draw(st.integers(1000, 2999))}-"
+                                   f"{d.draw(st.integers(1, 12)):02d}-"
+                                   f"{d.draw(st.integers(1, 28)):02d}"),
+                "uuid": lambda d: "-".join(
+                    "".join(d.draw(st.sampled_from("0123456789abcdef"))
+                            for _ in range(n)) for n in (8, 4, 4, 4, 12)),
+                "ipv4": lambda d: ".".join(
+                    str(d.draw(st.integers(0, 255))) for _ in range(4)),
+            }
+            if "pattern" in schema:
+                return gens[schema["pattern"]](data)
+            return fmts[schema["format"]](data)
         return data.draw(st.text(
             alphabet=st.characters(min_codepoint=32, max_codepoint=126,
                                    blacklist_characters='"\\'),
