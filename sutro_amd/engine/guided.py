@@ -17,6 +17,7 @@ byte-path is walkable by tokens — no token-level dead ends.
 
 from __future__ import annotations
 
+import math
 from dataclasses import dataclass
 from typing import Dict, FrozenSet, List, Optional, Set, Tuple
 
@@ -575,6 +576,178 @@ def _digits_range(a: str, b: str) -> _Node:
     return alt(*opts)
 
 
+def _frac_eq(f: str) -> _Node:
+    """Fraction digit strings whose zero-padded value equals f exactly
+    (the literal plus every trailing-zero truncation)."""
+    opts: List[_Node] = [lit(f)]
+    for k in range(1, 8):
+        if f[k:] == "0" * (8 - k):
+            opts.append(lit(f[:k]))
+    return alt(*opts)
+
+
+def _frac_ge(flo: str) -> _Node:
+    """Fraction digit strings d1..dk (1..8 digits) whose zero-padded 8-digit
+    value is >= flo (8 digits). A string may stop early only where the
+    remaining flo digits are all zero."""
+    opts: List[_Node] = []
+    for i in range(8):
+        d = int(flo[i])
+        pre = [lit(flo[:i])] if i else []
+        if d < 9:
+            # digit > flo[i], then 0..7-i free digits
+            opts.append(cat(*pre, crange(str(d + 1), "9"),
+                            rep(crange("0", "9"), 0, 7 - i)))
+        if flo[i + 1:] == "0" * (7 - i):
+            # stopping right after matching digit i is >= (rest of flo is 0)
+            opts.append(cat(*pre, lit(flo[i])))
+    opts.append(lit(flo))
+    return alt(*opts)
+
+
+def _frac_le(fhi: str) -> _Node:
+    """Fraction digit strings d1..dk (1..8 digits) whose zero-padded value
+    is <= fhi (8 digits). Stopping early pads with zeros (always <= when the
+    prefix matches); trailing free zeros allowed after a smaller digit."""
+    opts: List[_Node] = []
+    for i in range(8):
+        d = int(fhi[i])
+        pre = [lit(fhi[:i])] if i else []
+        if d > 0:
+            opts.append(cat(*pre, crange("0", str(d - 1)),
+                            rep(crange("0", "9"), 0, 7 - i)))
+        # stop exactly at digit i (prefix of fhi pads to <= fhi)
+        opts.append(cat(*pre, lit(fhi[i])))
+    return alt(*opts)
+
+
+def _nonneg_number_range(lo8: int, hi8: int) -> _Node:
+    """Decimal number strings I[.F] (F 1..8 digits, value on the 1e-8 grid)
+    with lo8 <= value*1e8 <= hi8, 0 <= lo8 <= hi8."""
+    loI, loF = divmod(lo8, 10**8)
+    hiI, hiF = divmod(hi8, 10**8)
+    frac_any = cat(cls("."), rep(crange("0", "9"), 1, 8))
+    opts: List[_Node] = []
+    if hiI - loI >= 2:
+        mid = _digits_range(str(loI + 1), str(hiI - 1))
+        opts.append(cat(mid, opt(frac_any)))
+    lo_digits = f"{loF:08d}"
+    hi_digits = f"{hiF:08d}"
+    if loI == hiI:
+        both = [cat(cls("."), alt_intersect_ge_le(lo_digits, hi_digits))] \
+            if loF > 0 else [cat(cls("."), _frac_le(hi_digits))]
+        if loF == 0:
+            both.append(cat())  # bare integer == loI
+        opts.append(cat(lit(str(loI)), alt(*both)))
+    else:
+        # I == loI: fraction >= loF (bare integer only if loF == 0)
+        lo_parts: List[_Node] = [cat(cls("."), _frac_ge(lo_digits))]
+        if loF == 0:
+            lo_parts.append(cat())
+            lo_parts[0] = cat(cls("."), rep(crange("0", "9"), 1, 8))
+        opts.append(cat(lit(str(loI)), alt(*lo_parts)))
+        # I == hiI: fraction <= hiF; bare integer always fine
+        opts.append(cat(lit(str(hiI)),
+                        alt(cat(), cat(cls("."), _frac_le(hi_digits)))))
+    return alt(*opts)
+
+
+def alt_intersect_ge_le(lo_digits: str, hi_digits: str) -> _Node:
+    """Fraction strings with lo <= padded value <= hi (same integer part).
+    Built digit-wise on the common prefix."""
+    assert lo_digits <= hi_digits
+    i = 0
+    while i < 8 and lo_digits[i] == hi_digits[i]:
+        i += 1
+    if i == 8:
+        return _frac_eq(lo_digits)  # lo == hi: exactly that value
+    pre = [lit(lo_digits[:i])] if i else []
+    da, db = int(lo_digits[i]), int(hi_digits[i])
+    opts: List[_Node] = []
+    # first differing digit == da: rest >= lo-rest (within remaining width)
+    opts.append(cat(lit(lo_digits[i]),
+                    _frac_ge_w(lo_digits[i + 1:], 7 - i)))
+    if db - da >= 2:
+        opts.append(cat(crange(str(da + 1), str(db - 1)),
+                        rep(crange("0", "9"), 0, 7 - i)))
+    opts.append(cat(lit(hi_digits[i]), opt(_frac_le_w(hi_digits[i + 1:],
+                                                      7 - i))))
+    # stopping AT the boundary digit: prefix+da pads to < lo unless rest of
+    # lo is zero; prefix+db pads to <= hi always — handled inside the helpers
+    full = cat(*pre, alt(*opts))
+    # stopping INSIDE the common prefix: pads with zeros — valid iff that
+    # equals lo (<= hi holds: it is a prefix of hi too)
+    stops = [lit(lo_digits[:k]) for k in range(1, i + 1)
+             if lo_digits[k:] == "0" * (8 - k)]
+    return alt(*stops, full) if stops else full
+
+
+def _frac_ge_w(flo: str, width: int) -> _Node:
+    """Like _frac_ge but over 0..width digits (suffix position inside a
+    longer fraction)."""
+    if width == 0 or int(flo or "0") == 0:
+        return rep(crange("0", "9"), 0, width)
+    opts: List[_Node] = []
+    for i in range(width):
+        d = int(flo[i])
+        pre = [lit(flo[:i])] if i else []
+        if d < 9:
+            opts.append(cat(*pre, crange(str(d + 1), "9"),
+                            rep(crange("0", "9"), 0, width - 1 - i)))
+        if flo[i + 1:] == "0" * (width - 1 - i):
+            opts.append(cat(*pre, lit(flo[i])))
+    opts.append(lit(flo))
+    return alt(*opts)
+
+
+def _frac_le_w(fhi: str, width: int) -> _Node:
+    """Like _frac_le but over 0..width digits."""
+    if width == 0:
+        return cat()
+    opts: List[_Node] = [cat()]
+    for i in range(width):
+        d = int(fhi[i])
+        pre = [lit(fhi[:i])] if i else []
+        if d > 0:
+            opts.append(cat(*pre, crange("0", str(d - 1)),
+                            rep(crange("0", "9"), 0, width - 1 - i)))
+        opts.append(cat(*pre, lit(fhi[i])))
+    return alt(*opts)
+
+
+def _number_range(lo, hi, excl_lo: bool = False,
+                  excl_hi: bool = False) -> _Node:
+    """Decimal numbers (up to 8 fraction digits) with optional bounds —
+    exact on the 1e-8 grid the regex can emit (exclusive bounds move one
+    grid step inward)."""
+    if lo is None and hi is None:
+        return _NUMBER
+    scale = 10**8
+    lo8 = None if lo is None else math.ceil(float(lo) * scale - 1e-6)
+    hi8 = None if hi is None else math.floor(float(hi) * scale + 1e-6)
+    if excl_lo and lo8 is not None and abs(float(lo) * scale - lo8) < 1e-6:
+        lo8 += 1
+    if excl_hi and hi8 is not None and abs(float(hi) * scale - hi8) < 1e-6:
+        hi8 -= 1
+    big = 10**18
+    if lo8 is None:
+        lo8 = -big
+    if hi8 is None:
+        hi8 = big
+    opts: List[_Node] = []
+    if lo8 < 0:
+        neg_lo8 = 1 if hi8 >= 0 else -hi8
+        neg_hi8 = -lo8
+        if neg_hi8 >= neg_lo8:
+            opts.append(cat(cls("-"), _nonneg_number_range(neg_lo8, neg_hi8)))
+        if hi8 >= 0 >= lo8:
+            opts.append(lit("0"))  # plain zero (also "-0"-free canonical)
+    if hi8 >= 0:
+        nn_lo8 = max(0, lo8)
+        opts.append(_nonneg_number_range(nn_lo8, hi8))
+    return alt(*opts)
+
+
 def _int_range(lo, hi) -> _Node:
     """Integers with optional inclusive bounds — exact at ANY magnitude
     (the old path only enforced bounds it could enumerate)."""
@@ -660,7 +833,20 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
             return alt(*[lit(str(v)) for v in range(int(lo), int(hi) + 1)])
         return _int_range(lo, hi)
     if t == "number":
-        return _NUMBER
+        lo = schema.get("minimum")
+        hi = schema.get("maximum")
+        elo = schema.get("exclusiveMinimum")
+        ehi = schema.get("exclusiveMaximum")
+        if lo is None and elo is not None:
+            lo = elo
+        if hi is None and ehi is not None:
+            hi = ehi
+        if lo is None and hi is None:
+            return _NUMBER
+        if lo is not None and hi is not None and lo > hi:
+            raise ValueError(f"empty number range [{lo}, {hi}]")
+        return _number_range(lo, hi, excl_lo=elo is not None,
+                             excl_hi=ehi is not None)
     if t == "boolean":
         return alt(lit("true"), lit("false"))
     if t == "null":

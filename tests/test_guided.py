@@ -383,3 +383,63 @@ def test_string_format_support():
             chk(json.loads(b.decode()))
             produced += 1
         assert produced > 30, fmt
+
+
+def test_number_bounds_exact():
+    """number minimum/maximum (and exclusive*) hold exactly on the 1e-8
+    decimal grid: generated walks stay inside, in-range grid values are
+    accepted, out-of-range values rejected."""
+    import json
+    import random
+
+    from sutro_amd.engine.guided import compile_dfa, schema_to_regex
+
+    rng = random.Random(2)
+
+    def walk(dfa, max_steps=120):
+        st, out = 0, []
+        for _ in range(max_steps):
+            opts = list(dfa.transitions[st].items())
+            if st in dfa.accepting and (not opts or rng.random() < 0.35):
+                return bytes(out)
+            if not opts:
+                return bytes(out) if st in dfa.accepting else None
+            b, t = rng.choice(opts)
+            out.append(b)
+            st = t
+        return None
+
+    def fmt(x):
+        s = f"{x:.8f}".rstrip("0").rstrip(".")
+        return s if s not in ("", "-") else "0"
+
+    for lo, hi in [(0.25, 0.75), (-1.5, 2.5), (10, 10.5), (2.5, 2.5),
+                   (0.123, 0.1234), (None, -3.25), (0.1, None)]:
+        sch = {"type": "number"}
+        if lo is not None:
+            sch["minimum"] = lo
+        if hi is not None:
+            sch["maximum"] = hi
+        dfa = compile_dfa(schema_to_regex(sch))
+        for _ in range(200):
+            b = walk(dfa)
+            if b is None:
+                continue
+            v = json.loads(b.decode())
+            assert (lo is None or v >= lo - 1e-12), (lo, hi, b)
+            assert (hi is None or v <= hi + 1e-12), (lo, hi, b)
+        lo8 = int(round((lo if lo is not None else -100) * 10**8))
+        hi8 = int(round((hi if hi is not None else 100) * 10**8))
+        for _ in range(200):
+            v8 = rng.randint(lo8, hi8)
+            assert dfa.matches(fmt(v8 / 10**8).encode()), (lo, hi, v8)
+        if lo is not None:
+            assert not dfa.matches(fmt((lo8 - 1) / 10**8).encode())
+        if hi is not None:
+            assert not dfa.matches(fmt((hi8 + 1) / 10**8).encode())
+
+    dfa = compile_dfa(schema_to_regex(
+        {"type": "number", "exclusiveMinimum": 0, "maximum": 1}))
+    assert not dfa.matches(b"0")
+    assert dfa.matches(b"0.00000001")
+    assert dfa.matches(b"1")
