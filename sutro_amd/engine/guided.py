@@ -852,6 +852,23 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
     if t == "null":
         return lit("null")
     if t == "array":
+        if "prefixItems" in schema:
+            # pydantic tuples: fixed positional item schemas
+            parts: List[_Node] = [lit("[")]
+            for i, sub in enumerate(schema["prefixItems"]):
+                if i:
+                    parts.append(lit(","))
+                parts.append(schema_to_regex(sub, defs, depth + 1))
+            extra = schema.get("items")
+            hi = min(int(schema.get("maxItems", len(schema["prefixItems"]))),
+                     _MAX_ARR)
+            if extra not in (None, False) and hi > len(schema["prefixItems"]):
+                more = cat(lit(","),
+                           schema_to_regex(extra if isinstance(extra, dict)
+                                           else {}, defs, depth + 1))
+                parts.append(rep(more, 0, hi - len(schema["prefixItems"])))
+            parts.append(lit("]"))
+            return cat(*parts)
         items_schema = schema.get("items", {})
         # uniqueItems over a small enum: enumerate permutations (exact
         # distinctness is beyond a DFA product otherwise) — rank() uses this
@@ -882,6 +899,14 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
     if t == "object" or "properties" in schema:
         props = schema.get("properties", {})
         if not props:
+            ap = schema.get("additionalProperties")
+            if isinstance(ap, dict):
+                # pydantic Dict[str, X]: 0..8 entries with short free keys
+                key = cat(cls('"'), rep(_STR_CHAR, 1, 24), cls('"'))
+                entry = cat(key, lit(":"),
+                            schema_to_regex(ap, defs, depth + 1))
+                body = cat(entry, rep(cat(lit(","), entry), 0, 7))
+                return cat(lit("{"), opt(body), lit("}"))
             return lit("{}")
         parts: List[_Node] = [lit("{")]
         for i, (name, sub) in enumerate(props.items()):

@@ -497,3 +497,38 @@ def test_pydantic_v2_model_end_to_end():
         uuid_mod.UUID(obj["uid"])
         assert obj["tags"] is None or (isinstance(obj["tags"], list)
                                        and len(obj["tags"]) <= 3)
+
+
+def test_pydantic_tuple_and_dict_fields():
+    """prefixItems (tuples) and additionalProperties (dicts) generate
+    correctly typed values."""
+    import json
+    import random
+    from typing import Dict, Tuple
+
+    from pydantic import BaseModel
+
+    from sutro_amd.engine.guided import GuidedFSM
+    from sutro_amd.engine.tokenizer import EOS_ID, get_tokenizer
+
+    class M(BaseModel):
+        t: Tuple[str, int]
+        d: Dict[str, int]
+
+    fsm = GuidedFSM.from_schema(M.model_json_schema(),
+                                tokenizer=get_tokenizer(2048))
+    rng = random.Random(1)
+    for _ in range(8):
+        st, out = fsm.start_state(), []
+        for _ in range(4000):
+            allowed = fsm.mask_for(st).nonzero().flatten().tolist()
+            assert allowed
+            tk = rng.choice(allowed)
+            if tk == EOS_ID:
+                break
+            out.extend(fsm.tokenizer.token_bytes(tk))
+            st = fsm.advance(st, tk)
+        obj = json.loads(bytes(out).decode())
+        assert len(obj["t"]) == 2
+        assert isinstance(obj["t"][0], str) and isinstance(obj["t"][1], int)
+        assert all(isinstance(v, int) for v in obj["d"].values())
