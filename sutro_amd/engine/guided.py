@@ -804,6 +804,13 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
         return alt(*opts)
     if "const" in schema:
         return schema_to_regex({"enum": [schema["const"]]}, defs, depth)
+    if "allOf" in schema and len(schema["allOf"]) == 1:
+        # single-element allOf (pydantic v1-style $ref wrapping): merge the
+        # sibling constraints over the inner schema
+        inner = schema["allOf"][0]
+        merged = {**inner, **{k: v for k, v in schema.items()
+                              if k not in ("allOf", "title", "description")}}
+        return schema_to_regex(merged, defs, depth + 1)
     if "anyOf" in schema or "oneOf" in schema:
         options = schema.get("anyOf") or schema.get("oneOf")
         return alt(*[schema_to_regex(o, defs, depth + 1) for o in options])
