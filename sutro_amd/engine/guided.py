@@ -915,14 +915,47 @@ def schema_to_regex(schema: dict, defs: Optional[dict] = None, depth: int = 0) -
                 body = cat(entry, rep(cat(lit(","), entry), 0, 7))
                 return cat(lit("{"), opt(body), lit("}"))
             return lit("{}")
-        parts: List[_Node] = [lit("{")]
-        for i, (name, sub) in enumerate(props.items()):
-            if i > 0:
-                parts.append(lit(","))
-            parts.append(lit(f'"{_escape_json(name)}":'))
-            parts.append(schema_to_regex(sub, defs, depth + 1))
-        parts.append(lit("}"))
-        return cat(*parts)
+        req = schema.get("required")
+        # no "required" key => every property emitted (a pydantic model
+        # always lists its required fields; internal/compact schemas keep
+        # the always-emit behavior)
+        required = set(props) if req is None else set(req)
+        entries = [(name, schema_to_regex(sub, defs, depth + 1),
+                    name in required) for name, sub in props.items()]
+
+        def prop_node(idx: int) -> _Node:
+            name, val, _ = entries[idx]
+            return cat(lit(f'"{_escape_json(name)}":'), val)
+
+        if all(r for _, _, r in entries):
+            parts: List[_Node] = [lit("{")]
+            for i in range(len(entries)):
+                if i > 0:
+                    parts.append(lit(","))
+                parts.append(prop_node(i))
+            parts.append(lit("}"))
+            return cat(*parts)
+
+        # optional omission, declaration order, linear construction: pick
+        # the FIRST emitted property j (any optional before the first
+        # required, or that required itself); everything after j appears as
+        # ",prop" — mandatory when required, optional otherwise
+        n = len(entries)
+
+        def suffix_from(k: int) -> _Node:
+            parts: List[_Node] = []
+            for i in range(k, n):
+                item = cat(lit(","), prop_node(i))
+                parts.append(item if entries[i][2] else opt(item))
+            return cat(*parts)
+
+        first_req = next((i for i, e in enumerate(entries) if e[2]), n)
+        heads = [cat(prop_node(j), suffix_from(j + 1))
+                 for j in range(min(first_req + 1, n))]
+        body = alt(*heads)
+        if not required:
+            body = opt(body)  # fully-optional object may be {}
+        return cat(lit("{"), body, lit("}"))
     # unconstrained: any scalar JSON value
     return alt(
         cat(cls('"'), _json_string_body(), cls('"')),

@@ -57,11 +57,20 @@ def arr(inner):
 
 
 def obj(inner):
+    def build(props, req_mask):
+        properties = {k: v for k, v in props}
+        schema = {"type": "object", "properties": properties}
+        keys = list(properties)
+        required = [k for k, m in zip(keys, req_mask) if m]
+        if len(required) < len(keys):  # exercise optional omission
+            schema["required"] = required
+        return schema
+
     return st.builds(
-        lambda props: {"type": "object",
-                       "properties": {k: v for k, v in props}},
+        build,
         st.lists(st.tuples(names, inner), min_size=1, max_size=3,
-                 unique_by=lambda t: t[0]))
+                 unique_by=lambda t: t[0]),
+        st.lists(st.booleans(), min_size=3, max_size=3))
 
 
 schemas = st.recursive(leaf, lambda inner: st.one_of(arr(inner), obj(inner)),
@@ -177,9 +186,14 @@ def validate(schema, value):
             validate(schema["items"], v)
     elif t == "object":
         assert isinstance(value, dict)
-        for k, sub in schema.get("properties", {}).items():
-            assert k in value, (k, value)
-            validate(sub, value[k])
+        props = schema.get("properties", {})
+        required = set(props) if "required" not in schema \
+            else set(schema["required"])
+        assert set(value) <= set(props), (schema, value)
+        assert required <= set(value), (schema, value)
+        for k, sub in props.items():
+            if k in value:
+                validate(sub, value[k])
 
 
 @settings(max_examples=60, deadline=None)
@@ -253,8 +267,14 @@ def sample_value(schema, data):
         return [sample_value(schema["items"], data)
                 for _ in range(data.draw(st.integers(lo, hi)))]
     if t == "object":
-        return {k: sample_value(sub, data)
-                for k, sub in schema.get("properties", {}).items()}
+        props = schema.get("properties", {})
+        required = set(props) if "required" not in schema \
+            else set(schema["required"])
+        out = {}
+        for k, sub in props.items():
+            if k in required or data.draw(st.booleans()):
+                out[k] = sample_value(sub, data)
+        return out
     raise AssertionError(f"unhandled schema {schema}")
 
 
