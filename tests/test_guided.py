@@ -532,3 +532,31 @@ def test_pydantic_tuple_and_dict_fields():
         assert len(obj["t"]) == 2
         assert isinstance(obj["t"][0], str) and isinstance(obj["t"][1], int)
         assert all(isinstance(v, int) for v in obj["d"].values())
+
+
+def test_required_optional_properties_exhaustive():
+    """`required` honored: every declaration-ordered subset containing the
+    required keys is accepted, everything else rejected."""
+    import itertools
+    import json
+
+    from sutro_amd.engine.guided import compile_dfa, schema_to_regex
+
+    schema = {"type": "object",
+              "properties": {"a": {"enum": [1]}, "b": {"enum": [2]},
+                             "c": {"enum": [3]}, "d": {"enum": [4]}},
+              "required": ["b", "d"]}
+    dfa = compile_dfa(schema_to_regex(schema))
+    names, vals = ["a", "b", "c", "d"], {"a": 1, "b": 2, "c": 3, "d": 4}
+    for r in range(5):
+        for keys in itertools.combinations(names, r):
+            txt = json.dumps({k: vals[k] for k in keys},
+                             separators=(",", ":"))
+            valid = {"b", "d"} <= set(keys)
+            assert dfa.matches(txt.encode()) == valid, (keys, valid)
+    assert not dfa.matches(b'{"d":4,"b":2}')  # declaration order enforced
+    # fully-optional object may be {}
+    d2 = compile_dfa(schema_to_regex(
+        {"type": "object", "properties": {"x": {"enum": [0]}},
+         "required": []}))
+    assert d2.matches(b"{}") and d2.matches(b'{"x":0}')
