@@ -504,11 +504,8 @@ def _digits_ge(s: str, allow_longer: bool = True) -> _Node:
         d = int(s[i])
         if d < 9:
             lead = [lit(s[:i])] if i else []
-            first = crange(str(d + 1), "9")
-            if i == 0 and n > 1:
-                pass  # first digit 1..9 handled by crange (d>=1 there)
-            opts.append(cat(*lead, first, rep(crange("0", "9"), n - 1 - i,
-                                              n - 1 - i)))
+            opts.append(cat(*lead, crange(str(d + 1), "9"),
+                            rep(crange("0", "9"), n - 1 - i, n - 1 - i)))
     opts.append(lit(s))
     if allow_longer:
         # any number with more digits (no leading zero), up to 18 digits
@@ -634,7 +631,7 @@ def _nonneg_number_range(lo8: int, hi8: int) -> _Node:
     lo_digits = f"{loF:08d}"
     hi_digits = f"{hiF:08d}"
     if loI == hiI:
-        both = [cat(cls("."), alt_intersect_ge_le(lo_digits, hi_digits))] \
+        both = [cat(cls("."), _frac_between(lo_digits, hi_digits))] \
             if loF > 0 else [cat(cls("."), _frac_le(hi_digits))]
         if loF == 0:
             both.append(cat())  # bare integer == loI
@@ -652,9 +649,10 @@ def _nonneg_number_range(lo8: int, hi8: int) -> _Node:
     return alt(*opts)
 
 
-def alt_intersect_ge_le(lo_digits: str, hi_digits: str) -> _Node:
-    """Fraction strings with lo <= padded value <= hi (same integer part).
-    Built digit-wise on the common prefix."""
+def _frac_between(lo_digits: str, hi_digits: str) -> _Node:
+    """Fraction digit strings with lo <= zero-padded value <= hi (used when
+    both bounds share the integer part). Built digit-wise on the common
+    prefix."""
     assert lo_digits <= hi_digits
     i = 0
     while i < 8 and lo_digits[i] == hi_digits[i]:
