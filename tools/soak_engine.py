@@ -26,6 +26,14 @@ SCHEMAS = [
     {"type": "object", "properties": {
         "label": {"enum": ["a", "b", "c"]},
         "score": {"type": "integer", "minimum": 0, "maximum": 20}}},
+    {"type": "object", "properties": {
+        "label": {"enum": ["a", "b"]},
+        "note": {"type": "string", "maxLength": 6},
+        "n": {"type": "number", "minimum": -2.5, "maximum": 7.25}},
+     "required": ["label"]},
+    {"type": "object", "properties": {
+        "sku": {"type": "string", "pattern": r"^[A-Z]-\d{2}$"},
+        "big": {"type": "integer", "minimum": 10**6, "maximum": 10**9}}},
     {"enum": ["red", "green", "blue"]},
     {"type": "array", "items": {"type": "integer", "minimum": -5,
                                 "maximum": 5}, "minItems": 1, "maxItems": 3},
@@ -42,9 +50,13 @@ def validate(schema, value):
     t = schema.get("type")
     if t == "object":
         assert isinstance(value, dict)
-        for kname, sub in schema.get("properties", {}).items():
-            assert kname in value
-            validate(sub, value[kname])
+        props = schema.get("properties", {})
+        required = set(props) if "required" not in schema \
+            else set(schema["required"])
+        assert required <= set(value) <= set(props), (schema, value)
+        for kname, sub in props.items():
+            if kname in value:
+                validate(sub, value[kname])
     elif t == "array":
         assert isinstance(value, list)
         assert schema.get("minItems", 0) <= len(value) <= schema.get(
@@ -57,10 +69,21 @@ def validate(schema, value):
             "maximum", 1 << 60)
     elif t == "boolean":
         assert isinstance(value, bool)
+    elif t == "number":
+        assert isinstance(value, (int, float))
+        if "minimum" in schema:
+            assert value >= schema["minimum"] - 1e-9
+        if "maximum" in schema:
+            assert value <= schema["maximum"] + 1e-9
     elif t == "string":
         assert isinstance(value, str)
-        assert schema.get("minLength", 0) <= len(value) <= schema.get(
-            "maxLength", 1 << 30)
+        if "pattern" in schema:
+            import re
+
+            assert re.search(schema["pattern"], value), (schema, value)
+        else:
+            assert schema.get("minLength", 0) <= len(value) <= schema.get(
+                "maxLength", 1 << 30)
 
 
 def run_wave(rng: random.Random, wave_seed: int):
