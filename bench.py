@@ -178,8 +178,8 @@ def run_e2e(args, rank: int, world: int, local_rank: int, have_gpu: bool):
     df = client.get_job_results(job_id)
     t1 = time.time()
     assert len(df) == rows, f"{len(df)} != {rows}"
-    job = client._local_transport().service.jobs[job_id]
-    out_tokens = float(job.output_tokens)
+    job = client.do_request("GET", f"/jobs/{job_id}").json()["job"]
+    out_tokens = float(job["output_tokens"])
     elapsed = t1 - t0
     if dist is not None:
         te = torch.tensor([elapsed], dtype=torch.float64)
