@@ -756,6 +756,14 @@ class Sutro(ObservabilityMixin, EmbeddingTemplates, ClassificationTemplates,
                 time.sleep(POLL_INTERVAL)
         raise TimeoutError(f"job {job_id} did not complete within {timeout}s")
 
+    def fancy_tqdm(self, *args, **kwargs):
+        """Styled tqdm factory, also exposed as a client method exactly like
+        the reference (`/root/reference/sutro/sdk.py:966-1023` duplicates
+        `common.fancy_tqdm` onto the class)."""
+        from .common import fancy_tqdm as _fancy
+
+        return _fancy(*args, **kwargs)
+
     # ---- cache management (CLI) ----
 
     def _clear_job_results_cache(self) -> int:
